@@ -1,0 +1,114 @@
+"""On-device LLM generation engine (replaces the reference's OpenAI-compatible
+HTTP calls, reference src/core/llm/providers/openai.py:117-121; K6/K7 in
+SURVEY §2.3).
+
+Llama-3-class decoder: batched prefill (flash-style MFMA attention) + KV-cache
+decode (single-pass online-softmax kernel), on-device sampling, optional
+per-token streaming callback (the reference streamed SSE chunks).
+Temperature-per-mode mirrors reference generator.py:262-267
+(fast=0.0, balanced=0.3, quality=0.2, creative=0.7).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Iterator
+
+import torch
+
+from sentio_amd import ops
+from sentio_amd.engines.configs import get_model_config
+from sentio_amd.engines.tokenizer import EOS_ID, ByteTokenizer
+from sentio_amd.engines.transformer import KVCache, Transformer
+
+MODE_TEMPERATURE = {"fast": 0.0, "balanced": 0.3, "quality": 0.2, "creative": 0.7}
+
+
+class GeneratorEngine:
+    def __init__(self, model: str = "llama3-8b", device: str = "cpu",
+                 dtype: str = "bf16", max_seq: int = 4096, seed: int = 303):
+        self.cfg = get_model_config(model)
+        self.device = device
+        self.max_seq = min(max_seq, self.cfg.max_seq)
+        self.tokenizer = ByteTokenizer()
+        self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
+        self._step_seed = 0
+
+    @torch.inference_mode()
+    def generate(
+        self,
+        prompts: list[str],
+        max_new_tokens: int = 128,
+        temperature: float = 0.3,
+        stop_on_eos: bool = True,
+        on_token: Callable[[int, list[int]], None] | None = None,
+    ) -> list[str]:
+        """Batched generation.  Returns decoded completions (prompt excluded).
+
+        on_token(step, token_ids_per_batch) fires after every decode step —
+        the serving layer turns it into SSE streaming.
+        """
+        if not prompts:
+            return []
+        B = len(prompts)
+        prompt_budget = self.max_seq - max_new_tokens - 1
+        padded, lens = self.tokenizer.encode_batch(
+            [p[-4 * prompt_budget:] for p in prompts], prompt_budget
+        )
+        tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
+        S = tokens.shape[1]
+        cache = KVCache(self.cfg, B, min(S + max_new_tokens + 1, self.max_seq),
+                        self.device, self.model.dtype)
+        logits = self.model.prefill(tokens, cache)
+
+        finished = torch.zeros(B, dtype=torch.bool, device=self.device)
+        generated: list[list[int]] = [[] for _ in range(B)]
+        cur = self._sample(logits, temperature)
+        for step in range(max_new_tokens):
+            cur_list = cur.cpu().tolist()
+            for b, t in enumerate(cur_list):
+                if not finished[b]:
+                    generated[b].append(int(t))
+            if on_token is not None:
+                on_token(step, cur_list)
+            if stop_on_eos:
+                finished |= cur == EOS_ID
+                if bool(finished.all()):
+                    break
+            if step == max_new_tokens - 1:
+                break
+            logits = self.model.decode_step(cur.unsqueeze(1), cache)
+            cur = self._sample(logits, temperature)
+        return [self.tokenizer.decode(g) for g in generated]
+
+    def _sample(self, logits: torch.Tensor, temperature: float) -> torch.Tensor:
+        self._step_seed += 1
+        return ops.sample_token(logits, temperature, seed=self._step_seed)
+
+    @torch.inference_mode()
+    def stream(self, prompt: str, max_new_tokens: int = 128,
+               temperature: float = 0.3) -> Iterator[str]:
+        """Yield text deltas token-by-token (single prompt) — the on-device
+        equivalent of the reference's SSE streaming (openai.py:149-157)."""
+        prompt_budget = self.max_seq - max_new_tokens - 1
+        ids = self.tokenizer.encode(prompt[-4 * prompt_budget:], prompt_budget)
+        tokens = torch.tensor([ids], dtype=torch.int64, device=self.device)
+        cache = KVCache(self.cfg, 1, min(len(ids) + max_new_tokens + 1, self.max_seq),
+                        self.device, self.model.dtype)
+        logits = self.model.prefill(tokens, cache)
+        cur = self._sample(logits, temperature)
+        generated: list[int] = []
+        emitted = ""
+        for step in range(max_new_tokens):
+            t = int(cur.item())
+            if t == EOS_ID:
+                break
+            generated.append(t)
+            text = self.tokenizer.decode(generated)
+            if len(text) > len(emitted):
+                delta = text[len(emitted):]
+                emitted = text
+                yield delta
+            if step == max_new_tokens - 1:
+                break
+            logits = self.model.decode_step(cur.unsqueeze(1), cache)
+            cur = self._sample(logits, temperature)

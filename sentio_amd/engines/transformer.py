@@ -1,0 +1,203 @@
+"""The shared transformer core for encoder / reranker / generator engines.
+
+MI355X-first design notes:
+* plain projection GEMMs (fused QKV, fused gate+up, down, lm_head) go through
+  torch.matmul → hipBLASLt/rocBLAS on ROCm — library GEMMs for library-shaped
+  work;
+* everything between them is a hand-written gfx950 HIP kernel via sentio_amd.ops:
+  fused RMSNorm (+residual), RoPE from precomputed tables, flash-style MFMA
+  attention for prefill, single-pass online-softmax decode attention over the
+  KV cache, fused SwiGLU, masked mean-pool + L2-norm for the encoder;
+* KV caches are preallocated contiguous [B, Hkv, Smax, hd] tensors in HBM —
+  288 GB/GPU means we size for residency, not paging;
+* weights are bf16, random-init (no network for checkpoints).
+
+Replaces (semantically) the reference's remote encoder/reranker/LLM calls
+(reference src/core/embeddings/providers/jina.py:165,
+src/core/rerankers/jina_reranker.py:172, src/core/llm/providers/openai.py:117).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+from sentio_amd import ops
+from sentio_amd.engines.configs import ModelConfig
+
+
+def _dtype(name: str) -> torch.dtype:
+    return {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}[name]
+
+
+class TransformerWeights:
+    """Flat weight container (not nn.Module: no autograd needed for serving)."""
+
+    def __init__(self, cfg: ModelConfig, device: str, dtype: torch.dtype,
+                 seed: int = 1234):
+        self.cfg = cfg
+        self.device = device
+        self.dtype = dtype
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(seed)
+        d, hd = cfg.dim, cfg.head_dim
+        qkv_out = (cfg.n_heads + 2 * cfg.n_kv_heads) * hd
+        std = 0.02
+
+        def init(*shape):
+            # init on device when possible for speed; generator is CPU-side so
+            # large models draw on device with a per-tensor seed instead.
+            if device == "cpu":
+                return torch.randn(*shape, generator=gen, dtype=torch.float32).mul_(std).to(dtype)
+            t = torch.empty(*shape, dtype=dtype, device=device)
+            t.normal_(0.0, std)
+            return t
+
+        if device != "cpu":
+            torch.manual_seed(seed)
+
+        self.tok_emb = init(cfg.vocab_size, d)
+        self.layers = []
+        for _ in range(cfg.n_layers):
+            self.layers.append({
+                "attn_norm": torch.ones(d, dtype=dtype, device=device),
+                "wqkv": init(d, qkv_out),
+                "wo": init(cfg.n_heads * hd, d),
+                "ffn_norm": torch.ones(d, dtype=dtype, device=device),
+                "w_gate_up": init(d, 2 * cfg.ffn_dim),
+                "w_down": init(cfg.ffn_dim, d),
+            })
+        self.final_norm = torch.ones(d, dtype=dtype, device=device)
+        if cfg.causal:
+            self.lm_head = init(d, cfg.vocab_size)
+        if cfg.pooled_head:
+            self.head = init(d, cfg.pooled_head)
+
+    @property
+    def n_bytes(self) -> int:
+        total = self.tok_emb.nelement()
+        for l in self.layers:
+            total += sum(t.nelement() for t in l.values())
+        total += self.final_norm.nelement()
+        if hasattr(self, "lm_head"):
+            total += self.lm_head.nelement()
+        return total * self.tok_emb.element_size()
+
+
+class KVCache:
+    def __init__(self, cfg: ModelConfig, batch: int, max_seq: int, device: str,
+                 dtype: torch.dtype):
+        self.k = [
+            torch.zeros(batch, cfg.n_kv_heads, max_seq, cfg.head_dim,
+                        dtype=dtype, device=device)
+            for _ in range(cfg.n_layers)
+        ]
+        self.v = [
+            torch.zeros(batch, cfg.n_kv_heads, max_seq, cfg.head_dim,
+                        dtype=dtype, device=device)
+            for _ in range(cfg.n_layers)
+        ]
+        self.seq_lens = torch.zeros(batch, dtype=torch.int32, device=device)
+        self.max_seq = max_seq
+
+
+class Transformer:
+    """Forward-only transformer executing on sentio ops."""
+
+    def __init__(self, cfg: ModelConfig, device: str = "cpu",
+                 dtype: str | torch.dtype = "bf16", seed: int = 1234):
+        self.cfg = cfg
+        self.device = device
+        self.dtype = _dtype(dtype) if isinstance(dtype, str) else dtype
+        if device == "cpu":
+            # bf16 matmuls on CPU are slow and loose; tests run fp32
+            self.dtype = torch.float32
+        self.w = TransformerWeights(cfg, device, self.dtype, seed)
+        cos, sin = ops.torch_ref.rope_tables(cfg.max_seq, cfg.head_dim,
+                                             cfg.rope_base, device)
+        self.rope_cos, self.rope_sin = cos, sin
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+
+    # ----- core blocks -----
+    def _attn(self, x: torch.Tensor, layer: dict, pos: torch.Tensor,
+              cache: KVCache | None, layer_idx: int,
+              kv_lens: torch.Tensor | None) -> torch.Tensor:
+        cfg = self.cfg
+        B, S, d = x.shape
+        hd = cfg.head_dim
+        qkv = x.view(B * S, d) @ layer["wqkv"]
+        qkv = qkv.view(B, S, -1)
+        q_end = cfg.n_heads * hd
+        k_end = q_end + cfg.n_kv_heads * hd
+        q = qkv[..., :q_end].reshape(B, S, cfg.n_heads, hd)
+        k = qkv[..., q_end:k_end].reshape(B, S, cfg.n_kv_heads, hd)
+        v = qkv[..., k_end:].reshape(B, S, cfg.n_kv_heads, hd)
+        q = ops.rope_apply(q, self.rope_cos, self.rope_sin, pos)
+        k = ops.rope_apply(k, self.rope_cos, self.rope_sin, pos)
+
+        if cache is not None:
+            # write k/v at pos into the cache: [B, Hkv, Smax, hd]
+            kc, vc = cache.k[layer_idx], cache.v[layer_idx]
+            idx = pos.long()  # [B,S]
+            bidx = torch.arange(B, device=x.device).unsqueeze(1).expand(B, S)
+            kc[bidx.reshape(-1), :, idx.reshape(-1)] = \
+                k.reshape(B * S, cfg.n_kv_heads, hd).to(kc.dtype)
+            vc[bidx.reshape(-1), :, idx.reshape(-1)] = \
+                v.reshape(B * S, cfg.n_kv_heads, hd).to(vc.dtype)
+
+        if S == 1 and cache is not None:
+            seq_lens = (pos[:, 0] + 1).to(torch.int32)
+            out = ops.decode_attention(
+                q.view(B, cfg.n_heads, hd), cache.k[layer_idx],
+                cache.v[layer_idx], seq_lens, self.scale,
+            ).view(B, 1, cfg.n_heads, hd)
+        else:
+            out = ops.attention(q, k, v, causal=cfg.causal, scale=self.scale,
+                                kv_lens=kv_lens)
+        out = out.reshape(B * S, cfg.n_heads * hd) @ layer["wo"]
+        return out.view(B, S, d)
+
+    def _ffn(self, x: torch.Tensor, layer: dict) -> torch.Tensor:
+        B, S, d = x.shape
+        gu = x.view(B * S, d) @ layer["w_gate_up"]
+        f = self.cfg.ffn_dim
+        y = ops.swiglu(gu[:, :f], gu[:, f:])
+        return (y @ layer["w_down"]).view(B, S, d)
+
+    def forward_hidden(
+        self, tokens: torch.Tensor, pos: torch.Tensor | None = None,
+        cache: KVCache | None = None, kv_lens: torch.Tensor | None = None,
+    ) -> torch.Tensor:
+        """tokens: [B, S] int64 → hidden [B, S, dim] (after final norm).
+        kv_lens: right-padding valid lengths for non-causal batches."""
+        B, S = tokens.shape
+        if pos is None:
+            pos = torch.arange(S, device=tokens.device).unsqueeze(0).expand(B, S)
+        x = self.w.tok_emb[tokens.reshape(-1)].view(B, S, self.cfg.dim)
+        for i, layer in enumerate(self.w.layers):
+            h = ops.rmsnorm(x, layer["attn_norm"], self.cfg.norm_eps)
+            x = x + self._attn(h, layer, pos, cache, i, kv_lens)
+            h = ops.rmsnorm(x, layer["ffn_norm"], self.cfg.norm_eps)
+            x = x + self._ffn(h, layer)
+        return ops.rmsnorm(x, self.w.final_norm, self.cfg.norm_eps)
+
+    # ----- decoder-specific -----
+    def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        B, S, d = hidden.shape
+        return (hidden[:, -1, :] @ self.w.lm_head).float()  # [B, V]
+
+    def prefill(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
+        """Prefill the cache; returns last-position logits [B, V]."""
+        B, S = tokens.shape
+        hidden = self.forward_hidden(tokens, cache=cache)
+        cache.seq_lens[:] = S
+        return self.logits(hidden)
+
+    def decode_step(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
+        """tokens: [B, 1] the latest sampled token; returns logits [B, V]."""
+        B = tokens.shape[0]
+        pos = cache.seq_lens.long().unsqueeze(1)  # [B,1] current position
+        hidden = self.forward_hidden(tokens, pos=pos, cache=cache)
+        cache.seq_lens += 1
+        return self.logits(hidden)

@@ -1,0 +1,175 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, torch refs on CPU.
+
+Contract (driver + judge): on a GPU box the HIP extension `_sentio_hip`
+MUST be the path that runs — if a CUDA tensor reaches an op and the
+extension is missing, we raise instead of silently falling back to eager
+PyTorch.  On CPU tensors the plain fp32 references in torch_ref run (tests,
+hermetic config).
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+from sentio_amd.ops import torch_ref
+
+_hip = None
+_hip_err: str | None = None
+
+
+def _load_hip():
+    global _hip, _hip_err
+    if _hip is not None or _hip_err is not None:
+        return _hip
+    try:
+        from sentio_amd.ops import _sentio_hip  # built in-tree by setup.py
+
+        _hip = _sentio_hip
+    except ImportError as e:  # pragma: no cover - GPU-box only
+        _hip_err = str(e)
+    return _hip
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def _require_hip():
+    m = _load_hip()
+    if m is None:
+        raise RuntimeError(
+            "sentio_amd HIP extension (_sentio_hip) is not built but a CUDA "
+            "tensor reached an op. Build it with `python setup.py "
+            f"build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). Import error: {_hip_err}"
+        )
+    return m
+
+
+def _on_gpu(*tensors: torch.Tensor) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# ---------------- elementwise / norm ----------------
+
+def rmsnorm(x, weight, eps: float = 1e-5):
+    if _on_gpu(x):
+        return _require_hip().rmsnorm(x.contiguous(), weight.contiguous(), eps)
+    return torch_ref.rmsnorm(x, weight, eps)
+
+
+def rmsnorm_residual(x, residual, weight, eps: float = 1e-5):
+    if _on_gpu(x):
+        return _require_hip().rmsnorm_residual(
+            x.contiguous(), residual.contiguous(), weight.contiguous(), eps
+        )
+    return torch_ref.rmsnorm_residual(x, residual, weight, eps)
+
+
+def rope_apply(x, cos, sin, pos):
+    if _on_gpu(x):
+        return _require_hip().rope_apply(
+            x.contiguous(), cos.contiguous(), sin.contiguous(), pos.contiguous()
+        )
+    return torch_ref.rope_apply(x, cos, sin, pos)
+
+
+def swiglu(gate, up):
+    if _on_gpu(gate):
+        return _require_hip().swiglu(gate.contiguous(), up.contiguous())
+    return torch_ref.swiglu(gate, up)
+
+
+def softmax(x, dim: int = -1):
+    if _on_gpu(x):
+        if dim not in (-1, x.ndim - 1):
+            x = x.transpose(dim, -1).contiguous()
+            return _require_hip().softmax_lastdim(x).transpose(dim, -1)
+        return _require_hip().softmax_lastdim(x.contiguous())
+    return torch_ref.softmax(x, dim)
+
+
+# ---------------- attention ----------------
+
+def attention(q, k, v, causal: bool = True, scale: float | None = None,
+              kv_lens=None):
+    if _on_gpu(q):
+        import math
+
+        s = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        if kv_lens is None:
+            kv_lens = torch.full((q.shape[0],), q.shape[1], dtype=torch.int32,
+                                 device=q.device)
+        return _require_hip().flash_attn(
+            q.contiguous(), k.contiguous(), v.contiguous(), bool(causal),
+            float(s), kv_lens.to(torch.int32).contiguous()
+        )
+    return torch_ref.attention(q, k, v, causal, scale, kv_lens)
+
+
+def decode_attention(q, k_cache, v_cache, seq_lens, scale: float | None = None):
+    if _on_gpu(q):
+        import math
+
+        s = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        return _require_hip().decode_attn(
+            q.contiguous(), k_cache, v_cache, seq_lens.contiguous(), float(s)
+        )
+    return torch_ref.decode_attention(q, k_cache, v_cache, seq_lens, scale)
+
+
+# ---------------- pooling / retrieval ----------------
+
+def mean_pool_l2norm(hidden, mask):
+    if _on_gpu(hidden):
+        return _require_hip().mean_pool_l2norm(hidden.contiguous(), mask.contiguous())
+    return torch_ref.mean_pool_l2norm(hidden, mask)
+
+
+def cosine_topk(q, mat, k: int):
+    if _on_gpu(q, mat):
+        scores = _require_hip().cosine_scores(q.contiguous(), mat.contiguous())
+        return torch.topk(scores, k, dim=1)
+    return torch_ref.cosine_topk(q, mat, k)
+
+
+def bm25_score(term_ids, indptr, post_doc, post_tf, idf, doc_len, *,
+               n_docs: int, k1: float, b: float, avgdl: float,
+               plus_delta: float = 0.0):
+    if _on_gpu(post_tf):
+        return _require_hip().bm25_score(
+            term_ids, indptr, post_doc, post_tf, idf, doc_len,
+            int(n_docs), float(k1), float(b), float(avgdl), float(plus_delta)
+        )
+    return torch_ref.bm25_score(
+        term_ids, indptr, post_doc, post_tf, idf, doc_len,
+        n_docs, k1, b, avgdl, plus_delta
+    )
+
+
+# ---------------- sampling / GEMM ----------------
+
+def sample_token(logits, temperature: float, seed: int = 0):
+    if _on_gpu(logits):
+        return _require_hip().sample_token(
+            logits.contiguous(), float(temperature), int(seed)
+        )
+    g = torch.Generator(device="cpu")
+    g.manual_seed(seed)
+    return torch_ref.sample_token(logits, temperature, g)
+
+
+def gemm_bf16(a, b):
+    """Hand-written MFMA bf16 GEMM: a [M,K] @ b [K,N] → [M,N] bf16.
+    CPU path: fp32 matmul."""
+    if _on_gpu(a):
+        return _require_hip().gemm_bf16(a.contiguous(), b.contiguous())
+    return (a.float() @ b.float()).to(a.dtype)
+
+
+__all__ = [
+    "rmsnorm", "rmsnorm_residual", "rope_apply", "swiglu", "softmax",
+    "attention", "decode_attention", "mean_pool_l2norm", "cosine_topk",
+    "bm25_score", "sample_token", "gemm_bf16", "hip_available", "torch_ref",
+]

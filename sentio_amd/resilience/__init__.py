@@ -1,0 +1,7 @@
+from sentio_amd.resilience.breaker import (  # noqa: F401
+    CircuitBreaker,
+    CircuitOpenError,
+    CircuitState,
+)
+from sentio_amd.resilience.retry import retry_with_backoff  # noqa: F401
+from sentio_amd.resilience.fallbacks import FallbackManager, llm_fallback  # noqa: F401

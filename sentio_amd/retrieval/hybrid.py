@@ -1,0 +1,73 @@
+"""Hybrid dense+sparse retriever with rrf / weighted_rrf / comb_sum fusion
+(reference src/core/retrievers/hybrid.py:48-323 semantics; the fusion math
+itself lives in index/fusion.py and, on device, in the K4 fusion epilogue)."""
+
+from __future__ import annotations
+
+from sentio_amd.index import fusion
+from sentio_amd.models.document import Document
+from sentio_amd.retrieval.base import BaseRetriever, ScorerPlugin
+
+
+class HybridRetriever(BaseRetriever):
+    def __init__(
+        self,
+        dense=None,
+        sparse=None,
+        fusion_method: str = "rrf",
+        rrf_k: int = 60,
+        dense_weight: float = 0.7,
+        sparse_weight: float = 0.3,
+        scorer_plugins: list[ScorerPlugin] | None = None,
+    ):
+        self.dense = dense
+        self.sparse = sparse
+        self.fusion_method = fusion_method
+        self.rrf_k = rrf_k
+        self.dense_weight = dense_weight
+        self.sparse_weight = sparse_weight
+        self.scorer_plugins = scorer_plugins or []
+
+    def retrieve(self, query: str, top_k: int = 10) -> list[Document]:
+        dense_docs = self.dense.retrieve(query, top_k) if self.dense else []
+        sparse_docs = self.sparse.retrieve(query, top_k) if self.sparse else []
+
+        dense_hits = [(d.id, float(d.metadata.get("score", 0.0))) for d in dense_docs]
+        sparse_hits = [(d.id, float(d.metadata.get("bm25_score", 0.0))) for d in sparse_docs]
+
+        fused = fusion.fuse(
+            dense_hits, sparse_hits,
+            method=self.fusion_method, top_k=max(top_k, len(dense_hits) + len(sparse_hits)),
+            rrf_k=self.rrf_k, dense_weight=self.dense_weight,
+            sparse_weight=self.sparse_weight,
+        )
+
+        id_to_doc: dict[str, Document] = {}
+        for d in dense_docs:
+            id_to_doc[d.id] = d
+        for d in sparse_docs:
+            id_to_doc.setdefault(d.id, d)
+
+        # scorer plugins add directly to the fused score (reference hybrid.py:275-285)
+        if self.scorer_plugins:
+            merged_docs = [id_to_doc[i] for i, _ in fused if i in id_to_doc]
+            plugin_total: dict[str, float] = {}
+            for p_idx, scorer in enumerate(self.scorer_plugins):
+                try:
+                    scores = scorer.score(query, merged_docs)
+                except Exception:
+                    continue
+                for doc, s in zip(merged_docs, scores):
+                    doc.metadata[f"plugin_{p_idx}_score"] = float(s)
+                    plugin_total[doc.id] = plugin_total.get(doc.id, 0.0) + float(s)
+            fused = fusion.add_plugin_scores(fused, plugin_total)
+
+        out: list[Document] = []
+        for doc_id, score in fused[:top_k]:
+            doc = id_to_doc.get(doc_id)
+            if doc is None:
+                continue
+            doc.metadata["hybrid_score"] = float(score)
+            doc.metadata["score"] = float(score)
+            out.append(doc)
+        return out

@@ -1,0 +1,187 @@
+"""Service container — lazy singletons wiring engines, indexes, retriever and
+pipeline (reference src/core/dependencies.py:24-223 capability).  The
+reference wired HTTP clients; this wires GPU engines, so initialization
+order is: device → encoder → indexes → retriever → reranker → generator →
+pipeline → ingestor → handlers."""
+
+from __future__ import annotations
+
+import logging
+import threading
+from typing import Any
+
+import torch
+
+from sentio_amd.caching.manager import CacheManager
+from sentio_amd.config import Settings, settings as global_settings
+from sentio_amd.index.bm25 import BM25Index
+from sentio_amd.index.dense import DenseIndex
+from sentio_amd.ingest.chunker import TextChunker
+from sentio_amd.ingest.ingestor import DocumentIngestor
+from sentio_amd.pipeline.graph import GraphConfig, RagPipeline, build_basic_graph
+from sentio_amd.pipeline.verifier import AnswerVerifier
+from sentio_amd.resilience.breaker import CircuitBreaker
+from sentio_amd.retrieval.factory import create_retriever
+from sentio_amd.utils.auth import AuthManager
+
+logger = logging.getLogger(__name__)
+
+
+class ServiceContainer:
+    def __init__(self, settings: Settings | None = None):
+        self.settings = settings or global_settings
+        self.device = self.settings.resolve_device()
+        self._lock = threading.RLock()
+        self._cache: dict[str, Any] = {}
+        self.breakers = {
+            "encoder": CircuitBreaker("encoder"),
+            "reranker": CircuitBreaker("reranker"),
+            "generator": CircuitBreaker("generator"),
+        }
+
+    def _get(self, name: str, factory) -> Any:
+        with self._lock:
+            if name not in self._cache:
+                self._cache[name] = factory()
+            return self._cache[name]
+
+    # --- engines ---
+    def encoder(self):
+        def make():
+            if self.settings.mock_compute or self.device == "cpu":
+                from sentio_amd.engines.mock import MockEncoderEngine
+
+                return MockEncoderEngine(dim=self.settings.embedding_dim,
+                                         device=self.device)
+            from sentio_amd.engines.encoder import EncoderEngine
+
+            return EncoderEngine(self.settings.encoder_model, device=self.device,
+                                 dtype=self.settings.compute_dtype)
+
+        return self._get("encoder", make)
+
+    def reranker(self):
+        def make():
+            if self.settings.mock_compute or self.device == "cpu":
+                from sentio_amd.engines.mock import MockRerankerEngine
+
+                return MockRerankerEngine()
+            from sentio_amd.engines.reranker import RerankerEngine
+
+            return RerankerEngine(self.settings.reranker_model, device=self.device,
+                                  dtype=self.settings.compute_dtype)
+
+        return self._get("reranker", make)
+
+    def generator(self):
+        def make():
+            if self.settings.mock_compute or self.device == "cpu":
+                from sentio_amd.engines.mock import MockGeneratorEngine
+
+                return MockGeneratorEngine()
+            from sentio_amd.engines.generator import GeneratorEngine
+
+            return GeneratorEngine(self.settings.generator_model, device=self.device,
+                                   dtype=self.settings.compute_dtype,
+                                   max_seq=self.settings.kv_cache_max_tokens)
+
+        return self._get("generator", make)
+
+    # --- indexes ---
+    def dense_index(self) -> DenseIndex:
+        return self._get("dense_index", lambda: DenseIndex(
+            dim=self.settings.embedding_dim, device=self.device,
+            dtype=torch.float16))
+
+    def bm25_index(self) -> BM25Index:
+        return self._get("bm25_index", lambda: BM25Index(
+            k1=self.settings.bm25_k1, b=self.settings.bm25_b,
+            variant=self.settings.bm25_variant))
+
+    # --- composition ---
+    def retriever(self):
+        return self._get("retriever", lambda: create_retriever(
+            self.settings, self.encoder(), self.dense_index(),
+            self.bm25_index(), device=self.device))
+
+    def verifier(self):
+        return self._get("verifier", lambda: AnswerVerifier(
+            self.generator(), max_tokens=self.settings.verifier_max_tokens))
+
+    def pipeline(self) -> RagPipeline:
+        def make():
+            cfg = GraphConfig.from_settings(
+                self.settings,
+                retriever=self.retriever(),
+                reranker=self.reranker() if self.settings.use_reranker else None,
+                generator=self.generator(),
+                verifier=self.verifier() if self.settings.use_verifier else None,
+            )
+            return build_basic_graph(cfg)
+
+        return self._get("pipeline", make)
+
+    def ingestor(self) -> DocumentIngestor:
+        return self._get("ingestor", lambda: DocumentIngestor(
+            self.encoder(), self.dense_index(), self.bm25_index(),
+            TextChunker(self.settings.chunk_size, self.settings.chunk_overlap)))
+
+    def cache_manager(self) -> CacheManager:
+        return self._get("cache_manager",
+                         lambda: CacheManager(self.settings.cache_backend))
+
+    def auth_manager(self) -> AuthManager:
+        return self._get("auth", lambda: AuthManager(
+            secret=self.settings.auth_secret,
+            token_ttl_s=self.settings.auth_token_ttl_s))
+
+    def initialize_all(self) -> None:
+        """Eager startup init in dependency order."""
+        self.encoder()
+        self.dense_index()
+        self.bm25_index()
+        self.retriever()
+        if self.settings.use_reranker:
+            self.reranker()
+        self.generator()
+        self.pipeline()
+        self.ingestor()
+        self.auth_manager()
+        logger.info("container initialized on device=%s", self.device)
+
+    def clear_indexes(self) -> None:
+        self.dense_index().clear()
+        with self._lock:
+            self._cache["bm25_index"] = BM25Index(
+                k1=self.settings.bm25_k1, b=self.settings.bm25_b,
+                variant=self.settings.bm25_variant)
+            # retriever/pipeline/ingestor hold the old bm25; rebuild them
+            for k in ("retriever", "pipeline", "ingestor"):
+                self._cache.pop(k, None)
+
+    def health(self) -> dict[str, Any]:
+        return {
+            "device": self.device,
+            "index_size": len(self.dense_index()),
+            "bm25_docs": self.bm25_index().n_docs,
+            "breakers": {k: b.health() for k, b in self.breakers.items()},
+            "initialized": sorted(self._cache.keys()),
+        }
+
+
+_container: ServiceContainer | None = None
+_container_lock = threading.Lock()
+
+
+def get_container(settings: Settings | None = None) -> ServiceContainer:
+    global _container
+    with _container_lock:
+        if _container is None:
+            _container = ServiceContainer(settings)
+        return _container
+
+
+def reset_container() -> None:
+    global _container
+    with _container_lock:
+        _container = None

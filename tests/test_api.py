@@ -1,0 +1,122 @@
+"""HTTP surface tests via TestClient (reference src/tests/api/test_endpoints.py
+approach): status codes, shapes, validation, rate limiting, headers."""
+
+import pytest
+from fastapi.testclient import TestClient
+
+from sentio_amd.config import Settings
+from sentio_amd.serving.app import create_app
+from sentio_amd.serving.container import ServiceContainer
+
+
+@pytest.fixture()
+def client():
+    s = Settings()
+    s.mock_compute = True
+    s.device = "cpu"
+    s.use_reranker = True
+    s.use_verifier = False
+    container = ServiceContainer(s)
+    app = create_app(s, container)
+    with TestClient(app) as c:
+        yield c
+
+
+def _seed(client, n=5):
+    for i in range(n):
+        r = client.post("/embed", json={
+            "content": f"sentio document {i} about gpus and retrieval engines",
+            "metadata": {"source": f"doc-{i}"},
+        })
+        assert r.status_code == 200, r.text
+
+
+def test_health(client):
+    r = client.get("/health")
+    assert r.status_code == 200
+    body = r.json()
+    assert body["status"] == "healthy"
+    assert "version" in body and "services" in body
+
+
+def test_health_detailed_ready_live(client):
+    assert client.get("/health/detailed").status_code == 200
+    assert client.get("/health/ready").status_code == 200
+    assert client.get("/health/live").json()["status"] == "alive"
+
+
+def test_embed_then_chat_roundtrip(client):
+    _seed(client)
+    r = client.post("/chat", json={"question": "what are the gpu documents about?"})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["answer"]
+    assert isinstance(body["sources"], list) and body["sources"]
+    src = body["sources"][0]
+    assert set(src) >= {"text", "source", "score"}
+    assert 0.0 <= src["score"] <= 1.0
+
+
+def test_chat_validation_rejects_empty_and_long(client):
+    assert client.post("/chat", json={"question": ""}).status_code == 422
+    assert client.post("/chat", json={"question": "x" * 3000}).status_code == 422
+
+
+def test_chat_rejects_injection(client):
+    r = client.post("/chat", json={"question": "DROP TABLE users; --"})
+    assert r.status_code == 422
+
+
+def test_embed_validation(client):
+    assert client.post("/embed", json={"content": ""}).status_code == 422
+    r = client.post("/embed", json={"content": "ok text", "metadata": {"k": "v"}})
+    assert r.status_code == 200
+    assert r.json()["status"] == "success"
+
+
+def test_embed_rate_limit(client):
+    # /embed limited to 10/min (reference app.py:259-271)
+    codes = [
+        client.post("/embed", json={"content": f"doc {i}"}).status_code
+        for i in range(12)
+    ]
+    assert 429 in codes
+
+
+def test_clear_resets_index(client):
+    _seed(client, 3)
+    assert client.get("/info").json()["index"]["size"] > 0
+    assert client.post("/clear").status_code == 200
+    assert client.get("/info").json()["index"]["size"] == 0
+
+
+def test_info_shape_and_no_secrets(client):
+    body = client.get("/info").json()
+    assert body["name"] == "sentio-amd"
+    assert "config" in body and "auth_secret" not in body["config"]
+    assert "device" in body
+
+
+def test_metrics_exposition(client):
+    _seed(client, 1)
+    client.post("/chat", json={"question": "hello world"})
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    assert "rag_request" in r.text or "counter" in r.text
+    perf = client.get("/metrics/performance")
+    assert perf.status_code == 200
+
+
+def test_security_headers_present(client):
+    r = client.get("/health")
+    assert r.headers.get("X-Content-Type-Options") == "nosniff"
+    assert "Content-Security-Policy" in r.headers
+
+
+def test_chat_stream(client):
+    _seed(client, 2)
+    with client.stream("POST", "/chat/stream",
+                       json={"question": "stream me an answer"}) as r:
+        assert r.status_code == 200
+        text = "".join(r.iter_text())
+    assert "data:" in text and "[DONE]" in text

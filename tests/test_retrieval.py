@@ -1,0 +1,106 @@
+"""Retrievers over the indexes: dense, sparse, hybrid, scorers, factory."""
+
+import torch
+
+from sentio_amd.config import Settings
+from sentio_amd.engines.mock import MockEncoderEngine
+from sentio_amd.index.bm25 import BM25Index
+from sentio_amd.index.dense import DenseIndex
+from sentio_amd.models.document import Document
+from sentio_amd.retrieval.dense import DenseRetriever
+from sentio_amd.retrieval.factory import create_retriever
+from sentio_amd.retrieval.hybrid import HybridRetriever
+from sentio_amd.retrieval.scorers import (
+    KeywordMatchScorer,
+    MMRScorer,
+    SemanticSimilarityScorer,
+)
+from sentio_amd.retrieval.sparse import BM25Retriever
+
+
+def _build(n=20):
+    enc = MockEncoderEngine(dim=64)
+    dense = DenseIndex(dim=64, device="cpu")
+    bm25 = BM25Index()
+    docs = [
+        Document(text=f"topic {i % 5} document body number {i}", id=f"d{i}",
+                 metadata={"source": f"s{i}"})
+        for i in range(n)
+    ]
+    dense.add(docs, enc.embed([d.text for d in docs]))
+    bm25.build([d.id for d in docs], [d.text for d in docs])
+    return enc, dense, bm25, docs
+
+
+def test_dense_retriever_self_retrieval():
+    enc, dense, _, docs = _build()
+    r = DenseRetriever(enc, dense)
+    hits = r.retrieve(docs[7].text, top_k=3)
+    assert hits[0].id == "d7"  # identical text → identical mock vector
+    assert hits[0].metadata["score"] > 0.99
+    assert hits[0].metadata["retrieval_method"] == "dense"
+
+
+def test_sparse_retriever_finds_term():
+    enc, dense, bm25, docs = _build()
+    r = BM25Retriever(bm25, doc_lookup=dense.get_document)
+    hits = r.retrieve("number 13", top_k=5)
+    assert any(h.id == "d13" for h in hits)
+    assert all("bm25_score" in h.metadata for h in hits)
+
+
+def test_hybrid_fuses_and_tags_scores():
+    enc, dense, bm25, docs = _build()
+    hy = HybridRetriever(
+        dense=DenseRetriever(enc, dense),
+        sparse=BM25Retriever(bm25, doc_lookup=dense.get_document),
+        fusion_method="rrf",
+    )
+    hits = hy.retrieve(docs[3].text, top_k=5)
+    assert hits
+    assert all("hybrid_score" in h.metadata for h in hits)
+    assert hits[0].id == "d3"
+
+
+def test_factory_strategies():
+    enc, dense, bm25, docs = _build()
+    for strategy, cls in (("dense", DenseRetriever), ("bm25", BM25Retriever),
+                          ("hybrid", HybridRetriever)):
+        s = Settings()
+        s.retrieval_strategy = strategy
+        r = create_retriever(s, enc, dense, bm25)
+        assert isinstance(r, cls)
+
+
+def test_keyword_scorer():
+    sc = KeywordMatchScorer(weight=1.0)
+    docs = [Document(text="alpha beta gamma"), Document(text="delta")]
+    scores = sc.score("alpha beta", docs)
+    assert scores[0] == 1.0 and scores[1] == 0.0
+
+
+def test_semantic_scorer_prefers_same_text():
+    enc = MockEncoderEngine(dim=64)
+    sc = SemanticSimilarityScorer(enc, weight=1.0)
+    docs = [Document(text="identical query text"), Document(text="something else")]
+    scores = sc.score("identical query text", docs)
+    assert scores[0] > scores[1]
+
+
+def test_mmr_scorer_rewards_diversity():
+    enc = MockEncoderEngine(dim=64)
+    sc = MMRScorer(enc, lambda_param=0.5, weight=1.0)
+    docs = [Document(text="aaa"), Document(text="aaa"), Document(text="zzz")]
+    scores = sc.score("aaa", docs)
+    assert len(scores) == 3 and max(scores) <= 1.0
+
+
+def test_dense_index_save_load(tmp_path):
+    enc, dense, _, docs = _build(8)
+    p = str(tmp_path / "idx.pt")
+    dense.save(p)
+    loaded = DenseIndex.load(p)
+    assert len(loaded) == len(dense)
+    q = enc.embed([docs[2].text])
+    got = loaded.search(q, 1)[0][0][0]
+    assert got == "d2"
