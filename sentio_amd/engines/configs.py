@@ -60,6 +60,10 @@ MODEL_CONFIGS: dict[str, ModelConfig] = {
         name="tiny-decoder", dim=64, n_layers=2, n_heads=4, n_kv_heads=2,
         ffn_dim=128, vocab_size=512, max_seq=1024, rope_base=10000.0,
     ),
+    "tiny-decoder64": ModelConfig(  # head_dim 64 — GPU-kernel-compatible tiny
+        name="tiny-decoder64", dim=256, n_layers=2, n_heads=4, n_kv_heads=2,
+        ffn_dim=512, vocab_size=512, max_seq=1024, rope_base=10000.0,
+    ),
     # --- encoders (jina-v3 class: 1024-dim output, reference jina.py:23-27) ---
     "sentio-encoder-base": ModelConfig(
         name="sentio-encoder-base", dim=1024, n_layers=24, n_heads=16,

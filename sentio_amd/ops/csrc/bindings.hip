@@ -1,0 +1,238 @@
+// Torch bindings for the sentio gfx950 kernels.  The only TU that includes
+// torch headers (slow compile); kernels live in sibling .hip TUs exposed
+// through the C API in sentio_kernels.h.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+hipError_t sentio_rmsnorm(const void*, const void*, void*, long, int, float,
+                          hipStream_t);
+hipError_t sentio_rmsnorm_residual(const void*, const void*, const void*,
+                                   void*, void*, long, int, float, hipStream_t);
+hipError_t sentio_swiglu(const void*, const void*, void*, long, hipStream_t);
+hipError_t sentio_rope(const void*, void*, const float*, const float*,
+                       const int*, int, int, int, int, hipStream_t);
+hipError_t sentio_softmax(const void*, void*, long, int, hipStream_t);
+hipError_t sentio_mean_pool_l2norm(const void*, const unsigned char*, float*,
+                                   int, int, int, hipStream_t);
+hipError_t sentio_sample(const float*, long*, int, int, float, unsigned,
+                         hipStream_t);
+hipError_t sentio_cosine_scores_f16(const void*, const void*, float*, long,
+                                    int, int, hipStream_t);
+hipError_t sentio_cosine_scores_bf16(const void*, const void*, float*, long,
+                                     int, int, hipStream_t);
+hipError_t sentio_bm25(const long*, const long*, const long*, const int*,
+                       const float*, const float*, const float*, float*, int,
+                       long, float, float, float, float, hipStream_t);
+hipError_t sentio_flash_attn(const void*, const void*, const void*, void*,
+                             const int*, int, int, int, int, int, float, int,
+                             hipStream_t);
+hipError_t sentio_decode_attn(const void*, const void*, const void*, void*,
+                              const int*, int, int, int, int, int, float,
+                              hipStream_t);
+hipError_t sentio_gemm_bf16(const void*, const void*, void*, int, int, int,
+                            hipStream_t);
+}
+
+namespace {
+
+void check_hip(hipError_t e, const char* what) {
+  TORCH_CHECK(e == hipSuccess, "sentio HIP kernel '", what,
+              "' failed: ", hipGetErrorString(e));
+}
+
+hipStream_t stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16_cuda(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+  check_bf16_cuda(x, "x");
+  check_bf16_cuda(w, "w");
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  check_hip(sentio_rmsnorm(x.data_ptr(), w.data_ptr(), y.data_ptr(), rows, D,
+                           (float)eps, stream()), "rmsnorm");
+  return y;
+}
+
+std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res,
+                                            torch::Tensor w, double eps) {
+  check_bf16_cuda(x, "x");
+  const int D = x.size(-1);
+  const long rows = x.numel() / D;
+  auto y = torch::empty_like(x);
+  auto h = torch::empty_like(x);
+  check_hip(sentio_rmsnorm_residual(x.data_ptr(), res.data_ptr(), w.data_ptr(),
+                                    y.data_ptr(), h.data_ptr(), rows, D,
+                                    (float)eps, stream()), "rmsnorm_residual");
+  return {y, h};
+}
+
+torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
+  check_bf16_cuda(gate, "gate");
+  auto out = torch::empty_like(gate);
+  check_hip(sentio_swiglu(gate.data_ptr(), up.data_ptr(), out.data_ptr(),
+                          gate.numel(), stream()), "swiglu");
+  return out;
+}
+
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosT, torch::Tensor sinT,
+                         torch::Tensor pos) {
+  check_bf16_cuda(x, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be [B,S,H,D]");
+  TORCH_CHECK(cosT.scalar_type() == torch::kFloat, "cos table must be f32");
+  auto pos_i = pos.to(torch::kInt).contiguous();
+  auto y = torch::empty_like(x);
+  check_hip(sentio_rope(x.data_ptr(), y.data_ptr(),
+                        cosT.data_ptr<float>(), sinT.data_ptr<float>(),
+                        pos_i.data_ptr<int>(), x.size(0), x.size(1), x.size(2),
+                        x.size(3), stream()), "rope");
+  return y;
+}
+
+torch::Tensor softmax_lastdim(torch::Tensor x) {
+  check_bf16_cuda(x, "x");
+  const int D = x.size(-1);
+  auto y = torch::empty_like(x);
+  check_hip(sentio_softmax(x.data_ptr(), y.data_ptr(), x.numel() / D, D,
+                           stream()), "softmax");
+  return y;
+}
+
+torch::Tensor mean_pool_l2norm(torch::Tensor hidden, torch::Tensor mask) {
+  check_bf16_cuda(hidden, "hidden");
+  TORCH_CHECK(hidden.dim() == 3, "hidden must be [B,S,D]");
+  auto m = mask.to(torch::kUInt8).contiguous();
+  auto out = torch::empty({hidden.size(0), hidden.size(2)},
+                          hidden.options().dtype(torch::kFloat));
+  check_hip(sentio_mean_pool_l2norm(hidden.data_ptr(), m.data_ptr<uint8_t>(),
+                                    out.data_ptr<float>(), hidden.size(0),
+                                    hidden.size(1), hidden.size(2), stream()),
+            "mean_pool_l2norm");
+  return out;
+}
+
+torch::Tensor sample_token(torch::Tensor logits, double temperature,
+                           int64_t seed) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2, "logits must be [B,V] GPU");
+  auto l = logits.to(torch::kFloat).contiguous();
+  auto out = torch::empty({l.size(0)}, l.options().dtype(torch::kLong));
+  check_hip(sentio_sample(l.data_ptr<float>(), out.data_ptr<int64_t>(),
+                          l.size(0), l.size(1), (float)temperature,
+                          (unsigned)seed, stream()), "sample");
+  return out;
+}
+
+torch::Tensor cosine_scores(torch::Tensor q, torch::Tensor mat) {
+  TORCH_CHECK(q.is_cuda() && mat.is_cuda(), "must be on GPU");
+  TORCH_CHECK(q.scalar_type() == mat.scalar_type(), "q/mat dtype mismatch");
+  const long N = mat.size(0);
+  const int D = mat.size(1);
+  const int B = q.size(0);
+  auto scores = torch::empty({B, N}, q.options().dtype(torch::kFloat));
+  hipError_t e;
+  if (q.scalar_type() == torch::kHalf)
+    e = sentio_cosine_scores_f16(mat.data_ptr(), q.data_ptr(),
+                                 scores.data_ptr<float>(), N, D, B, stream());
+  else if (q.scalar_type() == torch::kBFloat16)
+    e = sentio_cosine_scores_bf16(mat.data_ptr(), q.data_ptr(),
+                                  scores.data_ptr<float>(), N, D, B, stream());
+  else
+    TORCH_CHECK(false, "cosine_scores expects fp16/bf16");
+  check_hip(e, "cosine_scores");
+  return scores;
+}
+
+torch::Tensor bm25_score(torch::Tensor term_ids, torch::Tensor indptr,
+                         torch::Tensor post_doc, torch::Tensor post_tf,
+                         torch::Tensor idf, torch::Tensor doc_len,
+                         int64_t n_docs, double k1, double b, double avgdl,
+                         double plus_delta) {
+  auto scores = torch::zeros({n_docs}, post_tf.options().dtype(torch::kFloat));
+  const int T = term_ids.size(0);
+  if (T == 0) return scores;
+  auto tids = term_ids.to(torch::kLong).contiguous();
+  auto starts = indptr.index({tids});
+  auto ends = indptr.index({tids + 1});
+  auto lens = ends - starts;
+  auto qoff = torch::zeros({T + 1}, tids.options());
+  qoff.index_put_({torch::indexing::Slice(1, T + 1)}, torch::cumsum(lens, 0));
+  const long total = qoff[-1].item<long>();
+  if (total == 0) return scores;
+  check_hip(sentio_bm25(tids.data_ptr<int64_t>(), qoff.data_ptr<int64_t>(),
+                        starts.contiguous().data_ptr<int64_t>(),
+                        post_doc.data_ptr<int>(), post_tf.data_ptr<float>(),
+                        idf.data_ptr<float>(), doc_len.data_ptr<float>(),
+                        scores.data_ptr<float>(), T, total, (float)k1,
+                        (float)b, (float)avgdl, (float)plus_delta, stream()),
+            "bm25");
+  return scores;
+}
+
+torch::Tensor flash_attn(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                         bool causal, double scale, torch::Tensor kv_lens) {
+  check_bf16_cuda(q, "q");
+  check_bf16_cuda(k, "k");
+  check_bf16_cuda(v, "v");
+  TORCH_CHECK(q.dim() == 4, "q must be [B,S,H,D]");
+  const int B = q.size(0), S = q.size(1), H = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  auto out = torch::empty_like(q);
+  auto kl = kv_lens.to(torch::kInt).contiguous();
+  check_hip(sentio_flash_attn(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                              out.data_ptr(), kl.data_ptr<int>(), B, S, H,
+                              Hkv, D, (float)scale, causal ? 1 : 0, stream()),
+            "flash_attn");
+  return out;
+}
+
+torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                          torch::Tensor seq_lens, double scale) {
+  check_bf16_cuda(q, "q");
+  TORCH_CHECK(q.dim() == 3, "q must be [B,H,D]");
+  TORCH_CHECK(kc.dim() == 4, "k cache must be [B,Hkv,Smax,D]");
+  const int B = q.size(0), H = q.size(1), D = q.size(2);
+  const int Hkv = kc.size(1), Smax = kc.size(2);
+  auto out = torch::empty_like(q);
+  auto sl = seq_lens.to(torch::kInt).contiguous();
+  check_hip(sentio_decode_attn(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                               out.data_ptr(), sl.data_ptr<int>(), B, H, Hkv,
+                               Smax, D, (float)scale, stream()), "decode_attn");
+  return out;
+}
+
+torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b) {
+  check_bf16_cuda(a, "a");
+  check_bf16_cuda(b, "b");
+  const int M = a.size(0), K = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == K, "inner dims mismatch");
+  auto c = torch::empty({M, N}, a.options());
+  check_hip(sentio_gemm_bf16(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K,
+                             stream()), "gemm_bf16");
+  return c;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm);
+  m.def("rmsnorm_residual", &rmsnorm_residual);
+  m.def("swiglu", &swiglu);
+  m.def("rope_apply", &rope_apply);
+  m.def("softmax_lastdim", &softmax_lastdim);
+  m.def("mean_pool_l2norm", &mean_pool_l2norm);
+  m.def("sample_token", &sample_token);
+  m.def("cosine_scores", &cosine_scores);
+  m.def("bm25_score", &bm25_score);
+  m.def("flash_attn", &flash_attn);
+  m.def("decode_attn", &decode_attn);
+  m.def("gemm_bf16", &gemm_bf16);
+}

@@ -1,0 +1,134 @@
+// Retrieval kernels: fused cosine scoring scan over the in-HBM index (K2)
+// and BM25 CSR-postings scoring (K3).  Memory-bandwidth-bound by design:
+// the cosine scan reads each index row exactly once and amortizes it over
+// all B queries of the batch (queries staged in LDS).
+#include "common.h"
+
+// mat [N, D] f16-or-bf16 row-major (L2-normalized rows), q [B, D] same dtype
+// (normalized), scores [B, N] f32.
+// Block = 4 waves; queries staged once in LDS; each wave scans rows
+// grid-strided, lane l owns elements [l*E, l*E+E) of a row (E = D/64).
+// Wave reduces B dots per row via shfl.
+// E = D/64 is a template constant so the row cache rv[] stays in registers
+// (runtime-indexed ext arrays spill to scratch — guide §5.4 rule 20).
+template <typename T, int E>
+__global__ void cosine_scores_kernel(const T* __restrict__ mat,
+                                     const T* __restrict__ q,
+                                     float* __restrict__ scores,
+                                     long N, int B) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* qs = reinterpret_cast<float*>(smem);  // [B][D] f32
+  constexpr int D = E * WAVE;
+
+  // cooperative query staging
+  for (int i = threadIdx.x; i < B * D; i += blockDim.x)
+    qs[i] = (float)q[i];
+  __syncthreads();
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const long wave_global = (long)blockIdx.x * waves_per_block + wid;
+  const long wave_count = (long)gridDim.x * waves_per_block;
+
+  for (long row = wave_global; row < N; row += wave_count) {
+    const T* r = mat + row * (long)D + lane * E;
+    float rv[E];
+#pragma unroll
+    for (int e = 0; e < E; ++e) rv[e] = (float)r[e];
+    for (int b = 0; b < B; ++b) {
+      const float* qb = qs + (long)b * D + lane * E;
+      float acc = 0.f;
+#pragma unroll
+      for (int e = 0; e < E; ++e) acc += rv[e] * qb[e];
+      acc = wave_sum(acc);
+      if (lane == 0) scores[(long)b * N + row] = acc;
+    }
+  }
+}
+
+template <typename T>
+hipError_t launch_cosine(const void* mat, const void* q, float* scores, long N,
+                         int D, int B, hipStream_t stream) {
+  if (D % WAVE != 0) return hipErrorInvalidValue;
+  size_t lds = (size_t)B * D * sizeof(float);
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  long blocks = (N + 3) / 4;
+  if (blocks > 4096) blocks = 4096;
+  dim3 g((unsigned)blocks), blk(256);
+  switch (D / WAVE) {
+#define CASE_E(EV)                                                         \
+  case EV:                                                                 \
+    hipLaunchKernelGGL((cosine_scores_kernel<T, EV>), g, blk, lds, stream, \
+                       (const T*)mat, (const T*)q, scores, N, B);          \
+    break;
+    CASE_E(1) CASE_E(2) CASE_E(4) CASE_E(6) CASE_E(8)
+    CASE_E(12) CASE_E(16) CASE_E(24) CASE_E(32)
+#undef CASE_E
+    default:
+      return hipErrorInvalidValue;
+  }
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------- BM25
+// Grid-stride over the query's postings union.  For posting j of term t:
+//   contrib = idf[t] * (tf*(k1+1)/(tf + k1*(1-b+b*dl/avgdl)) + delta)
+// atomically accumulated into scores[doc].  T (query terms) is small, so a
+// linear scan finds the owning term.
+__global__ void bm25_kernel(const long* __restrict__ term_ids,
+                            const long* __restrict__ qoff,   // [T+1] prefix
+                            const long* __restrict__ starts, // [T] indptr[t]
+                            const int* __restrict__ post_doc,
+                            const float* __restrict__ post_tf,
+                            const float* __restrict__ idf,
+                            const float* __restrict__ doc_len,
+                            float* __restrict__ scores,
+                            int T, long total, float k1, float b,
+                            float inv_avgdl, float delta) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int t = 0;
+    while (t + 1 < T && i >= qoff[t + 1]) ++t;
+    const long p = starts[t] + (i - qoff[t]);
+    const int doc = post_doc[p];
+    const float tf = post_tf[p];
+    const float denom = tf + k1 * (1.f - b + b * doc_len[doc] * inv_avgdl);
+    const float contrib =
+        idf[term_ids[t]] * (tf * (k1 + 1.f) / denom + delta);
+    atomicAdd(&scores[doc], contrib);
+  }
+}
+
+extern "C" {
+
+hipError_t sentio_cosine_scores_f16(const void* mat, const void* q,
+                                    float* scores, long N, int D, int B,
+                                    hipStream_t stream) {
+  return launch_cosine<__half>(mat, q, scores, N, D, B, stream);
+}
+
+hipError_t sentio_cosine_scores_bf16(const void* mat, const void* q,
+                                     float* scores, long N, int D, int B,
+                                     hipStream_t stream) {
+  return launch_cosine<__hip_bfloat16>(mat, q, scores, N, D, B, stream);
+}
+
+hipError_t sentio_bm25(const long* term_ids, const long* qoff,
+                       const long* starts, const int* post_doc,
+                       const float* post_tf, const float* idf,
+                       const float* doc_len, float* scores, int T, long total,
+                       float k1, float b, float avgdl, float delta,
+                       hipStream_t stream) {
+  if (total == 0) return hipSuccess;
+  long blocks = (total + 255) / 256;
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(bm25_kernel, dim3((unsigned)blocks), dim3(256), 0, stream,
+                     term_ids, qoff, starts, post_doc, post_tf, idf, doc_len,
+                     scores, T, total, k1, b, 1.f / avgdl, delta);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+}  // extern "C"
