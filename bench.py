@@ -1,0 +1,389 @@
+"""Flagship benchmark: end-to-end /chat serving steps on N GPUs of one node.
+
+Measures BASELINE.json's north-star metric — end-to-end /chat QPS (+p50) on
+a hybrid-RAG pipeline with a Llama-3-8B-class generator — on synthetic
+corpora and random-init weights (no network for datasets/checkpoints).
+
+One step = one batch of `--batch` concurrent /chat requests per rank through
+the full pipeline: query embed (encoder engine) → sharded dense cosine top-k
+(RCCL all-gather merge) → GPU BM25 → RRF fusion → cross-encoder rerank →
+selection → batched generation (prefill + `--gen-tokens` KV-cache decode
+steps) [→ optional verifier].  Weak scaling: each GPU holds a fixed
+`--docs-per-gpu` corpus shard and serves its own request batch; value is the
+whole-job aggregate QPS over all ranks.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched under torch.distributed.run with one rank per GPU over RCCL.
+Rank 0 prints ONE JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+import torch
+
+from sentio_amd.index.bm25 import BM25Index
+from sentio_amd.index.dense import DenseIndex
+from sentio_amd.index import fusion
+from sentio_amd.models.document import Document
+from sentio_amd.parallel import dist as D
+from sentio_amd.parallel.shard import ShardedIndex
+from sentio_amd.pipeline.context import prepare_context
+from sentio_amd.pipeline.prompt_builder import PromptBuilder
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--batch", type=int, default=32,
+                   help="concurrent /chat requests per rank per step")
+    p.add_argument("--docs-per-gpu", type=int, default=1_250_000,
+                   help="synthetic corpus shard per GPU (8 GPUs -> 10M total)")
+    p.add_argument("--gen-tokens", type=int, default=128)
+    p.add_argument("--model", type=str, default="llama3-8b")
+    p.add_argument("--encoder", type=str, default="sentio-encoder-base")
+    p.add_argument("--reranker", type=str, default="sentio-reranker-base")
+    p.add_argument("--top-k", type=int, default=10)
+    p.add_argument("--rerank-top-k", type=int, default=5)
+    p.add_argument("--select-top-k", type=int, default=3)
+    p.add_argument("--verify", action="store_true",
+                   help="include the verifier pass (config #4)")
+    p.add_argument("--verify-tokens", type=int, default=64)
+    p.add_argument("--vocab-terms", type=int, default=30000)
+    p.add_argument("--seq-len", type=int, default=2048,
+                   help="generation context budget (prompt cap)")
+    return p.parse_args()
+
+
+class SyntheticCorpus:
+    """Deterministic on-the-fly document payloads — 10M stored texts would
+    be pure host-RAM waste; content derives from the doc ref."""
+
+    WORDS = ["gpu", "hbm", "kernel", "index", "retrieval", "engine", "tensor",
+             "wave", "cache", "stream", "graph", "shard", "fusion", "decode"]
+
+    def __init__(self, shard: int):
+        self.shard = shard
+
+    def text_for(self, idx: int) -> str:
+        rng = np.random.RandomState((self.shard << 20) ^ idx)
+        words = rng.choice(self.WORDS, size=60)
+        return (f"synthetic document {self.shard}:{idx} " + " ".join(words))
+
+    def doc_for(self, ref: str) -> Document:
+        shard_s, idx_s = ref.split(":", 1)
+        return Document(text=self.text_for(int(idx_s)),
+                        metadata={"source": f"doc-{ref}"}, id=ref)
+
+
+def build_synthetic_indexes(n_docs: int, dim: int, vocab: int, device: str,
+                            rank: int):
+    """Dense: random unit vectors in HBM.  BM25: synthetic CSR postings
+    (zipf-ish doc frequencies) built directly on device-compatible arrays."""
+    g = torch.Generator(device="cpu")
+    g.manual_seed(1000 + rank)
+    dense = DenseIndex(dim=dim, device=device,
+                       dtype=torch.float16 if device != "cpu" else torch.float32)
+    # bulk vector fill without Document objects (payloads are synthetic)
+    chunk = 262144
+    dense._ensure_capacity(n_docs)
+    for i0 in range(0, n_docs, chunk):
+        n = min(chunk, n_docs - i0)
+        v = torch.randn(n, dim, device=device)
+        v = v / v.norm(dim=1, keepdim=True)
+        dense._vecs[i0:i0 + n] = v.to(dense.dtype)
+    dense._size = n_docs
+    dense.doc_ids = None  # replaced by int refs below
+
+    # BM25 postings: ~20 distinct terms per doc
+    per_doc = 20
+    nnz = n_docs * per_doc
+    rng = np.random.RandomState(2000 + rank)
+    # zipf-flavored term draw, clipped to vocab
+    terms = (rng.zipf(1.3, size=nnz) - 1) % vocab
+    docs = np.repeat(np.arange(n_docs, dtype=np.int32), per_doc)
+    order = np.argsort(terms, kind="stable")
+    terms_sorted = terms[order]
+    post_doc = docs[order]
+    post_tf = (rng.randint(1, 5, size=nnz)).astype(np.float32)
+    counts = np.bincount(terms_sorted, minlength=vocab)
+    indptr = np.zeros(vocab + 1, np.int64)
+    np.cumsum(counts, out=indptr[1:])
+
+    bm = BM25Index()
+    bm.doc_ids = None
+    bm.doc_len = rng.randint(40, 200, size=n_docs).astype(np.float32)
+    bm.vocab = {f"term{i}": i for i in range(vocab)}
+    bm.indptr = indptr
+    bm.post_doc = post_doc
+    bm.post_tf = post_tf
+    df = counts.astype(np.float64)
+    bm.idf = np.log((n_docs - df + 0.5) / (df + 0.5) + 1.0).astype(np.float32)
+    return dense, bm
+
+
+def dense_search_ids(dense: DenseIndex, q: torch.Tensor, k: int):
+    """Index-level search returning integer row ids (bulk synthetic corpus
+    has no Document list)."""
+    if dense._size == 0:
+        return [[] for _ in range(q.shape[0])]
+    k = min(k, dense._size)
+    qq = q.to(dense.device, torch.float32)
+    qq = qq / qq.norm(dim=1, keepdim=True).clamp_min(1e-12)
+    if dense.device != "cpu":
+        from sentio_amd import ops
+
+        vals, idx = ops.cosine_topk(qq.to(dense.dtype), dense._vecs[: dense._size], k)
+    else:
+        scores = qq @ dense._vecs[: dense._size].T.float()
+        vals, idx = torch.topk(scores, k, dim=1)
+    return vals, idx
+
+
+def bm25_search_ids(bm: BM25Index, term_ids: np.ndarray, k: int, device: str):
+    import torch as T
+
+    if device != "cpu":
+        from sentio_amd import ops
+
+        if not bm._device_arrays:
+            bm._device_arrays = {
+                "indptr": T.from_numpy(bm.indptr).to(device),
+                "post_doc": T.from_numpy(bm.post_doc).to(device),
+                "post_tf": T.from_numpy(bm.post_tf).to(device),
+                "idf": T.from_numpy(bm.idf).to(device),
+                "doc_len": T.from_numpy(bm.doc_len).to(device),
+            }
+        a = bm._device_arrays
+        tids = T.from_numpy(term_ids).to(device)
+        scores = ops.bm25_score(
+            tids, a["indptr"], a["post_doc"], a["post_tf"], a["idf"],
+            a["doc_len"], n_docs=len(bm.doc_len), k1=bm.k1, b=bm.b,
+            avgdl=float(bm.doc_len.mean()), plus_delta=0.0)
+        vals, idx = T.topk(scores, min(k, scores.shape[0]))
+        return vals.cpu().tolist(), idx.cpu().tolist()
+    # CPU fallback
+    scores = np.zeros(len(bm.doc_len), np.float32)
+    avgdl = max(float(bm.doc_len.mean()), 1e-9)
+    den = bm.k1 * (1 - bm.b + bm.b * bm.doc_len / avgdl)
+    for t in term_ids:
+        lo, hi = bm.indptr[t], bm.indptr[t + 1]
+        d = bm.post_doc[lo:hi]
+        tf = bm.post_tf[lo:hi]
+        scores[d] += bm.idf[t] * tf * (bm.k1 + 1) / (tf + den[d])
+    idx = np.argsort(-scores)[:k]
+    return scores[idx].tolist(), idx.tolist()
+
+
+def main():
+    args = parse_args()
+    rank, world = D.init_distributed()
+    on_gpu = torch.cuda.is_available()
+    device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}" if on_gpu else "cpu"
+    if on_gpu:
+        torch.cuda.set_device(device)
+
+    # CPU plumbing fallback (no GPU in the dev container)
+    if not on_gpu:
+        args.model = "tiny-decoder64"
+        args.encoder = "tiny-encoder"
+        args.reranker = "tiny-reranker"
+        args.docs_per_gpu = min(args.docs_per_gpu, 2000)
+        args.batch = min(args.batch, 2)
+        args.gen_tokens = min(args.gen_tokens, 8)
+
+    from sentio_amd.engines.encoder import EncoderEngine
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.engines.reranker import RerankerEngine
+
+    t_init = time.time()
+    encoder = EncoderEngine(args.encoder, device=device, max_seq=128)
+    reranker = RerankerEngine(args.reranker, device=device, max_seq=256)
+    generator = GeneratorEngine(args.model, device=device,
+                                max_seq=args.seq_len + args.gen_tokens + 8)
+    dense, bm = build_synthetic_indexes(
+        args.docs_per_gpu, encoder.dim, args.vocab_terms, device, rank)
+    corpus = SyntheticCorpus(rank)
+    builder = PromptBuilder("balanced")
+    if on_gpu:
+        torch.cuda.synchronize()
+    init_s = time.time() - t_init
+
+    rng = np.random.RandomState(42 + rank)
+
+    def make_queries(step: int) -> list[str]:
+        out = []
+        for b in range(args.batch):
+            terms = rng.randint(0, args.vocab_terms, size=4)
+            out.append(
+                f"what does term{terms[0]} term{terms[1]} mean for "
+                f"term{terms[2]} term{terms[3]} retrieval step {step}?")
+        return out
+
+    def chat_step(step: int) -> None:
+        queries = make_queries(step)
+        # 1. embed queries (one encoder batch)
+        qv = encoder.embed(queries)
+        # 2. dense search over ALL ranks' shards (SPMD all-gather)
+        if world > 1:
+            q_all = D.all_gather_tensor(qv)
+        else:
+            q_all = qv
+        vals, idx = dense_search_ids(dense, q_all, args.top_k)
+        vals = vals.cpu(); idx = idx.cpu()
+        if world > 1:
+            gathered_v = D.all_gather_objects(vals.numpy())
+            gathered_i = D.all_gather_objects(idx.numpy())
+        else:
+            gathered_v = [vals.numpy()]; gathered_i = [idx.numpy()]
+        # 3. sparse search (local shard) + gather
+        sparse_hits_local = []
+        for q in queries:
+            tids = np.array([int(tok[4:]) for tok in q.split()
+                             if tok.startswith("term") and tok[4:].isdigit()],
+                            np.int64)
+            sv, si = bm25_search_ids(bm, tids, args.top_k, device)
+            sparse_hits_local.append(list(zip(si, sv)))
+        if world > 1:
+            sparse_all = D.all_gather_objects(sparse_hits_local)
+        else:
+            sparse_all = [sparse_hits_local]
+
+        # 4. per-query merge + fusion (owner = this rank's queries)
+        base = rank * args.batch
+        batch_docs = []
+        for qi in range(args.batch):
+            row = base + qi
+            dense_cand = []
+            for shard in range(world):
+                for j in range(gathered_i[shard].shape[1]):
+                    dense_cand.append((f"{shard}:{gathered_i[shard][row][j]}",
+                                       float(gathered_v[shard][row][j])))
+            dense_cand.sort(key=lambda x: x[1], reverse=True)
+            dense_cand = dense_cand[: args.top_k]
+            sparse_cand = []
+            for shard in range(world):
+                for di, s in sparse_all[shard][qi]:
+                    sparse_cand.append((f"{shard}:{di}", float(s)))
+            sparse_cand.sort(key=lambda x: x[1], reverse=True)
+            sparse_cand = sparse_cand[: args.top_k]
+            fused = fusion.fuse(dense_cand, sparse_cand, method="rrf",
+                                top_k=args.top_k, rrf_k=60)
+            docs = [corpus.doc_for(ref) for ref, _ in fused]
+            for d, (_, s) in zip(docs, fused):
+                d.metadata["score"] = s
+            batch_docs.append(docs)
+
+        # 5. rerank all queries' candidates in one cross-encoder batch
+        pair_texts = []
+        spans = []
+        for qi, docs in enumerate(batch_docs):
+            top_n = min(len(docs), 2 * args.rerank_top_k)
+            spans.append((len(pair_texts), top_n))
+            pair_texts.extend(f"{queries[qi]}\n{d.text}" for d in docs[:top_n])
+        scores = reranker.score_pairs("", pair_texts) if pair_texts else []
+        # NOTE: pair text already contains the query; score_pairs prefixes
+        # query="" so the packed text is the pair.
+        reranked = []
+        for qi, (off, n) in enumerate(spans):
+            cand = batch_docs[qi][:n]
+            sc = scores[off: off + n]
+            order = sorted(range(n), key=lambda i: sc[i], reverse=True)
+            keep = []
+            for i in order[: args.rerank_top_k]:
+                cand[i].metadata["score"] = float(sc[i])
+                keep.append(cand[i])
+            reranked.append(keep)
+
+        # 6. select + prompts
+        prompts = []
+        for qi in range(args.batch):
+            docs = reranked[qi][: args.select_top_k]
+            ctx = prepare_context(docs)
+            prompts.append(builder.system_prompt() + "\n\n" +
+                           builder.build_qa_prompt(queries[qi], ctx))
+        # 7. batched generation
+        answers = generator.generate(prompts, max_new_tokens=args.gen_tokens,
+                                     temperature=0.3, stop_on_eos=False)
+        # 8. optional verify
+        if args.verify:
+            vprompts = [builder.build_verify_prompt(
+                query=queries[qi], context=prepare_context(reranked[qi]),
+                answer=answers[qi]) for qi in range(args.batch)]
+            generator.generate(vprompts, max_new_tokens=args.verify_tokens,
+                               temperature=0.0, stop_on_eos=False)
+
+    # ---- warmup ----
+    for w in range(args.warmup):
+        chat_step(w)
+    D.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+
+    # ---- timed ----
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        chat_step(1000 + s)
+    D.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    e = torch.tensor([elapsed])
+    if D.is_distributed():
+        import torch.distributed as tdist
+
+        tdist.all_reduce(e, op=tdist.ReduceOp.MAX)
+    elapsed = float(e.item())
+
+    total_requests = args.batch * world * args.steps
+    qps = total_requests / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        result = {
+            "metric": "chat_qps",
+            "value": round(qps, 3),
+            "unit": "requests/s",
+            "n_gpus": world if on_gpu else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "encoder": args.encoder,
+                "reranker": args.reranker,
+                "global_batch": args.batch * world,
+                "seq_len": args.seq_len,
+                "gen_tokens": args.gen_tokens,
+                "docs_per_gpu": args.docs_per_gpu,
+                "total_docs": args.docs_per_gpu * world,
+                "parallelism": f"dp{world}+index-shard{world}",
+                "pipeline": "embed>hybrid(dense+bm25+rrf)>rerank>select>generate"
+                            + (">verify" if args.verify else ""),
+                "p50_ms_per_request_batch": round(ms_per_step, 1),
+                "init_s": round(init_s, 1),
+                "device": "cuda" if on_gpu else "cpu-plumbing",
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

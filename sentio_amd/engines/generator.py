@@ -25,13 +25,19 @@ MODE_TEMPERATURE = {"fast": 0.0, "balanced": 0.3, "quality": 0.2, "creative": 0.
 
 class GeneratorEngine:
     def __init__(self, model: str = "llama3-8b", device: str = "cpu",
-                 dtype: str = "bf16", max_seq: int = 4096, seed: int = 303):
+                 dtype: str = "bf16", max_seq: int = 4096, seed: int = 303,
+                 tp=None):
         self.cfg = get_model_config(model)
         self.device = device
         self.max_seq = min(max_seq, self.cfg.max_seq)
         self.tokenizer = ByteTokenizer()
-        self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
+        self.model = Transformer(self.cfg, device=device, dtype=dtype,
+                                 seed=seed, tp=tp)
         self._step_seed = 0
+
+    def _new_cache(self, batch: int, max_seq: int) -> KVCache:
+        return KVCache(self.cfg, batch, max_seq, self.device, self.model.dtype,
+                       n_kv_heads=self.model.hkv_local)
 
     @torch.inference_mode()
     def generate(
@@ -56,8 +62,7 @@ class GeneratorEngine:
         )
         tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
         S = tokens.shape[1]
-        cache = KVCache(self.cfg, B, min(S + max_new_tokens + 1, self.max_seq),
-                        self.device, self.model.dtype)
+        cache = self._new_cache(B, min(S + max_new_tokens + 1, self.max_seq))
         logits = self.model.prefill(tokens, cache)
 
         finished = torch.zeros(B, dtype=torch.bool, device=self.device)
@@ -92,8 +97,7 @@ class GeneratorEngine:
         prompt_budget = self.max_seq - max_new_tokens - 1
         ids = self.tokenizer.encode(prompt[-4 * prompt_budget:], prompt_budget)
         tokens = torch.tensor([ids], dtype=torch.int64, device=self.device)
-        cache = KVCache(self.cfg, 1, min(len(ids) + max_new_tokens + 1, self.max_seq),
-                        self.device, self.model.dtype)
+        cache = self._new_cache(1, min(len(ids) + max_new_tokens + 1, self.max_seq))
         logits = self.model.prefill(tokens, cache)
         cur = self._sample(logits, temperature)
         generated: list[int] = []

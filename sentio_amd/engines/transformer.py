@@ -25,6 +25,7 @@ import torch
 
 from sentio_amd import ops
 from sentio_amd.engines.configs import ModelConfig
+from sentio_amd.parallel.tp import TPContext, shard_columns, shard_qkv, shard_rows
 
 
 def _dtype(name: str) -> torch.dtype:
@@ -35,15 +36,17 @@ class TransformerWeights:
     """Flat weight container (not nn.Module: no autograd needed for serving)."""
 
     def __init__(self, cfg: ModelConfig, device: str, dtype: torch.dtype,
-                 seed: int = 1234):
+                 seed: int = 1234, tp: TPContext | None = None):
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
+        self.tp = tp or TPContext()
         gen = torch.Generator(device="cpu")
         gen.manual_seed(seed)
         d, hd = cfg.dim, cfg.head_dim
         qkv_out = (cfg.n_heads + 2 * cfg.n_kv_heads) * hd
         std = 0.02
+        f = cfg.ffn_dim
 
         def init(*shape):
             # init on device when possible for speed; generator is CPU-side so
@@ -57,16 +60,29 @@ class TransformerWeights:
         if device != "cpu":
             torch.manual_seed(seed)
 
+        tp = self.tp
         self.tok_emb = init(cfg.vocab_size, d)
         self.layers = []
         for _ in range(cfg.n_layers):
+            # full tensors are drawn identically on every rank (same seed and
+            # order), then sliced — TP=N matches TP=1 numerically.
+            wqkv = shard_qkv(init(d, qkv_out), cfg.n_heads, cfg.n_kv_heads, hd, tp)
+            wo = shard_rows(init(cfg.n_heads * hd, d), tp)
+            wgu_full = init(d, 2 * f)
+            if tp.enabled:
+                w_gate_up = torch.cat(
+                    [shard_columns(wgu_full[:, :f], tp),
+                     shard_columns(wgu_full[:, f:], tp)], dim=1).contiguous()
+            else:
+                w_gate_up = wgu_full
+            w_down = shard_rows(init(f, d), tp)
             self.layers.append({
                 "attn_norm": torch.ones(d, dtype=dtype, device=device),
-                "wqkv": init(d, qkv_out),
-                "wo": init(cfg.n_heads * hd, d),
+                "wqkv": wqkv,
+                "wo": wo,
                 "ffn_norm": torch.ones(d, dtype=dtype, device=device),
-                "w_gate_up": init(d, 2 * cfg.ffn_dim),
-                "w_down": init(cfg.ffn_dim, d),
+                "w_gate_up": w_gate_up,
+                "w_down": w_down,
             })
         self.final_norm = torch.ones(d, dtype=dtype, device=device)
         if cfg.causal:
@@ -87,14 +103,15 @@ class TransformerWeights:
 
 class KVCache:
     def __init__(self, cfg: ModelConfig, batch: int, max_seq: int, device: str,
-                 dtype: torch.dtype):
+                 dtype: torch.dtype, n_kv_heads: int | None = None):
+        nkv = n_kv_heads or cfg.n_kv_heads
         self.k = [
-            torch.zeros(batch, cfg.n_kv_heads, max_seq, cfg.head_dim,
+            torch.zeros(batch, nkv, max_seq, cfg.head_dim,
                         dtype=dtype, device=device)
             for _ in range(cfg.n_layers)
         ]
         self.v = [
-            torch.zeros(batch, cfg.n_kv_heads, max_seq, cfg.head_dim,
+            torch.zeros(batch, nkv, max_seq, cfg.head_dim,
                         dtype=dtype, device=device)
             for _ in range(cfg.n_layers)
         ]
@@ -106,14 +123,21 @@ class Transformer:
     """Forward-only transformer executing on sentio ops."""
 
     def __init__(self, cfg: ModelConfig, device: str = "cpu",
-                 dtype: str | torch.dtype = "bf16", seed: int = 1234):
+                 dtype: str | torch.dtype = "bf16", seed: int = 1234,
+                 tp: TPContext | None = None):
         self.cfg = cfg
         self.device = device
+        self.tp = tp or TPContext()
+        if self.tp.enabled:
+            assert cfg.n_heads % self.tp.world == 0
+            assert cfg.n_kv_heads % self.tp.world == 0
+        self.h_local = cfg.n_heads // self.tp.world
+        self.hkv_local = cfg.n_kv_heads // self.tp.world
         self.dtype = _dtype(dtype) if isinstance(dtype, str) else dtype
         if device == "cpu":
             # bf16 matmuls on CPU are slow and loose; tests run fp32
             self.dtype = torch.float32
-        self.w = TransformerWeights(cfg, device, self.dtype, seed)
+        self.w = TransformerWeights(cfg, device, self.dtype, seed, self.tp)
         cos, sin = ops.torch_ref.rope_tables(cfg.max_seq, cfg.head_dim,
                                              cfg.rope_base, device)
         self.rope_cos, self.rope_sin = cos, sin
@@ -126,44 +150,48 @@ class Transformer:
         cfg = self.cfg
         B, S, d = x.shape
         hd = cfg.head_dim
+        H, Hkv = self.h_local, self.hkv_local
         qkv = x.view(B * S, d) @ layer["wqkv"]
         qkv = qkv.view(B, S, -1)
-        q_end = cfg.n_heads * hd
-        k_end = q_end + cfg.n_kv_heads * hd
-        q = qkv[..., :q_end].reshape(B, S, cfg.n_heads, hd)
-        k = qkv[..., q_end:k_end].reshape(B, S, cfg.n_kv_heads, hd)
-        v = qkv[..., k_end:].reshape(B, S, cfg.n_kv_heads, hd)
+        q_end = H * hd
+        k_end = q_end + Hkv * hd
+        q = qkv[..., :q_end].reshape(B, S, H, hd)
+        k = qkv[..., q_end:k_end].reshape(B, S, Hkv, hd)
+        v = qkv[..., k_end:].reshape(B, S, Hkv, hd)
         q = ops.rope_apply(q, self.rope_cos, self.rope_sin, pos)
         k = ops.rope_apply(k, self.rope_cos, self.rope_sin, pos)
 
         if cache is not None:
-            # write k/v at pos into the cache: [B, Hkv, Smax, hd]
+            # write k/v at pos into the cache: [B, Hkv_local, Smax, hd]
             kc, vc = cache.k[layer_idx], cache.v[layer_idx]
             idx = pos.long()  # [B,S]
             bidx = torch.arange(B, device=x.device).unsqueeze(1).expand(B, S)
             kc[bidx.reshape(-1), :, idx.reshape(-1)] = \
-                k.reshape(B * S, cfg.n_kv_heads, hd).to(kc.dtype)
+                k.reshape(B * S, Hkv, hd).to(kc.dtype)
             vc[bidx.reshape(-1), :, idx.reshape(-1)] = \
-                v.reshape(B * S, cfg.n_kv_heads, hd).to(vc.dtype)
+                v.reshape(B * S, Hkv, hd).to(vc.dtype)
 
         if S == 1 and cache is not None:
             seq_lens = (pos[:, 0] + 1).to(torch.int32)
             out = ops.decode_attention(
-                q.view(B, cfg.n_heads, hd), cache.k[layer_idx],
+                q.view(B, H, hd), cache.k[layer_idx],
                 cache.v[layer_idx], seq_lens, self.scale,
-            ).view(B, 1, cfg.n_heads, hd)
+            ).view(B, 1, H, hd)
         else:
             out = ops.attention(q, k, v, causal=cfg.causal, scale=self.scale,
                                 kv_lens=kv_lens)
-        out = out.reshape(B * S, cfg.n_heads * hd) @ layer["wo"]
+        out = out.reshape(B * S, H * hd) @ layer["wo"]
+        out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
         return out.view(B, S, d)
 
     def _ffn(self, x: torch.Tensor, layer: dict) -> torch.Tensor:
         B, S, d = x.shape
         gu = x.view(B * S, d) @ layer["w_gate_up"]
-        f = self.cfg.ffn_dim
-        y = ops.swiglu(gu[:, :f], gu[:, f:])
-        return (y @ layer["w_down"]).view(B, S, d)
+        f = self.cfg.ffn_dim // self.tp.world
+        y = ops.swiglu(gu[:, :f].contiguous(), gu[:, f:].contiguous())
+        out = y @ layer["w_down"]
+        out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
+        return out.view(B, S, d)
 
     def forward_hidden(
         self, tokens: torch.Tensor, pos: torch.Tensor | None = None,

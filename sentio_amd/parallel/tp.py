@@ -1,0 +1,80 @@
+"""Tensor parallelism for the generator.
+
+Megatron-style decomposition over the node's GPUs:
+* attention: QKV projection column-parallel (head-sharded), output
+  projection row-parallel → one all-reduce per layer;
+* FFN: gate+up column-parallel, down row-parallel → one all-reduce.
+xGMI is point-to-point (7 links x ~153 GB/s), so the per-layer all-reduce
+of [B·S, dim] bf16 activations is the bandwidth bound RCCL must schedule;
+bucket sizes stay whole-tensor (decode activations are tiny).
+
+The Transformer engine consumes this through `TPContext`, which tells the
+weight container to materialize only this rank's shard (sliced from the
+same seeded generator so TP=N matches TP=1 bit-for-bit in fp32).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+
+@dataclass
+class TPContext:
+    rank: int = 0
+    world: int = 1
+    group: object = None
+
+    @property
+    def enabled(self) -> bool:
+        return self.world > 1
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.enabled:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.group)
+        return t
+
+    @classmethod
+    def from_env(cls) -> "TPContext":
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            return cls(rank=dist.get_rank(), world=dist.get_world_size())
+        return cls()
+
+
+def shard_columns(w: torch.Tensor, tp: TPContext) -> torch.Tensor:
+    """Column-parallel shard of [in, out]: split the out dim."""
+    if not tp.enabled:
+        return w
+    out = w.shape[1]
+    assert out % tp.world == 0, "out dim must divide TP degree"
+    step = out // tp.world
+    return w[:, tp.rank * step : (tp.rank + 1) * step].contiguous()
+
+
+def shard_rows(w: torch.Tensor, tp: TPContext) -> torch.Tensor:
+    """Row-parallel shard of [in, out]: split the in dim."""
+    if not tp.enabled:
+        return w
+    inp = w.shape[0]
+    assert inp % tp.world == 0, "in dim must divide TP degree"
+    step = inp // tp.world
+    return w[tp.rank * step : (tp.rank + 1) * step].contiguous()
+
+
+def shard_qkv(w: torch.Tensor, n_heads: int, n_kv: int, head_dim: int,
+              tp: TPContext) -> torch.Tensor:
+    """Shard the fused QKV [dim, (H + 2*Hkv)*hd] by heads: this rank keeps
+    H/world query heads and Hkv/world kv heads (Hkv % world == 0)."""
+    if not tp.enabled:
+        return w
+    assert n_heads % tp.world == 0 and n_kv % tp.world == 0
+    hl = n_heads // tp.world
+    kl = n_kv // tp.world
+    q_end = n_heads * head_dim
+    k_end = q_end + n_kv * head_dim
+    q = w[:, tp.rank * hl * head_dim : (tp.rank + 1) * hl * head_dim]
+    k = w[:, q_end + tp.rank * kl * head_dim : q_end + (tp.rank + 1) * kl * head_dim]
+    v = w[:, k_end + tp.rank * kl * head_dim : k_end + (tp.rank + 1) * kl * head_dim]
+    return torch.cat([q, k, v], dim=1).contiguous()
