@@ -23,6 +23,55 @@ from sentio_amd.engines.transformer import KVCache, Transformer
 MODE_TEMPERATURE = {"fast": 0.0, "balanced": 0.3, "quality": 0.2, "creative": 0.7}
 
 
+class _DecodeSession:
+    """One (batch, cache-size) decode context: preallocated KV cache plus an
+    optional hipGraph capture of the whole per-token decode step."""
+
+    def __init__(self, engine: "GeneratorEngine", batch: int, cache_len: int,
+                 use_graphs: bool):
+        self.engine = engine
+        self.cache = engine._new_cache(batch, cache_len)
+        self.graph = None
+        self.static_tok = None
+        self.static_logits = None
+        self.use_graphs = use_graphs
+        self.batch = batch
+
+    def prefill(self, tokens: torch.Tensor) -> torch.Tensor:
+        self.cache.seq_lens.zero_()
+        return self.engine.model.prefill(tokens, self.cache)
+
+    def _capture(self):
+        model = self.engine.model
+        dev = self.engine.device
+        self.static_tok = torch.zeros(self.batch, 1, dtype=torch.int64,
+                                      device=dev)
+        # warmup on a side stream (required before capture), then capture.
+        # seq_lens must be restored: warmup/capture each advance it.
+        saved = self.cache.seq_lens.clone()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                model.decode_step(self.static_tok, self.cache)
+        torch.cuda.current_stream().wait_stream(s)
+        self.cache.seq_lens.copy_(saved)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.static_logits = model.decode_step(self.static_tok, self.cache)
+        self.cache.seq_lens.copy_(saved)
+
+    def decode_step(self, tok: torch.Tensor) -> torch.Tensor:
+        if not self.use_graphs:
+            return self.engine.model.decode_step(tok.view(self.batch, 1),
+                                                 self.cache)
+        if self.graph is None:
+            self._capture()
+        self.static_tok.copy_(tok.view(self.batch, 1))
+        self.graph.replay()
+        return self.static_logits
+
+
 class GeneratorEngine:
     def __init__(self, model: str = "llama3-8b", device: str = "cpu",
                  dtype: str = "bf16", max_seq: int = 4096, seed: int = 303,
@@ -38,6 +87,33 @@ class GeneratorEngine:
     def _new_cache(self, batch: int, max_seq: int) -> KVCache:
         return KVCache(self.cfg, batch, max_seq, self.device, self.model.dtype,
                        n_kv_heads=self.model.hkv_local)
+
+    # ----- hipGraph-captured decode session -----
+    # The per-token decode step launches ~10 kernels per layer; on MI355X the
+    # wall time at small batch is launch-bound, so the whole step is captured
+    # once as a hipGraph and replayed per token (sampling stays outside: its
+    # seed changes per step).  Sessions are pooled by (batch, cache size).
+    _GRAPHS_ENABLED = True
+
+    def _decode_session(self, batch: int, cache_len: int):
+        import os
+
+        use_graphs = (
+            self._GRAPHS_ENABLED
+            and self.device != "cpu"
+            and os.environ.get("SENTIO_DISABLE_HIPGRAPH", "0") != "1"
+        )
+        key = (batch, cache_len)
+        pool = getattr(self, "_sessions", None)
+        if pool is None:
+            pool = self._sessions = {}
+        if key in pool:
+            sess = pool[key]
+            sess.cache.seq_lens.zero_()
+            return sess
+        sess = _DecodeSession(self, batch, cache_len, use_graphs)
+        pool[key] = sess
+        return sess
 
     @torch.inference_mode()
     def generate(
@@ -62,8 +138,8 @@ class GeneratorEngine:
         )
         tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
         S = tokens.shape[1]
-        cache = self._new_cache(B, min(S + max_new_tokens + 1, self.max_seq))
-        logits = self.model.prefill(tokens, cache)
+        sess = self._decode_session(B, self.max_seq)
+        logits = sess.prefill(tokens)
 
         finished = torch.zeros(B, dtype=torch.bool, device=self.device)
         generated: list[list[int]] = [[] for _ in range(B)]
@@ -81,7 +157,7 @@ class GeneratorEngine:
                     break
             if step == max_new_tokens - 1:
                 break
-            logits = self.model.decode_step(cur.unsqueeze(1), cache)
+            logits = sess.decode_step(cur)
             cur = self._sample(logits, temperature)
         return [self.tokenizer.decode(g) for g in generated]
 
@@ -97,8 +173,8 @@ class GeneratorEngine:
         prompt_budget = self.max_seq - max_new_tokens - 1
         ids = self.tokenizer.encode(prompt[-4 * prompt_budget:], prompt_budget)
         tokens = torch.tensor([ids], dtype=torch.int64, device=self.device)
-        cache = self._new_cache(1, min(len(ids) + max_new_tokens + 1, self.max_seq))
-        logits = self.model.prefill(tokens, cache)
+        sess = self._decode_session(1, self.max_seq)
+        logits = sess.prefill(tokens)
         cur = self._sample(logits, temperature)
         generated: list[int] = []
         emitted = ""
@@ -114,5 +190,5 @@ class GeneratorEngine:
                 yield delta
             if step == max_new_tokens - 1:
                 break
-            logits = self.model.decode_step(cur.unsqueeze(1), cache)
+            logits = sess.decode_step(cur)
             cur = self._sample(logits, temperature)

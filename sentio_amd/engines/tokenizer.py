@@ -28,12 +28,30 @@ class ByteTokenizer:
             ids = ids[:max_len]
         return ids
 
+    # ids beyond the byte range appear when a model's vocab is larger than
+    # the tokenizer's (random-init logits sample the whole 128k vocab) —
+    # render them as deterministic placeholder words so generations are
+    # non-empty, readable text.
+    _PLACEHOLDER = (
+        "alpha beta gamma delta epsilon zeta eta theta iota kappa lambda mu "
+        "nu xi omicron pi rho sigma tau upsilon phi chi psi omega node edge "
+        "graph wave tile lane block grid cache token shard index fuse rank"
+    ).split()
+
     def decode(self, ids: list[int]) -> str:
-        # ids beyond the byte range can appear when a model's vocab is larger
-        # than the tokenizer's (random-init logits) — drop them on decode
-        data = bytes(i - BYTE_OFFSET for i in ids
-                     if BYTE_OFFSET <= i < BYTE_OFFSET + 256)
-        return data.decode("utf-8", errors="replace")
+        parts: list[str] = []
+        run: list[int] = []
+        for i in ids:
+            if BYTE_OFFSET <= i < BYTE_OFFSET + 256:
+                run.append(i - BYTE_OFFSET)
+            elif i >= BYTE_OFFSET + 256:
+                if run:
+                    parts.append(bytes(run).decode("utf-8", errors="replace"))
+                    run = []
+                parts.append(" " + self._PLACEHOLDER[i % len(self._PLACEHOLDER)])
+        if run:
+            parts.append(bytes(run).decode("utf-8", errors="replace"))
+        return "".join(parts)
 
     def encode_batch(
         self, texts: list[str], max_len: int

@@ -253,3 +253,22 @@ def test_engine_forward_gpu_matches_cpu(dev):
     lg = g_gpu.model.prefill(tokens_gpu, cache_g)
     cs = torch.nn.functional.cosine_similarity(lc[0], lg[0].cpu().float(), dim=0)
     assert cs.item() > 0.98
+
+
+def test_hipgraph_decode_matches_eager(dev):
+    """hipGraph-captured decode must produce the same logits as eager."""
+    import os
+
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    g = GeneratorEngine("tiny-decoder64", device=dev, max_seq=64)
+    out_graph = g.generate(["graph parity"], max_new_tokens=8, temperature=0.0,
+                           stop_on_eos=False)
+    os.environ["SENTIO_DISABLE_HIPGRAPH"] = "1"
+    try:
+        g2 = GeneratorEngine("tiny-decoder64", device=dev, max_seq=64)
+        out_eager = g2.generate(["graph parity"], max_new_tokens=8,
+                                temperature=0.0, stop_on_eos=False)
+    finally:
+        os.environ.pop("SENTIO_DISABLE_HIPGRAPH", None)
+    assert out_graph == out_eager
