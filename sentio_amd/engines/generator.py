@@ -141,25 +141,35 @@ class GeneratorEngine:
         sess = self._decode_session(B, self.max_seq)
         logits = sess.prefill(tokens)
 
-        finished = torch.zeros(B, dtype=torch.bool, device=self.device)
-        generated: list[list[int]] = [[] for _ in range(B)]
+        # tokens accumulate on-device; the host syncs only for EOS checks
+        # (every 16 steps) or the streaming callback — not per token.
+        toks_buf = torch.zeros(B, max_new_tokens, dtype=torch.int64,
+                               device=self.device)
         cur = self._sample(logits, temperature)
+        n_steps = max_new_tokens
         for step in range(max_new_tokens):
-            cur_list = cur.cpu().tolist()
-            for b, t in enumerate(cur_list):
-                if not finished[b]:
-                    generated[b].append(int(t))
+            toks_buf[:, step] = cur
             if on_token is not None:
-                on_token(step, cur_list)
-            if stop_on_eos:
-                finished |= cur == EOS_ID
-                if bool(finished.all()):
+                on_token(step, cur.cpu().tolist())
+            if stop_on_eos and (step % 16 == 15 or on_token is not None):
+                done = (toks_buf[:, : step + 1] == EOS_ID).any(dim=1)
+                if bool(done.all()):
+                    n_steps = step + 1
                     break
             if step == max_new_tokens - 1:
                 break
             logits = sess.decode_step(cur)
             cur = self._sample(logits, temperature)
-        return [self.tokenizer.decode(g) for g in generated]
+        rows = toks_buf[:, :n_steps].cpu().tolist()
+        out = []
+        for row in rows:
+            ids = []
+            for t in row:
+                if stop_on_eos and t == EOS_ID:
+                    break
+                ids.append(t)
+            out.append(self.tokenizer.decode(ids))
+        return out
 
     def _sample(self, logits: torch.Tensor, temperature: float) -> torch.Tensor:
         self._step_seed += 1
