@@ -38,7 +38,12 @@ __device__ __forceinline__ int vt_lds_off(int d, int key_byte) {
 // P tile [QBLK][KVBLK] bf16 with 16-byte row pad: stride = 64+16 = 80 B
 #define P_STRIDE 80
 
-__launch_bounds__(64)
+// 4 waves per block: waves handle 4 consecutive 16-row Q tiles of ONE
+// (batch, head) and SHARE the K/V LDS staging — 4x less HBM/LDS staging
+// traffic than wave-private tiles, cooperative 256-thread staging.
+#define FA_WAVES 4
+
+__launch_bounds__(256)
 __global__ void flash_attn_kernel(
     const bf16* __restrict__ q,    // [B, S, H, D]
     const bf16* __restrict__ k,    // [B, S, Hkv, D]
@@ -47,19 +52,22 @@ __global__ void flash_attn_kernel(
     const int* __restrict__ kv_lens,  // [B]
     int B, int S, int H, int Hkv, int D, float scale, int causal) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // carve: K tile | V^T tile | P tile
+  // carve: K tile | V^T tile | per-wave P tiles
   char* k_lds = smem;                                   // KVBLK * D * 2
   char* vt_lds = k_lds + KVBLK * D * 2;                 // D * VT_STRIDE
-  char* p_lds = vt_lds + D * VT_STRIDE;                 // QBLK * P_STRIDE
 
-  const int lane = threadIdx.x;
-  const int qt = blockIdx.x;            // q tile index
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  char* p_lds = vt_lds + D * VT_STRIDE + wid * QBLK * P_STRIDE;
+
+  const int qt = blockIdx.x * FA_WAVES + wid;  // this wave's q tile
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hkv = h / (H / Hkv);
   const int q0 = qt * QBLK;
   const int kvlen = min(kv_lens[b], S);
-  if (q0 >= S) return;
+  const bool active = q0 < S;           // inactive waves still hit barriers
+  const int q_hi = min(q0 + QBLK - 1, S - 1);
 
   const int DC = D / 32;                // feature chunks per mfma K-dim
   const int NB = D / 16;                // output column blocks
@@ -74,7 +82,7 @@ __global__ void flash_attn_kernel(
 #pragma unroll
     for (int dc = 0; dc < 4; ++dc) {
       if (dc < DC) {
-        if (qrow < S) {
+        if (active && qrow < S) {
           a_q[dc] = *reinterpret_cast<const bf16x8_t*>(qp + dc * 32);
         } else {
           bf16x8_t z = {};
@@ -94,17 +102,19 @@ __global__ void flash_attn_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
 
-  const int kv_hi = causal ? min(kvlen, q0 + QBLK) : kvlen;
+  // block-level kv bound: the LAST wave's causal horizon (waves skip their
+  // own out-of-horizon tiles compute-side)
+  const int block_q_hi = min(blockIdx.x * FA_WAVES * QBLK + FA_WAVES * QBLK,
+                             S);
+  const int kv_hi = causal ? min(kvlen, block_q_hi) : kvlen;
   const int n_kv_tiles = (kv_hi + KVBLK - 1) / KVBLK;
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int kv0 = kt * KVBLK;
-    // ---- stage K tile (swizzled) and V^T tile into LDS (one wave)
-    // K: rows kv0..kv0+31; each lane stages rows lane/2 (2 lanes per row,
-    // each lane covers D/2 bytes when D=128 → use generic loop
+    // ---- stage K tile (swizzled) and V^T tile into LDS (256 threads)
     {
       const int elems = KVBLK * D;      // bf16 elements in tile
-      for (int i = lane * 8; i < elems; i += WAVE * 8) {
+      for (int i = threadIdx.x * 8; i < elems; i += FA_WAVES * WAVE * 8) {
         const int row = i / D;
         const int d = i % D;
         const int key = kv0 + row;
@@ -132,7 +142,11 @@ __global__ void flash_attn_kernel(
           *reinterpret_cast<__bf16*>(vt_lds + vt_lds_off(d + j, row * 2)) = vv[j];
       }
     }
-    __syncthreads();  // single wave: compiles to s_waitcnt; keeps LDS ordered
+    __syncthreads();  // staging visible to every wave
+
+    // waves whose causal horizon ends before this kv tile skip compute but
+    // still execute every barrier (uniform control flow)
+    const bool compute = active && (!causal || kv0 <= q_hi);
 
     // ---- S = scale * Q K^T for the two 16-key halves
     f32x4_t s_acc[2];
@@ -156,6 +170,9 @@ __global__ void flash_attn_kernel(
     // lane holds: col = l&15 (+16*half), rows = (l>>4)*4 + r
     float p_val[2][4];
     float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) alpha[r] = 1.f;
+    if (compute)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow_local = (lane >> 4) * 4 + r;
@@ -185,46 +202,52 @@ __global__ void flash_attn_kernel(
       p_val[1][r] = p1;
     }
 
-    // ---- write P to LDS in C layout, reread as A fragments
-    __syncthreads();
+    // ---- write P to LDS (wave-private buffer) in C layout, reread as A
+    // fragments.  The in-wave ds-write -> ds-read ordering is provided by
+    // the block barrier's lgkmcnt drain; barriers stay uniform across waves.
+    if (compute) {
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+      for (int half = 0; half < 2; ++half) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow_local = (lane >> 4) * 4 + r;
-        const int key_local = half * 16 + (lane & 15);
-        *reinterpret_cast<__bf16*>(
-            p_lds + qrow_local * P_STRIDE + key_local * 2) =
-            (__bf16)p_val[half][r];
+        for (int r = 0; r < 4; ++r) {
+          const int qrow_local = (lane >> 4) * 4 + r;
+          const int key_local = half * 16 + (lane & 15);
+          *reinterpret_cast<__bf16*>(
+              p_lds + qrow_local * P_STRIDE + key_local * 2) =
+              (__bf16)p_val[half][r];
+        }
       }
     }
     __syncthreads();
-    // A frag for PV: P[row = l&15][k = key = kofs + j] — one frag covers all
-    // 32 keys of the tile (k = (l>>4)*8 + j spans 0..31)
-    const bf16x8_t a_p = *reinterpret_cast<const bf16x8_t*>(
-        p_lds + (lane & 15) * P_STRIDE + kofs * 2);
+    if (compute) {
+      // A frag for PV: P[row = l&15][k = key = kofs + j] — one frag covers
+      // all 32 keys of the tile (k = (l>>4)*8 + j spans 0..31)
+      const bf16x8_t a_p = *reinterpret_cast<const bf16x8_t*>(
+          p_lds + (lane & 15) * P_STRIDE + kofs * 2);
 
-    // ---- O = alpha*O + P V   (one mfma per 16-col block of V)
+      // ---- O = alpha*O + P V   (one mfma per 16-col block of V)
 #pragma unroll
-    for (int nb = 0; nb < MAXD / 16; ++nb) {
-      if (nb < NB) {
-        // rescale accumulator rows by alpha[r]
+      for (int nb = 0; nb < MAXD / 16; ++nb) {
+        if (nb < NB) {
+          // rescale accumulator rows by alpha[r]
 #pragma unroll
-        for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
-        // B frag: V^T[d = nb*16 + (l&15)][key = kofs + j]
-        bf16x8_t b_v;
+          for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
+          // B frag: V^T[d = nb*16 + (l&15)][key = kofs + j]
+          bf16x8_t b_v;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          b_v[j] = *reinterpret_cast<const __bf16*>(
-              vt_lds + vt_lds_off(nb * 16 + (lane & 15), (kofs + j) * 2));
-        o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_p, b_v, o_acc[nb], 0, 0, 0);
+          for (int j = 0; j < 8; ++j)
+            b_v[j] = *reinterpret_cast<const __bf16*>(
+                vt_lds + vt_lds_off(nb * 16 + (lane & 15), (kofs + j) * 2));
+          o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_p, b_v, o_acc[nb], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
   }
 
   // ---- epilogue: divide by l, store
+  if (!active) return;
 #pragma unroll
   for (int nb = 0; nb < MAXD / 16; ++nb) {
     if (nb < NB) {
@@ -249,73 +272,110 @@ __global__ void flash_attn_kernel(
 //   phase A: thread-per-key dot products (vectorized K row reads)
 //   phase B: thread-per-dim PV accumulation (coalesced V reads)
 #define DEC_CHUNK 256
+#define DEC_MAXG 8   // max query heads per kv head handled by one block
 
+// GQA-aware: one block per (batch, kv-head) streams the KV cache ONCE and
+// serves all G = H/Hkv query heads of the group — 1/G the HBM traffic of a
+// block-per-query-head layout.  Template on G so per-thread accumulator
+// arrays stay in registers (guide §5.4 rule 20).
+template <int G>
 __global__ void decode_attn_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc, bf16* __restrict__ out,
     const int* __restrict__ seq_lens,
     int H, int Hkv, int Smax, int D, float scale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* p_sh = reinterpret_cast<float*>(smem);            // [DEC_CHUNK]
-  float* q_sh = p_sh + DEC_CHUNK;                          // [D]
-  float* red = q_sh + D;                                   // [32] scratch
+  float* p_sh = reinterpret_cast<float*>(smem);            // [G][DEC_CHUNK]
+  float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
+  float* red = q_sh + G * D;                               // [32] scratch
 
-  const int h = blockIdx.x;
+  const int hkv = blockIdx.x;
   const int b = blockIdx.y;
-  const int hkv = h / (H / Hkv);
   const int slen = seq_lens[b];
-  const bf16* qb = q + ((long)b * H + h) * D;
   const bf16* kb = kc + ((long)b * Hkv + hkv) * Smax * (long)D;
   const bf16* vb = vc + ((long)b * Hkv + hkv) * Smax * (long)D;
 
-  for (int d = threadIdx.x; d < D; d += blockDim.x) q_sh[d] = bf2f(qb[d]);
+  for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
+    const int g = i / D, d = i % D;
+    q_sh[i] = bf2f(q[((long)b * H + hkv * G + g) * D + d]);
+  }
   __syncthreads();
 
-  float m_run = -INFINITY, l_run = 0.f;
-  // per-thread O accumulators over dims (D <= 256 with 256 threads)
-  float o0 = 0.f;
+  float m_run[G], l_run[G], alpha[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) { m_run[g] = -INFINITY; l_run[g] = 0.f; }
+  float o0[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) o0[g] = 0.f;
   const int myd = threadIdx.x;          // dim owned in phase B (if < D)
 
   for (int s0 = 0; s0 < slen; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, slen - s0);
-    // phase A: dot for own key
-    float sc = -INFINITY;
+    // phase A: this thread's key, dots vs all G query heads
+    float sc[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) sc[g] = -INFINITY;
     const int s = s0 + threadIdx.x;
     if (threadIdx.x < chunk) {
       const bf16* krow = kb + (long)s * D;
-      float acc = 0.f;
+#pragma unroll
+      for (int g = 0; g < G; ++g) sc[g] = 0.f;
       for (int d = 0; d < D; d += 8) {
         bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const short*>(krow) + d);
+        float kf[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc += bits2f(kv8[j]) * q_sh[d + j];
+        for (int j = 0; j < 8; ++j) kf[j] = bits2f(kv8[j]);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          float acc = 0.f;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) acc += kf[j] * q_sh[g * D + d + j];
+          sc[g] += acc;
+        }
       }
-      sc = acc * scale;
+#pragma unroll
+      for (int g = 0; g < G; ++g) sc[g] *= scale;
     }
-    // chunk max
-    float cmax = block_max(sc, red);
-    float m_new = fmaxf(m_run, cmax);
-    float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
-    if (m_new == -INFINITY) alpha = 1.f;
-    float p = (threadIdx.x < chunk && sc != -INFINITY) ? __expf(sc - m_new) : 0.f;
-    p_sh[threadIdx.x] = p;
-    float csum = block_sum(p, red);
-    l_run = l_run * alpha + csum;
-    m_run = m_new;
+    // per-head block max/sum + p staging
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      float cmax = block_max(sc[g], red);
+      float m_new = fmaxf(m_run[g], cmax);
+      float a = (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+      if (m_new == -INFINITY) a = 1.f;
+      alpha[g] = a;
+      float p = (threadIdx.x < chunk && sc[g] != -INFINITY)
+                    ? __expf(sc[g] - m_new) : 0.f;
+      p_sh[g * DEC_CHUNK + threadIdx.x] = p;
+      float csum = block_sum(p, red);
+      l_run[g] = l_run[g] * a + csum;
+      m_run[g] = m_new;
+    }
     __syncthreads();
-    // phase B: PV for own dim
+    // phase B: PV — V row element read ONCE, fed to all G heads
     if (myd < D) {
-      float acc = 0.f;
-      for (int j = 0; j < chunk; ++j)
-        acc += p_sh[j] * bf2f(vb[(long)(s0 + j) * D + myd]);
-      o0 = o0 * alpha + acc;
+      float acc[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) acc[g] = 0.f;
+      for (int j = 0; j < chunk; ++j) {
+        const float vv = bf2f(vb[(long)(s0 + j) * D + myd]);
+#pragma unroll
+        for (int g = 0; g < G; ++g)
+          acc[g] += p_sh[g * DEC_CHUNK + j] * vv;
+      }
+#pragma unroll
+      for (int g = 0; g < G; ++g) o0[g] = o0[g] * alpha[g] + acc[g];
     }
     __syncthreads();
   }
 
   if (myd < D) {
-    const float denom = l_run > 0.f ? l_run : 1.f;
-    out[((long)b * H + h) * D + myd] = f2bf(o0 / denom);
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const float denom = l_run[g] > 0.f ? l_run[g] : 1.f;
+      out[((long)b * H + hkv * G + g) * D + myd] = f2bf(o0[g] / denom);
+    }
   }
 }
 
@@ -326,9 +386,10 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
                              int H, int Hkv, int D, float scale, int causal,
                              hipStream_t stream) {
   if (D % 32 != 0 || D > MAXD) return hipErrorInvalidValue;
-  size_t lds = (size_t)KVBLK * D * 2 + (size_t)D * VT_STRIDE + QBLK * P_STRIDE;
-  dim3 grid((S + QBLK - 1) / QBLK, H, B);
-  hipLaunchKernelGGL(flash_attn_kernel, grid, dim3(64), lds, stream,
+  size_t lds = (size_t)KVBLK * D * 2 + (size_t)D * VT_STRIDE
+               + 4 * QBLK * P_STRIDE;
+  dim3 grid((S + 4 * QBLK - 1) / (4 * QBLK), H, B);
+  hipLaunchKernelGGL(flash_attn_kernel, grid, dim3(256), lds, stream,
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,
                      (bf16*)out, kv_lens, B, S, H, Hkv, D, scale, causal);
   HIP_CHECK_LAUNCH();
@@ -340,11 +401,24 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
                               int Hkv, int Smax, int D, float scale,
                               hipStream_t stream) {
   if (D > 256) return hipErrorInvalidValue;
-  size_t lds = (DEC_CHUNK + D + 32) * sizeof(float);
-  dim3 grid(H, B);
-  hipLaunchKernelGGL(decode_attn_kernel, grid, dim3(DEC_CHUNK), lds, stream,
-                     (const bf16*)q, (const bf16*)kc, (const bf16*)vc,
-                     (bf16*)out, seq_lens, H, Hkv, Smax, D, scale);
+  const int G = H / Hkv;
+  if (G > DEC_MAXG || H % Hkv) return hipErrorInvalidValue;
+  size_t lds = (size_t)(G * DEC_CHUNK + G * D + 32) * sizeof(float);
+  dim3 grid(Hkv, B);
+#define DEC_CASE(GV)                                                          \
+  case GV:                                                                    \
+    hipLaunchKernelGGL((decode_attn_kernel<GV>), grid, dim3(DEC_CHUNK), lds,  \
+                       stream, (const bf16*)q, (const bf16*)kc,               \
+                       (const bf16*)vc, (bf16*)out, seq_lens, H, Hkv, Smax,   \
+                       D, scale);                                             \
+    break;
+  switch (G) {
+    DEC_CASE(1) DEC_CASE(2) DEC_CASE(3) DEC_CASE(4)
+    DEC_CASE(5) DEC_CASE(6) DEC_CASE(7) DEC_CASE(8)
+    default:
+      return hipErrorInvalidValue;
+  }
+#undef DEC_CASE
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
