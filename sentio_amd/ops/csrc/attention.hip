@@ -278,12 +278,19 @@ __global__ void flash_attn_kernel(
 // serves all G = H/Hkv query heads of the group — 1/G the HBM traffic of a
 // block-per-query-head layout.  Template on G so per-thread accumulator
 // arrays stay in registers (guide §5.4 rule 20).
+// Split-S ("flash-decoding") layout: grid (Hkv, B, SPLITS); each split
+// streams its contiguous key range once for all G heads and writes
+// UNNORMALIZED partials (o = sum exp(s-m) v, plus m and l) to the
+// workspace; a combine kernel reduces the splits.  SPLITS is chosen to
+// fill the 256 CUs (Hkv*B alone is 0.5 blocks/CU at common shapes).
 template <int G>
-__global__ void decode_attn_kernel(
+__global__ void decode_attn_split_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
-    const bf16* __restrict__ vc, bf16* __restrict__ out,
+    const bf16* __restrict__ vc,
+    float* __restrict__ ws_o,      // [B, Hkv, SPLITS, G, D]
+    float* __restrict__ ws_ml,     // [B, Hkv, SPLITS, G, 2]
     const int* __restrict__ seq_lens,
-    int H, int Hkv, int Smax, int D, float scale) {
+    int H, int Hkv, int Smax, int D, float scale, int splits) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* p_sh = reinterpret_cast<float*>(smem);            // [G][DEC_CHUNK]
   float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
@@ -291,7 +298,11 @@ __global__ void decode_attn_kernel(
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
+  const int split = blockIdx.z;
   const int slen = seq_lens[b];
+  const int span = (Smax + splits - 1) / splits;
+  const int s_begin = split * span;
+  const int s_end = min(slen, s_begin + span);
   const bf16* kb = kc + ((long)b * Hkv + hkv) * Smax * (long)D;
   const bf16* vb = vc + ((long)b * Hkv + hkv) * Smax * (long)D;
 
@@ -309,8 +320,8 @@ __global__ void decode_attn_kernel(
   for (int g = 0; g < G; ++g) o0[g] = 0.f;
   const int myd = threadIdx.x;          // dim owned in phase B (if < D)
 
-  for (int s0 = 0; s0 < slen; s0 += DEC_CHUNK) {
-    const int chunk = min(DEC_CHUNK, slen - s0);
+  for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
+    const int chunk = min(DEC_CHUNK, s_end - s0);
     // phase A: this thread's key, dots vs all G query heads
     float sc[G];
 #pragma unroll
@@ -370,12 +381,50 @@ __global__ void decode_attn_kernel(
     __syncthreads();
   }
 
+  // write unnormalized partials for this split
+  const long base = (((long)b * Hkv + hkv) * splits + split) * G;
   if (myd < D) {
 #pragma unroll
+    for (int g = 0; g < G; ++g)
+      ws_o[(base + g) * D + myd] = o0[g];
+  }
+  if (threadIdx.x == 0) {
+#pragma unroll
     for (int g = 0; g < G; ++g) {
-      const float denom = l_run[g] > 0.f ? l_run[g] : 1.f;
-      out[((long)b * H + hkv * G + g) * D + myd] = f2bf(o0[g] / denom);
+      ws_ml[(base + g) * 2 + 0] = m_run[g];
+      ws_ml[(base + g) * 2 + 1] = l_run[g];
     }
+  }
+}
+
+// combine: one block per (b, h); threads over D
+__global__ void decode_attn_combine_kernel(
+    const float* __restrict__ ws_o, const float* __restrict__ ws_ml,
+    bf16* __restrict__ out, int H, int G, int D, int splits) {
+  const int h = blockIdx.x;
+  const int b = blockIdx.y;
+  const int hkv = h / G;
+  const int g = h % G;
+  const long base0 = (((long)b * (H / G) + hkv) * splits) * G + g;
+  // global max over splits
+  float M = -INFINITY;
+  for (int s = 0; s < splits; ++s)
+    M = fmaxf(M, ws_ml[(base0 + (long)s * G) * 2 + 0]);
+  float den = 0.f;
+  for (int s = 0; s < splits; ++s) {
+    const float m = ws_ml[(base0 + (long)s * G) * 2 + 0];
+    const float l = ws_ml[(base0 + (long)s * G) * 2 + 1];
+    if (m != -INFINITY) den += l * __expf(m - M);
+  }
+  if (den <= 0.f) den = 1.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float num = 0.f;
+    for (int s = 0; s < splits; ++s) {
+      const float m = ws_ml[(base0 + (long)s * G) * 2 + 0];
+      if (m != -INFINITY)
+        num += ws_o[(base0 + (long)s * G) * D + d] * __expf(m - M);
+    }
+    out[((long)b * H + h) * D + d] = f2bf(num / den);
   }
 }
 
@@ -397,20 +446,21 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
 }
 
 hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
-                              void* out, const int* seq_lens, int B, int H,
-                              int Hkv, int Smax, int D, float scale,
-                              hipStream_t stream) {
+                              void* out, const int* seq_lens,
+                              float* ws_o, float* ws_ml, int splits,
+                              int B, int H, int Hkv, int Smax, int D,
+                              float scale, hipStream_t stream) {
   if (D > 256) return hipErrorInvalidValue;
   const int G = H / Hkv;
   if (G > DEC_MAXG || H % Hkv) return hipErrorInvalidValue;
   size_t lds = (size_t)(G * DEC_CHUNK + G * D + 32) * sizeof(float);
-  dim3 grid(Hkv, B);
+  dim3 grid(Hkv, B, splits);
 #define DEC_CASE(GV)                                                          \
   case GV:                                                                    \
-    hipLaunchKernelGGL((decode_attn_kernel<GV>), grid, dim3(DEC_CHUNK), lds,  \
-                       stream, (const bf16*)q, (const bf16*)kc,               \
-                       (const bf16*)vc, (bf16*)out, seq_lens, H, Hkv, Smax,   \
-                       D, scale);                                             \
+    hipLaunchKernelGGL((decode_attn_split_kernel<GV>), grid, dim3(DEC_CHUNK), \
+                       lds, stream, (const bf16*)q, (const bf16*)kc,          \
+                       (const bf16*)vc, ws_o, ws_ml, seq_lens, H, Hkv, Smax,  \
+                       D, scale, splits);                                     \
     break;
   switch (G) {
     DEC_CASE(1) DEC_CASE(2) DEC_CASE(3) DEC_CASE(4)
@@ -419,6 +469,9 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
       return hipErrorInvalidValue;
   }
 #undef DEC_CASE
+  HIP_CHECK_LAUNCH();
+  hipLaunchKernelGGL(decode_attn_combine_kernel, dim3(H, B), dim3(128), 0,
+                     stream, ws_o, ws_ml, (bf16*)out, H, G, D, splits);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

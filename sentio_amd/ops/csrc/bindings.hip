@@ -29,8 +29,8 @@ hipError_t sentio_flash_attn(const void*, const void*, const void*, void*,
                              const int*, int, int, int, int, int, float, int,
                              hipStream_t);
 hipError_t sentio_decode_attn(const void*, const void*, const void*, void*,
-                              const int*, int, int, int, int, int, float,
-                              hipStream_t);
+                              const int*, float*, float*, int, int, int, int,
+                              int, int, float, hipStream_t);
 hipError_t sentio_gemm_bf16(const void*, const void*, void*, int, int, int,
                             hipStream_t);
 }
@@ -201,11 +201,22 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   TORCH_CHECK(kc.dim() == 4, "k cache must be [B,Hkv,Smax,D]");
   const int B = q.size(0), H = q.size(1), D = q.size(2);
   const int Hkv = kc.size(1), Smax = kc.size(2);
+  const int G = H / Hkv;
   auto out = torch::empty_like(q);
   auto sl = seq_lens.to(torch::kInt).contiguous();
+  // split-S so the grid fills the chip: target >= 2 blocks per CU
+  int splits = (int)((512 + (long)B * Hkv - 1) / ((long)B * Hkv));
+  int max_splits = std::max(1, Smax / 256);
+  splits = std::max(1, std::min(splits, max_splits));
+  auto ws_o = torch::empty({(long)B * Hkv * splits * G * D},
+                           q.options().dtype(torch::kFloat));
+  auto ws_ml = torch::empty({(long)B * Hkv * splits * G * 2},
+                            q.options().dtype(torch::kFloat));
   check_hip(sentio_decode_attn(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                               out.data_ptr(), sl.data_ptr<int>(), B, H, Hkv,
-                               Smax, D, (float)scale, stream()), "decode_attn");
+                               out.data_ptr(), sl.data_ptr<int>(),
+                               ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
+                               splits, B, H, Hkv, Smax, D, (float)scale,
+                               stream()), "decode_attn");
   return out;
 }
 
