@@ -179,3 +179,26 @@ def _tp_worker(rank, world):
 
 def test_tp2_matches_tp1():
     _run_workers(_tp_worker)
+
+
+# ---------------- weight broadcast (DP bootstrap) ----------------
+
+def _broadcast_worker(rank, world):
+    import torch
+
+    from sentio_amd.engines.checkpoint import broadcast_weights
+    from sentio_amd.engines.configs import MODEL_CONFIGS
+    from sentio_amd.engines.transformer import Transformer
+
+    # each rank gets a DIFFERENT seed; after broadcast all match rank 0
+    m = Transformer(MODEL_CONFIGS["tiny-decoder64"], device="cpu",
+                    seed=100 + rank)
+    broadcast_weights(m, src=0)
+    ref = Transformer(MODEL_CONFIGS["tiny-decoder64"], device="cpu", seed=100)
+    assert torch.equal(m.w.tok_emb, ref.w.tok_emb)
+    assert torch.equal(m.w.layers[0]["w_down"], ref.w.layers[0]["w_down"])
+    assert torch.equal(m.w.lm_head, ref.w.lm_head)
+
+
+def test_broadcast_weights_syncs_replicas():
+    _run_workers(_broadcast_worker, world=2)
