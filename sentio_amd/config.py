@@ -81,8 +81,9 @@ class Settings:
     rate_limit_embed_per_min: int = 10
     max_query_len: int = 2000               # reference security.py InputValidator
     max_document_len: int = 50000
-    cache_backend: str = "memory"           # memory | multi_tier
+    cache_backend: str = "memory"           # memory | multi_tier | disk
     cache_ttl_s: float = 300.0
+    health_interval_s: float = 30.0         # reference patterns.py:252-306 loop
 
     # --- models / engines (MI355X-native; sizes are the bench ladder's) ---
     embedding_dim: int = 1024               # jina-v3 class (reference jina.py:23-27)

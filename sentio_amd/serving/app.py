@@ -109,7 +109,11 @@ def create_app(settings: Settings | None = None,
     @asynccontextmanager
     async def lifespan(app: FastAPI):
         container.initialize_all()
+        hc = container.health_checker()
+        hc.run_checks()
+        hc.start()
         yield
+        hc.stop()
 
     app = FastAPI(title="sentio-amd", version=__import__("sentio_amd").__version__,
                   lifespan=lifespan)

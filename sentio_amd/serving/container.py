@@ -135,6 +135,61 @@ class ServiceContainer:
             secret=self.settings.auth_secret,
             token_ttl_s=self.settings.auth_token_ttl_s))
 
+    def health_checker(self):
+        """Periodic probes over the GPU engines and indexes
+        (reference patterns.py:252-306 HealthChecker role)."""
+        def make():
+            from sentio_amd.resilience.health import HealthChecker
+
+            hc = HealthChecker(interval_s=self.settings.health_interval_s)
+
+            def check_encoder() -> bool:
+                v = self.encoder().embed(["health probe"])
+                return v.shape[-1] == self.settings.embedding_dim
+
+            def check_device() -> bool:
+                if self.device == "cpu":
+                    return True
+                import torch
+
+                free, total = torch.cuda.mem_get_info()
+                return free > 0
+
+            hc.register("encoder", check_encoder)
+            hc.register("device", check_device)
+            hc.register("dense_index", lambda: len(self.dense_index()) >= 0)
+            for name, br in self.breakers.items():
+                hc.register(f"breaker:{name}",
+                            lambda b=br: b.state.value != "open")
+            return hc
+
+        return self._get("health_checker", make)
+
+    # --- index snapshots (HBM→disk; reference persisted only a BM25 pickle,
+    # sparse.py:102-157 — here the whole index state checkpoints) ---
+    def save_indexes(self, directory: str) -> dict[str, Any]:
+        import os
+
+        os.makedirs(directory, exist_ok=True)
+        dense_path = os.path.join(directory, "dense.pt")
+        bm25_path = os.path.join(directory, "bm25.npz")
+        self.dense_index().save(dense_path)
+        self.bm25_index().save(bm25_path)
+        return {"dense": dense_path, "bm25": bm25_path,
+                "docs": len(self.dense_index())}
+
+    def load_indexes(self, directory: str) -> dict[str, Any]:
+        import os
+
+        with self._lock:
+            self._cache["dense_index"] = DenseIndex.load(
+                os.path.join(directory, "dense.pt"), device=self.device)
+            self._cache["bm25_index"] = BM25Index.load(
+                os.path.join(directory, "bm25.npz"))
+            for k in ("retriever", "pipeline", "ingestor"):
+                self._cache.pop(k, None)
+        return {"docs": len(self.dense_index())}
+
     def initialize_all(self) -> None:
         """Eager startup init in dependency order."""
         self.encoder()

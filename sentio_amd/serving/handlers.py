@@ -135,6 +135,7 @@ class HealthHandler:
             "status": status,
             "timestamp": now,
             "checks": checks,
+            "periodic": self.container.health_checker().status(),
             "resources": resource_monitor.snapshot(),
         }
         self._cache = (now, result)
