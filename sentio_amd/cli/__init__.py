@@ -66,6 +66,24 @@ def bench_run(steps: int = 5, warmup: int = 1, gpus: int = 1):
                     "--warmup", str(warmup), "--gpus", str(gpus)], check=True)
 
 
+index_app = typer.Typer(help="Index snapshots (HBM→disk save / load)")
+app.add_typer(index_app, name="index")
+
+
+@index_app.command("save")
+def index_save(directory: str = typer.Argument(...)):
+    from sentio_amd.serving.container import get_container
+
+    typer.echo(json.dumps(get_container().save_indexes(directory), indent=2))
+
+
+@index_app.command("load")
+def index_load(directory: str = typer.Argument(...)):
+    from sentio_amd.serving.container import get_container
+
+    typer.echo(json.dumps(get_container().load_indexes(directory), indent=2))
+
+
 @app.command("chat")
 def chat_once(question: str = typer.Argument(...)):
     from sentio_amd.serving.container import get_container

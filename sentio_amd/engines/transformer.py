@@ -198,17 +198,58 @@ class Transformer:
         cache: KVCache | None = None, kv_lens: torch.Tensor | None = None,
     ) -> torch.Tensor:
         """tokens: [B, S] int64 → hidden [B, S, dim] (after final norm).
-        kv_lens: right-padding valid lengths for non-causal batches."""
+        kv_lens: right-padding valid lengths for non-causal batches.
+        Residual adds are fused into the next block's RMSNorm
+        (ops.rmsnorm_residual) — one kernel instead of add + norm."""
         B, S = tokens.shape
         if pos is None:
             pos = torch.arange(S, device=tokens.device).unsqueeze(0).expand(B, S)
         x = self.w.tok_emb[tokens.reshape(-1)].view(B, S, self.cfg.dim)
-        for i, layer in enumerate(self.w.layers):
-            h = ops.rmsnorm(x, layer["attn_norm"], self.cfg.norm_eps)
-            x = x + self._attn(h, layer, pos, cache, i, kv_lens)
-            h = ops.rmsnorm(x, layer["ffn_norm"], self.cfg.norm_eps)
-            x = x + self._ffn(h, layer)
-        return ops.rmsnorm(x, self.w.final_norm, self.cfg.norm_eps)
+        eps = self.cfg.norm_eps
+        layers = self.w.layers
+        n = len(layers)
+        normed = ops.rmsnorm(x, layers[0]["attn_norm"], eps)
+        for i, layer in enumerate(layers):
+            a = self._attn(normed, layer, pos, cache, i, kv_lens)
+            normed, x = ops.rmsnorm_residual(a, x, layer["ffn_norm"], eps)
+            f = self._ffn(normed, layer)
+            w_next = layers[i + 1]["attn_norm"] if i + 1 < n else self.w.final_norm
+            normed, x = ops.rmsnorm_residual(f, x, w_next, eps)
+        return normed
+
+    # ----- decode fast path: fused RoPE + KV-cache write + fused norms -----
+    def _attn_decode(self, normed: torch.Tensor, layer: dict, cache: KVCache,
+                     layer_idx: int, attn_lens: torch.Tensor) -> torch.Tensor:
+        B = normed.shape[0]
+        d = self.cfg.dim
+        hd = self.cfg.head_dim
+        qkv = normed.view(B, d) @ layer["wqkv"]
+        q = ops.decode_qkv_prep(qkv, cache.k[layer_idx], cache.v[layer_idx],
+                                self.rope_cos, self.rope_sin, cache.seq_lens)
+        out = ops.decode_attention(q, cache.k[layer_idx], cache.v[layer_idx],
+                                   attn_lens, self.scale)
+        out = out.view(B, self.h_local * hd) @ layer["wo"]
+        out = self.tp.all_reduce(out)
+        return out.view(B, 1, d)
+
+    def forward_decode(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
+        """One-token decode: tokens [B, 1] → hidden [B, 1, dim].  Uses the
+        fused decode_qkv_prep kernel (RoPE + cache write in one launch) and
+        advances cache.seq_lens on device (hipGraph-capturable)."""
+        B = tokens.shape[0]
+        x = self.w.tok_emb[tokens.reshape(-1)].view(B, 1, self.cfg.dim)
+        eps = self.cfg.norm_eps
+        layers = self.w.layers
+        n = len(layers)
+        attn_lens = cache.seq_lens + 1
+        normed = ops.rmsnorm(x, layers[0]["attn_norm"], eps)
+        for i, layer in enumerate(layers):
+            a = self._attn_decode(normed, layer, cache, i, attn_lens)
+            normed, x = ops.rmsnorm_residual(a, x, layer["ffn_norm"], eps)
+            f = self._ffn(normed, layer)
+            w_next = layers[i + 1]["attn_norm"] if i + 1 < n else self.w.final_norm
+            normed, x = ops.rmsnorm_residual(f, x, w_next, eps)
+        return normed
 
     # ----- decoder-specific -----
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
@@ -224,8 +265,6 @@ class Transformer:
 
     def decode_step(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
         """tokens: [B, 1] the latest sampled token; returns logits [B, V]."""
-        B = tokens.shape[0]
-        pos = cache.seq_lens.long().unsqueeze(1)  # [B,1] current position
-        hidden = self.forward_hidden(tokens, pos=pos, cache=cache)
+        hidden = self.forward_decode(tokens, cache)
         cache.seq_lens += 1
         return self.logits(hidden)

@@ -75,6 +75,16 @@ def rope_apply(x, cos, sin, pos):
     return torch_ref.rope_apply(x, cos, sin, pos)
 
 
+def decode_qkv_prep(qkv, k_cache, v_cache, cos, sin, seq_lens):
+    """Fused decode-token head prep: RoPE(q), RoPE(k)→cache, v→cache.
+    qkv: [B, (H+2*Hkv)*D] raw projection; seq_lens: [B] i32 current position.
+    Returns q [B, H, D]; writes k/v rows in place."""
+    if _on_gpu(qkv):
+        return _require_hip().decode_qkv_prep(
+            qkv.contiguous(), k_cache, v_cache, cos, sin, seq_lens)
+    return torch_ref.decode_qkv_prep(qkv, k_cache, v_cache, cos, sin, seq_lens)
+
+
 def swiglu(gate, up):
     if _on_gpu(gate):
         return _require_hip().swiglu(gate.contiguous(), up.contiguous())

@@ -295,6 +295,7 @@ __global__ void decode_attn_split_kernel(
   float* p_sh = reinterpret_cast<float*>(smem);            // [G][DEC_CHUNK]
   float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
   float* red = q_sh + G * D;                               // [32] scratch
+  float* o_sh = red + 32;                                  // [G][D] final reduce
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
@@ -309,16 +310,28 @@ __global__ void decode_attn_split_kernel(
   for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
     const int g = i / D, d = i % D;
     q_sh[i] = bf2f(q[((long)b * H + hkv * G + g) * D + d]);
+    o_sh[i] = 0.f;
   }
   __syncthreads();
 
   float m_run[G], l_run[G], alpha[G];
 #pragma unroll
   for (int g = 0; g < G; ++g) { m_run[g] = -INFINITY; l_run[g] = 0.f; }
-  float o0[G];
+
+  // phase-B layout: all 256 threads active — thread owns 8 dims (dgroup) of
+  // every 16th key row (jslot).  16-byte V loads; 16 lanes cover a 256B row,
+  // a wave streams 4 rows, the block keeps 16 rows in flight.  Per-thread
+  // partial O accumulates across ALL chunks (rescaled by alpha like the
+  // softmax state) and is LDS-reduced once at the end.
+  const int dgroup = threadIdx.x & 15;       // owns dims dgroup*8 .. +7
+  const int jslot = threadIdx.x >> 4;        // keys j ≡ jslot (mod 16)
+  const int dgD = D / 8;                     // dgroups that exist (D=128 → 16)
+  const bool dg_ok = dgroup < dgD;
+  float o_part[G][8];
 #pragma unroll
-  for (int g = 0; g < G; ++g) o0[g] = 0.f;
-  const int myd = threadIdx.x;          // dim owned in phase B (if < D)
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o_part[g][e] = 0.f;
 
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
@@ -364,29 +377,46 @@ __global__ void decode_attn_split_kernel(
       m_run[g] = m_new;
     }
     __syncthreads();
-    // phase B: PV — V row element read ONCE, fed to all G heads
-    if (myd < D) {
-      float acc[G];
+    // phase B: PV.  v loads are bf16x8 (16 B) and fully coalesced; the
+    // p_sh reads broadcast (16 lanes share one address).
 #pragma unroll
-      for (int g = 0; g < G; ++g) acc[g] = 0.f;
-      for (int j = 0; j < chunk; ++j) {
-        const float vv = bf2f(vb[(long)(s0 + j) * D + myd]);
+    for (int g = 0; g < G; ++g)
 #pragma unroll
-        for (int g = 0; g < G; ++g)
-          acc[g] += p_sh[g * DEC_CHUNK + j] * vv;
+      for (int e = 0; e < 8; ++e) o_part[g][e] *= alpha[g];
+    if (dg_ok) {
+      for (int j = jslot; j < chunk; j += 16) {
+        bf16x8 v8 = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
+            dgroup * 8);
+        float vf[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[e]);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float p = p_sh[g * DEC_CHUNK + j];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+        }
       }
-#pragma unroll
-      for (int g = 0; g < G; ++g) o0[g] = o0[g] * alpha[g] + acc[g];
     }
     __syncthreads();
   }
 
-  // write unnormalized partials for this split
-  const long base = (((long)b * Hkv + hkv) * splits + split) * G;
-  if (myd < D) {
+  // reduce the 16 jslot partials per (g, dim) through LDS float atomics
+  if (dg_ok) {
 #pragma unroll
     for (int g = 0; g < G; ++g)
-      ws_o[(base + g) * D + myd] = o0[g];
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        atomicAdd(&o_sh[g * D + dgroup * 8 + e], o_part[g][e]);
+  }
+  __syncthreads();
+
+  // write unnormalized partials for this split
+  const long base = (((long)b * Hkv + hkv) * splits + split) * G;
+  for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
+    const int g = i / D, d = i % D;
+    ws_o[(base + g) * D + d] = o_sh[i];
   }
   if (threadIdx.x == 0) {
 #pragma unroll
@@ -450,10 +480,10 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
                               float* ws_o, float* ws_ml, int splits,
                               int B, int H, int Hkv, int Smax, int D,
                               float scale, hipStream_t stream) {
-  if (D > 256) return hipErrorInvalidValue;
+  if (D > 256 || (D % 8)) return hipErrorInvalidValue;
   const int G = H / Hkv;
   if (G > DEC_MAXG || H % Hkv) return hipErrorInvalidValue;
-  size_t lds = (size_t)(G * DEC_CHUNK + G * D + 32) * sizeof(float);
+  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float);
   dim3 grid(Hkv, B, splits);
 #define DEC_CASE(GV)                                                          \
   case GV:                                                                    \

@@ -14,6 +14,9 @@ hipError_t sentio_swiglu(const void*, const void*, void*, long, hipStream_t);
 hipError_t sentio_rope(const void*, void*, const float*, const float*,
                        const int*, int, int, int, int, hipStream_t);
 hipError_t sentio_softmax(const void*, void*, long, int, hipStream_t);
+hipError_t sentio_decode_qkv_prep(const void*, void*, void*, void*,
+                                  const float*, const float*, const int*, int,
+                                  int, int, int, int, hipStream_t);
 hipError_t sentio_mean_pool_l2norm(const void*, const unsigned char*, float*,
                                    int, int, int, hipStream_t);
 hipError_t sentio_sample(const float*, long*, int, int, float, unsigned,
@@ -96,6 +99,30 @@ torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cosT, torch::Tensor sinT
                         pos_i.data_ptr<int>(), x.size(0), x.size(1), x.size(2),
                         x.size(3), stream()), "rope");
   return y;
+}
+
+torch::Tensor decode_qkv_prep(torch::Tensor qkv, torch::Tensor kc,
+                              torch::Tensor vc, torch::Tensor cosT,
+                              torch::Tensor sinT, torch::Tensor seq_lens) {
+  check_bf16_cuda(qkv, "qkv");
+  TORCH_CHECK(qkv.dim() == 2, "qkv must be [B, (H+2*Hkv)*D]");
+  TORCH_CHECK(kc.dim() == 4 && vc.dim() == 4, "caches must be [B,Hkv,Smax,D]");
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt, "seq_lens must be i32");
+  const int B = qkv.size(0);
+  const int Hkv = kc.size(1);
+  const int Smax = kc.size(2);
+  const int D = kc.size(3);
+  const int H = (int)(qkv.size(1) / D) - 2 * Hkv;
+  TORCH_CHECK(H > 0 && (long)(H + 2 * Hkv) * D == qkv.size(1),
+              "qkv width mismatch");
+  auto q_out = torch::empty({B, H, D}, qkv.options());
+  check_hip(sentio_decode_qkv_prep(qkv.data_ptr(), q_out.data_ptr(),
+                                   kc.data_ptr(), vc.data_ptr(),
+                                   cosT.data_ptr<float>(),
+                                   sinT.data_ptr<float>(),
+                                   seq_lens.data_ptr<int>(), B, H, Hkv, D,
+                                   Smax, stream()), "decode_qkv_prep");
+  return q_out;
 }
 
 torch::Tensor softmax_lastdim(torch::Tensor x) {
@@ -238,6 +265,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_residual", &rmsnorm_residual);
   m.def("swiglu", &swiglu);
   m.def("rope_apply", &rope_apply);
+  m.def("decode_qkv_prep", &decode_qkv_prep);
   m.def("softmax_lastdim", &softmax_lastdim);
   m.def("mean_pool_l2norm", &mean_pool_l2norm);
   m.def("sample_token", &sample_token);

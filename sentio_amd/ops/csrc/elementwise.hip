@@ -152,6 +152,42 @@ __global__ void rope_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
   }
 }
 
+// ------------------------------------------------------- decode qkv prep
+// Fused per-token decode head prep: input is the raw QKV projection
+// [B, (H+2*Hkv)*D] for S=1; this kernel applies RoPE to the q and k heads
+// and scatters k/v into the KV caches at row seq_lens[b] — replacing a
+// rope launch for q, a rope launch for k, and two indexed cache-write
+// launches per layer per token (the decode step's tiny-kernel tail).
+// Grid: (H + 2*Hkv, B); block: 128 threads.
+__global__ void decode_qkv_prep_kernel(
+    const bf16* __restrict__ qkv, bf16* __restrict__ q_out,
+    bf16* __restrict__ kc, bf16* __restrict__ vc,
+    const float* __restrict__ cosT, const float* __restrict__ sinT,
+    const int* __restrict__ seq_lens, int H, int Hkv, int D, int Smax) {
+  const int b = blockIdx.y;
+  const int h = blockIdx.x;            // 0..H+2*Hkv-1
+  const int pos = seq_lens[b];
+  const int half = D / 2;
+  const bf16* src = qkv + ((long)b * (H + 2 * Hkv) + h) * D;
+
+  if (h < H + Hkv) {                   // q or k head: rotate
+    bf16* dst = (h < H)
+        ? q_out + ((long)b * H + h) * D
+        : kc + (((long)b * Hkv + (h - H)) * Smax + pos) * (long)D;
+    for (int p = threadIdx.x; p < half; p += blockDim.x) {
+      const float c = cosT[(long)pos * half + p];
+      const float sn = sinT[(long)pos * half + p];
+      const float x0 = bf2f(src[2 * p]);
+      const float x1 = bf2f(src[2 * p + 1]);
+      dst[2 * p] = f2bf(x0 * c - x1 * sn);
+      dst[2 * p + 1] = f2bf(x0 * sn + x1 * c);
+    }
+  } else {                             // v head: copy into cache
+    bf16* dst = vc + (((long)b * Hkv + (h - H - Hkv)) * Smax + pos) * (long)D;
+    for (int d = threadIdx.x; d < D; d += blockDim.x) dst[d] = src[d];
+  }
+}
+
 // ---------------------------------------------------------------- softmax
 // x [N, D] bf16 -> y [N, D] bf16, row-wise, online in two block passes.
 __global__ void softmax_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
@@ -310,6 +346,19 @@ hipError_t sentio_rope(const void* x, void* y, const float* cosT,
   if (blocks > 2048) blocks = 2048;
   hipLaunchKernelGGL(rope_kernel, dim3((unsigned)blocks), dim3(256), 0, stream,
                      (const bf16*)x, (bf16*)y, cosT, sinT, pos, S, H, D, total);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sentio_decode_qkv_prep(const void* qkv, void* q_out, void* kc,
+                                  void* vc, const float* cosT,
+                                  const float* sinT, const int* seq_lens,
+                                  int B, int H, int Hkv, int D, int Smax,
+                                  hipStream_t stream) {
+  if (D % 2) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(decode_qkv_prep_kernel, dim3(H + 2 * Hkv, B), dim3(128),
+                     0, stream, (const bf16*)qkv, (bf16*)q_out, (bf16*)kc,
+                     (bf16*)vc, cosT, sinT, seq_lens, H, Hkv, D, Smax);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

@@ -189,3 +189,25 @@ def sample_token(
         return logits.float().argmax(dim=-1)
     probs = torch.softmax(logits.float() / temperature, dim=-1)
     return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+
+
+def decode_qkv_prep(
+    qkv: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+    cos: torch.Tensor, sin: torch.Tensor, seq_lens: torch.Tensor
+) -> torch.Tensor:
+    """CPU reference of the fused decode head prep (see ops.decode_qkv_prep):
+    splits the raw S=1 QKV projection, RoPE-rotates q and k, writes k/v into
+    the caches at row seq_lens[b], returns q [B, H, D]."""
+    B = qkv.shape[0]
+    Hkv, smax, D = k_cache.shape[1], k_cache.shape[2], k_cache.shape[3]
+    H = qkv.shape[1] // D - 2 * Hkv
+    pos = seq_lens.long().view(B, 1)
+    q = qkv[:, : H * D].view(B, 1, H, D)
+    k = qkv[:, H * D : (H + Hkv) * D].view(B, 1, Hkv, D)
+    v = qkv[:, (H + Hkv) * D :].view(B, 1, Hkv, D)
+    q = rope_apply(q, cos, sin, pos)
+    k = rope_apply(k, cos, sin, pos)
+    b_idx = torch.arange(B)
+    k_cache[b_idx, :, seq_lens.long()] = k.view(B, Hkv, D).to(k_cache.dtype)
+    v_cache[b_idx, :, seq_lens.long()] = v.view(B, Hkv, D).to(v_cache.dtype)
+    return q.view(B, H, D)
