@@ -187,8 +187,7 @@ class Transformer:
     def _ffn(self, x: torch.Tensor, layer: dict) -> torch.Tensor:
         B, S, d = x.shape
         gu = x.view(B * S, d) @ layer["w_gate_up"]
-        f = self.cfg.ffn_dim // self.tp.world
-        y = ops.swiglu(gu[:, :f].contiguous(), gu[:, f:].contiguous())
+        y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
         out = y @ layer["w_down"]
         out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
         return out.view(B, S, d)

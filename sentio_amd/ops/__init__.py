@@ -91,6 +91,13 @@ def swiglu(gate, up):
     return torch_ref.swiglu(gate, up)
 
 
+def swiglu_packed(gu):
+    """gu [..., 2F] packed [gate | up] → silu(gate) * up [..., F]."""
+    if _on_gpu(gu):
+        return _require_hip().swiglu_packed(gu.contiguous())
+    return torch_ref.swiglu_packed(gu)
+
+
 def softmax(x, dim: int = -1):
     if _on_gpu(x):
         if dim not in (-1, x.ndim - 1):
@@ -179,7 +186,7 @@ def gemm_bf16(a, b):
 
 
 __all__ = [
-    "rmsnorm", "rmsnorm_residual", "rope_apply", "swiglu", "softmax",
+    "rmsnorm", "rmsnorm_residual", "rope_apply", "decode_qkv_prep", "swiglu", "swiglu_packed", "softmax",
     "attention", "decode_attention", "mean_pool_l2norm", "cosine_topk",
     "bm25_score", "sample_token", "gemm_bf16", "hip_available", "torch_ref",
 ]

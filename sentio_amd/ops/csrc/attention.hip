@@ -335,42 +335,57 @@ __global__ void decode_attn_split_kernel(
 
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    // phase A: this thread's key, dots vs all G query heads
-    float sc[G];
+    // phase A: (jslot, dgroup) layout — 16 lanes read one K row as
+    // contiguous bf16x8 chunks (full 256 B lines; the old thread-per-key
+    // layout issued 16 B loads at 256 B stride and thrashed L1), partial
+    // dots reduced across the 16-lane group with 4 shfl_xor steps.
+    for (int it = 0; it < DEC_CHUNK / 16; ++it) {
+      const int jj = it * 16 + jslot;          // key index within chunk
+      float part[G];
 #pragma unroll
-    for (int g = 0; g < G; ++g) sc[g] = -INFINITY;
-    const int s = s0 + threadIdx.x;
-    if (threadIdx.x < chunk) {
-      const bf16* krow = kb + (long)s * D;
-#pragma unroll
-      for (int g = 0; g < G; ++g) sc[g] = 0.f;
-      for (int d = 0; d < D; d += 8) {
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const short*>(krow) + d);
+      for (int g = 0; g < G; ++g) part[g] = 0.f;
+      if (jj < chunk && dg_ok) {
+        bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(kb + (long)(s0 + jj) * D) +
+            dgroup * 8);
         float kf[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) kf[j] = bits2f(kv8[j]);
+        for (int e = 0; e < 8; ++e) kf[e] = bits2f(k8[e]);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
           float acc = 0.f;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) acc += kf[j] * q_sh[g * D + d + j];
-          sc[g] += acc;
+          for (int e = 0; e < 8; ++e)
+            acc += kf[e] * q_sh[g * D + dgroup * 8 + e];
+          part[g] = acc;
         }
       }
+      // reduce the 16 dgroup partials (lanes l ^ 1,2,4,8 share a key)
 #pragma unroll
-      for (int g = 0; g < G; ++g) sc[g] *= scale;
+      for (int g = 0; g < G; ++g) {
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          part[g] += __shfl_xor(part[g], off, WAVE);
+      }
+      if (dgroup == 0) {
+#pragma unroll
+        for (int g = 0; g < G; ++g)
+          p_sh[g * DEC_CHUNK + jj] = (jj < chunk) ? part[g] * scale
+                                                  : -INFINITY;
+      }
     }
-    // per-head block max/sum + p staging
+    __syncthreads();
+    // per-head block max/sum + in-place exp (scores → p in p_sh)
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-      float cmax = block_max(sc[g], red);
+      const float sc = p_sh[g * DEC_CHUNK + threadIdx.x];
+      const bool ok = threadIdx.x < chunk;
+      float cmax = block_max(ok ? sc : -INFINITY, red);
       float m_new = fmaxf(m_run[g], cmax);
       float a = (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
       if (m_new == -INFINITY) a = 1.f;
       alpha[g] = a;
-      float p = (threadIdx.x < chunk && sc[g] != -INFINITY)
-                    ? __expf(sc[g] - m_new) : 0.f;
+      float p = (ok && sc != -INFINITY) ? __expf(sc - m_new) : 0.f;
       p_sh[g * DEC_CHUNK + threadIdx.x] = p;
       float csum = block_sum(p, red);
       l_run[g] = l_run[g] * a + csum;

@@ -11,6 +11,7 @@ hipError_t sentio_rmsnorm(const void*, const void*, void*, long, int, float,
 hipError_t sentio_rmsnorm_residual(const void*, const void*, const void*,
                                    void*, void*, long, int, float, hipStream_t);
 hipError_t sentio_swiglu(const void*, const void*, void*, long, hipStream_t);
+hipError_t sentio_swiglu_packed(const void*, void*, long, int, hipStream_t);
 hipError_t sentio_rope(const void*, void*, const float*, const float*,
                        const int*, int, int, int, int, hipStream_t);
 hipError_t sentio_softmax(const void*, void*, long, int, hipStream_t);
@@ -84,6 +85,19 @@ torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
   auto out = torch::empty_like(gate);
   check_hip(sentio_swiglu(gate.data_ptr(), up.data_ptr(), out.data_ptr(),
                           gate.numel(), stream()), "swiglu");
+  return out;
+}
+
+torch::Tensor swiglu_packed(torch::Tensor gu) {
+  check_bf16_cuda(gu, "gu");
+  const int F2 = gu.size(-1);
+  TORCH_CHECK(F2 % 16 == 0, "packed width must be divisible by 16");
+  const long rows = gu.numel() / F2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = F2 / 2;
+  auto out = torch::empty(sizes, gu.options());
+  check_hip(sentio_swiglu_packed(gu.data_ptr(), out.data_ptr(), rows, F2 / 2,
+                                 stream()), "swiglu_packed");
   return out;
 }
 
@@ -264,6 +278,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_residual", &rmsnorm_residual);
   m.def("swiglu", &swiglu);
+  m.def("swiglu_packed", &swiglu_packed);
   m.def("rope_apply", &rope_apply);
   m.def("decode_qkv_prep", &decode_qkv_prep);
   m.def("softmax_lastdim", &softmax_lastdim);

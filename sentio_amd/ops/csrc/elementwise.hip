@@ -115,6 +115,29 @@ __global__ void swiglu_kernel(const bf16* __restrict__ gate,
     reinterpret_cast<bf16x8*>(out)[i] = o;
   }
 }
+// Packed variant: gu [N, 2F] rows = [gate | up]; out [N, F].
+// Avoids the two strided .contiguous() copies per FFN invocation.
+__global__ void swiglu_packed_kernel(const bf16* __restrict__ gu,
+                                     bf16* __restrict__ out, long rows, int F) {
+  const int f8 = F / 8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+       i < rows * (long)f8; i += (long)gridDim.x * blockDim.x) {
+    const long r = i / f8;
+    const int c8 = (int)(i % f8);
+    const bf16* row = gu + r * (long)(2 * F);
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(row + c8 * 8);
+    bf16x8 u = *reinterpret_cast<const bf16x8*>(row + F + c8 * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bits2f(g[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o[j] = f2bits(s * bits2f(u[j]));
+    }
+    *reinterpret_cast<bf16x8*>(out + r * (long)F + c8 * 8) = o;
+  }
+}
+
 __global__ void swiglu_tail_kernel(const bf16* __restrict__ gate,
                                    const bf16* __restrict__ up,
                                    bf16* __restrict__ out, long start, long n) {
@@ -346,6 +369,18 @@ hipError_t sentio_rope(const void* x, void* y, const float* cosT,
   if (blocks > 2048) blocks = 2048;
   hipLaunchKernelGGL(rope_kernel, dim3((unsigned)blocks), dim3(256), 0, stream,
                      (const bf16*)x, (bf16*)y, cosT, sinT, pos, S, H, D, total);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sentio_swiglu_packed(const void* gu, void* out, long rows, int F,
+                                hipStream_t stream) {
+  if (F % 8) return hipErrorInvalidValue;
+  long total = rows * (long)(F / 8);
+  long blocks = (total + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(swiglu_packed_kernel, dim3((unsigned)blocks), dim3(256),
+                     0, stream, (const bf16*)gu, (bf16*)out, rows, F);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

@@ -211,3 +211,9 @@ def decode_qkv_prep(
     k_cache[b_idx, :, seq_lens.long()] = k.view(B, Hkv, D).to(k_cache.dtype)
     v_cache[b_idx, :, seq_lens.long()] = v.view(B, Hkv, D).to(v_cache.dtype)
     return q.view(B, H, D)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+    """gu [..., 2F] rows packed [gate | up] → silu(gate) * up, [..., F]."""
+    f = gu.shape[-1] // 2
+    return swiglu(gu[..., :f], gu[..., f:])

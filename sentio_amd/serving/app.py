@@ -13,7 +13,12 @@ from contextlib import asynccontextmanager
 from typing import Any
 
 from fastapi import Depends, FastAPI, HTTPException, Request
-from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
+from fastapi.responses import (
+    HTMLResponse,
+    JSONResponse,
+    PlainTextResponse,
+    StreamingResponse,
+)
 from pydantic import BaseModel, Field, field_validator
 
 from sentio_amd.config import Settings
@@ -243,6 +248,12 @@ def create_app(settings: Settings | None = None,
             "metrics": metrics_collector.snapshot(),
             "monitors": performance_monitor.all_summaries(),
         }
+
+    @app.get("/ui")
+    async def ui_page():
+        from sentio_amd.serving.ui import UI_HTML
+
+        return HTMLResponse(UI_HTML)
 
     return app
 

@@ -120,3 +120,9 @@ def test_chat_stream(client):
         assert r.status_code == 200
         text = "".join(r.iter_text())
     assert "data:" in text and "[DONE]" in text
+
+
+def test_ui_page(client):
+    r = client.get("/ui")
+    assert r.status_code == 200
+    assert "sentio-amd" in r.text and "/chat" in r.text

@@ -324,3 +324,13 @@ def test_forward_decode_matches_prefill(dev):
     # bf16 path tolerance: top-1 must match, logits close
     assert (top_a[:, 0] == top_b[:, 0]).all()
     torch.testing.assert_close(logits_all, logits_dec, rtol=5e-2, atol=5e-1)
+
+
+def test_swiglu_packed(dev):
+    from sentio_amd import ops
+
+    torch.manual_seed(3)
+    gu = torch.randn(64, 512, dtype=torch.bfloat16, device=dev)
+    got = ops.swiglu_packed(gu)
+    want = ops.torch_ref.swiglu_packed(gu.cpu().float())
+    _cmp(got, want, rtol=2e-2, atol=2e-2)
