@@ -230,10 +230,22 @@ def main():
                 f"term{terms[2]} term{terms[3]} retrieval step {step}?")
         return out
 
+    stage_t: dict[str, float] = {}
+
+    def _mark(name: str, t0: float) -> float:
+        """Accumulate per-stage wall time (sync'd) for the stderr breakdown."""
+        if on_gpu:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        stage_t[name] = stage_t.get(name, 0.0) + (t1 - t0)
+        return t1
+
     def chat_step(step: int) -> None:
+        t0 = time.perf_counter()
         queries = make_queries(step)
         # 1. embed queries (one encoder batch)
         qv = encoder.embed(queries)
+        t0 = _mark("embed", t0)
         # 2. dense search over ALL ranks' shards (SPMD all-gather)
         if world > 1:
             q_all = D.all_gather_tensor(qv)
@@ -241,6 +253,7 @@ def main():
             q_all = qv
         vals, idx = dense_search_ids(dense, q_all, args.top_k)
         vals = vals.cpu(); idx = idx.cpu()
+        t0 = _mark("dense", t0)
         if world > 1:
             gathered_v = D.all_gather_objects(vals.numpy())
             gathered_i = D.all_gather_objects(idx.numpy())
@@ -258,6 +271,7 @@ def main():
             sparse_all = D.all_gather_objects(sparse_hits_local)
         else:
             sparse_all = [sparse_hits_local]
+        t0 = _mark("bm25", t0)
 
         # 4. per-query merge + fusion (owner = this rank's queries)
         base = rank * args.batch
@@ -283,6 +297,7 @@ def main():
             for d, (_, s) in zip(docs, fused):
                 d.metadata["score"] = s
             batch_docs.append(docs)
+        t0 = _mark("fuse", t0)
 
         # 5. rerank all queries' candidates in one cross-encoder batch
         pair_texts = []
@@ -304,6 +319,7 @@ def main():
                 cand[i].metadata["score"] = float(sc[i])
                 keep.append(cand[i])
             reranked.append(keep)
+        t0 = _mark("rerank", t0)
 
         # 6. select + prompts
         prompts = []
@@ -313,8 +329,10 @@ def main():
             prompts.append(builder.system_prompt() + "\n\n" +
                            builder.build_qa_prompt(queries[qi], ctx))
         # 7. batched generation
+        t0 = _mark("select", t0)
         answers = generator.generate(prompts, max_new_tokens=args.gen_tokens,
                                      temperature=0.3, stop_on_eos=False)
+        t0 = _mark("generate", t0)
         # 8. optional verify
         if args.verify:
             vprompts = [builder.build_verify_prompt(
@@ -352,6 +370,9 @@ def main():
     ms_per_step = elapsed / args.steps * 1e3
 
     if rank == 0:
+        per_step = {k: round(v / (args.steps + args.warmup) * 1e3, 1)
+                    for k, v in sorted(stage_t.items(), key=lambda kv: -kv[1])}
+        print(f"[stage ms/step] {per_step}", file=sys.stderr)
         result = {
             "metric": "chat_qps",
             "value": round(qps, 3),
