@@ -39,7 +39,7 @@ class RerankerEngine:
             kv_lens = torch.tensor(lens, dtype=torch.int32, device=self.device)
             hidden = self.model.forward_hidden(tokens, kv_lens=kv_lens)
             pooled = hidden[:, 0, :]  # first-token pooled representation
-            s = (pooled @ self.model.w.head).float().squeeze(-1)
+            s = torch.nn.functional.linear(pooled, self.model.w.head).float().squeeze(-1)
             scores.extend(torch.sigmoid(s).cpu().tolist())
         return scores
 
