@@ -296,6 +296,7 @@ __global__ void decode_attn_split_kernel(
   float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
   float* red = q_sh + G * D;                               // [32] scratch
   float* o_sh = red + 32;                                  // [G][D] final reduce
+  char* k_lds = reinterpret_cast<char*>(o_sh + G * D);     // [DEC_CHUNK][D] bf16 swizzled
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
@@ -333,46 +334,49 @@ __global__ void decode_attn_split_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) o_part[g][e] = 0.f;
 
+  const int Dbytes = D * 2;
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    // phase A: (jslot, dgroup) layout — 16 lanes read one K row as
-    // contiguous bf16x8 chunks (full 256 B lines; the old thread-per-key
-    // layout issued 16 B loads at 256 B stride and thrashed L1), partial
-    // dots reduced across the 16-lane group with 4 shfl_xor steps.
-    for (int it = 0; it < DEC_CHUNK / 16; ++it) {
-      const int jj = it * 16 + jslot;          // key index within chunk
-      float part[G];
+    // phase A1: cooperative coalesced K staging into XOR-swizzled LDS
+    // (256 threads issue independent bf16x8 global loads — full lines and
+    // deep memory-level parallelism; the swizzle kills the 16-way bank
+    // conflict a row-major [key][D] layout would give phase A2).
+    {
+      const int elems = chunk * D;
+      for (int i = threadIdx.x * 8; i < elems; i += DEC_CHUNK * 8) {
+        const int row = i / D;
+        const int d = i % D;
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(kb + (long)(s0 + row) * D) + d);
+        *reinterpret_cast<bf16x8*>(
+            k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = v;
+      }
+    }
+    __syncthreads();
+    // phase A2: thread-per-key dot vs all G query heads, K read from LDS
+    {
+      float sc[G];
 #pragma unroll
-      for (int g = 0; g < G; ++g) part[g] = 0.f;
-      if (jj < chunk && dg_ok) {
-        bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const short*>(kb + (long)(s0 + jj) * D) +
-            dgroup * 8);
-        float kf[8];
+      for (int g = 0; g < G; ++g) sc[g] = 0.f;
+      const int row = threadIdx.x;
+      if (row < chunk) {
+        for (int d = 0; d < D; d += 8) {
+          bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
+              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
 #pragma unroll
-        for (int e = 0; e < 8; ++e) kf[e] = bits2f(k8[e]);
+          for (int g = 0; g < G; ++g) {
+            float acc = 0.f;
 #pragma unroll
-        for (int g = 0; g < G; ++g) {
-          float acc = 0.f;
-#pragma unroll
-          for (int e = 0; e < 8; ++e)
-            acc += kf[e] * q_sh[g * D + dgroup * 8 + e];
-          part[g] = acc;
+            for (int e = 0; e < 8; ++e)
+              acc += bits2f(k8[e]) * q_sh[g * D + d + e];
+            sc[g] += acc;
+          }
         }
       }
-      // reduce the 16 dgroup partials (lanes l ^ 1,2,4,8 share a key)
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1)
-          part[g] += __shfl_xor(part[g], off, WAVE);
-      }
-      if (dgroup == 0) {
-#pragma unroll
-        for (int g = 0; g < G; ++g)
-          p_sh[g * DEC_CHUNK + jj] = (jj < chunk) ? part[g] * scale
-                                                  : -INFINITY;
-      }
+      for (int g = 0; g < G; ++g)
+        p_sh[g * DEC_CHUNK + threadIdx.x] =
+            (row < chunk) ? sc[g] * scale : -INFINITY;
     }
     __syncthreads();
     // per-head block max/sum + in-place exp (scores → p in p_sh)
@@ -498,7 +502,8 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
   if (D > 256 || (D % 8)) return hipErrorInvalidValue;
   const int G = H / Hkv;
   if (G > DEC_MAXG || H % Hkv) return hipErrorInvalidValue;
-  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float);
+  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float)
+               + (size_t)DEC_CHUNK * D * 2;   // swizzled K stage
   dim3 grid(Hkv, B, splits);
 #define DEC_CASE(GV)                                                          \
   case GV:                                                                    \

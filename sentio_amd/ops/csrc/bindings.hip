@@ -20,8 +20,8 @@ hipError_t sentio_decode_qkv_prep(const void*, void*, void*, void*,
                                   int, int, int, int, hipStream_t);
 hipError_t sentio_mean_pool_l2norm(const void*, const unsigned char*, float*,
                                    int, int, int, hipStream_t);
-hipError_t sentio_sample(const float*, long*, int, int, float, unsigned,
-                         hipStream_t);
+hipError_t sentio_sample(const float*, long*, float*, int*, int, int, float,
+                         unsigned, hipStream_t);
 hipError_t sentio_cosine_scores_f16(const void*, const void*, float*, long,
                                     int, int, hipStream_t);
 hipError_t sentio_cosine_scores_bf16(const void*, const void*, float*, long,
@@ -165,9 +165,13 @@ torch::Tensor sample_token(torch::Tensor logits, double temperature,
                            int64_t seed) {
   TORCH_CHECK(logits.is_cuda() && logits.dim() == 2, "logits must be [B,V] GPU");
   auto l = logits.to(torch::kFloat).contiguous();
-  auto out = torch::empty({l.size(0)}, l.options().dtype(torch::kLong));
+  const long B = l.size(0);
+  auto out = torch::empty({B}, l.options().dtype(torch::kLong));
+  auto ws_val = torch::empty({B * 16}, l.options());
+  auto ws_idx = torch::empty({B * 16}, l.options().dtype(torch::kInt));
   check_hip(sentio_sample(l.data_ptr<float>(), out.data_ptr<int64_t>(),
-                          l.size(0), l.size(1), (float)temperature,
+                          ws_val.data_ptr<float>(), ws_idx.data_ptr<int>(),
+                          B, l.size(1), (float)temperature,
                           (unsigned)seed, stream()), "sample");
   return out;
 }

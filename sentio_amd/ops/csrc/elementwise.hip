@@ -268,17 +268,26 @@ __global__ void mean_pool_l2norm_kernel(const bf16* __restrict__ hidden,
 // ---------------------------------------------------------------- sampling
 // logits [B, V] f32 -> token [B] i64.  Gumbel-argmax at T>0 (equivalent to
 // softmax sampling, single pass, no normalization); plain argmax at T<=0.
-__global__ void sample_kernel(const float* __restrict__ logits,
-                              long* __restrict__ out, int V,
-                              float invT, unsigned seed, int greedy) {
+// Two phases: the Gumbel-perturbed scan is transcendental-bound (2 logs per
+// element), so each row splits across SAMPLE_PARTS blocks; a tiny combine
+// kernel reduces the per-part winners.
+#define SAMPLE_PARTS 16
+__global__ void sample_part_kernel(const float* __restrict__ logits,
+                                   float* __restrict__ ws_val,
+                                   int* __restrict__ ws_idx, int V,
+                                   float invT, unsigned seed, int greedy) {
   __shared__ float s_val[32];
   __shared__ int s_idx[32];
   const int b = blockIdx.x;
+  const int part = blockIdx.y;
   const float* lb = logits + (long)b * V;
+  const int span = (V + SAMPLE_PARTS - 1) / SAMPLE_PARTS;
+  const int v_lo = part * span;
+  const int v_hi = min(V, v_lo + span);
 
   float best = -INFINITY;
-  int besti = 0;
-  for (int v = threadIdx.x; v < V; v += blockDim.x) {
+  int besti = v_lo;
+  for (int v = v_lo + threadIdx.x; v < v_hi; v += blockDim.x) {
     float sc = lb[v];
     if (!greedy) {
       unsigned h = hash_u32(seed, (unsigned)b, (unsigned)v);
@@ -306,8 +315,23 @@ __global__ void sample_kernel(const float* __restrict__ logits,
       if (s_val[w] > bv || (s_val[w] == bv && s_idx[w] < bi)) {
         bv = s_val[w]; bi = s_idx[w];
       }
-    out[b] = bi;
+    ws_val[b * SAMPLE_PARTS + part] = bv;
+    ws_idx[b * SAMPLE_PARTS + part] = bi;
   }
+}
+
+__global__ void sample_combine_kernel(const float* __restrict__ ws_val,
+                                      const int* __restrict__ ws_idx,
+                                      long* __restrict__ out) {
+  const int b = blockIdx.x;
+  float bv = -INFINITY;
+  int bi = 0;
+  for (int p = 0; p < SAMPLE_PARTS; ++p) {
+    const float v = ws_val[b * SAMPLE_PARTS + p];
+    const int i = ws_idx[b * SAMPLE_PARTS + p];
+    if (v > bv || (v == bv && i < bi)) { bv = v; bi = i; }
+  }
+  out[b] = bi;
 }
 
 // ---------------------------------------------------------------- C API
@@ -415,12 +439,16 @@ hipError_t sentio_mean_pool_l2norm(const void* hidden, const unsigned char* mask
   return hipSuccess;
 }
 
-hipError_t sentio_sample(const float* logits, long* out, int B, int V,
-                         float temperature, unsigned seed, hipStream_t stream) {
+hipError_t sentio_sample(const float* logits, long* out, float* ws_val,
+                         int* ws_idx, int B, int V, float temperature,
+                         unsigned seed, hipStream_t stream) {
   int greedy = temperature <= 0.f;
   float invT = greedy ? 1.f : 1.f / temperature;
-  hipLaunchKernelGGL(sample_kernel, dim3(B), dim3(256), 0, stream, logits, out,
-                     V, invT, seed, greedy);
+  hipLaunchKernelGGL(sample_part_kernel, dim3(B, SAMPLE_PARTS), dim3(256), 0,
+                     stream, logits, ws_val, ws_idx, V, invT, seed, greedy);
+  HIP_CHECK_LAUNCH();
+  hipLaunchKernelGGL(sample_combine_kernel, dim3(B), dim3(1), 0, stream,
+                     ws_val, ws_idx, out);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
