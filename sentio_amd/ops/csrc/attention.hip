@@ -346,7 +346,7 @@ __global__ void decode_attn_split_kernel(
       for (int i = threadIdx.x * 8; i < elems; i += DEC_CHUNK * 8) {
         const int row = i / D;
         const int d = i % D;
-        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+        bf16x8 v = nt_load8(
             reinterpret_cast<const short*>(kb + (long)(s0 + row) * D) + d);
         *reinterpret_cast<bf16x8*>(
             k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = v;
@@ -404,7 +404,7 @@ __global__ void decode_attn_split_kernel(
       for (int e = 0; e < 8; ++e) o_part[g][e] *= alpha[g];
     if (dg_ok) {
       for (int j = jslot; j < chunk; j += 16) {
-        bf16x8 v8 = *reinterpret_cast<const bf16x8*>(
+        bf16x8 v8 = nt_load8(
             reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
             dgroup * 8);
         float vf[8];

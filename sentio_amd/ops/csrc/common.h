@@ -18,6 +18,13 @@ typedef float  f32x16 __attribute__((ext_vector_type(16)));
 typedef short  bf16x8 __attribute__((ext_vector_type(8)));
 typedef short  bf16x4 __attribute__((ext_vector_type(4)));
 
+// Non-temporal 16 B load (MI355X_MICROARCH.md nt-weights: streamed data one
+// CU reads exactly once should bypass cache retention — measured -18% landing
+// time on weight streams; use for decode KV streaming, NOT for reused tiles).
+__device__ __forceinline__ bf16x8 nt_load8(const short* p) {
+  return __builtin_nontemporal_load(reinterpret_cast<const bf16x8*>(p));
+}
+
 __device__ __forceinline__ float bf2f(bf16 v) { return __bfloat162float(v); }
 __device__ __forceinline__ bf16 f2bf(float v) { return __float2bfloat16(v); }
 
