@@ -83,6 +83,8 @@ class Settings:
     max_document_len: int = 50000
     cache_backend: str = "memory"           # memory | multi_tier | disk
     cache_ttl_s: float = 300.0
+    embedding_cache_size: int = 2048        # reference base.py:23-106 LFU+TTL
+    embedding_cache_ttl_s: float = 3600.0
     health_interval_s: float = 30.0         # reference patterns.py:252-306 loop
 
     # --- models / engines (MI355X-native; sizes are the bench ladder's) ---

@@ -20,7 +20,8 @@ from sentio_amd.engines.transformer import Transformer
 
 class EncoderEngine:
     def __init__(self, model: str = "sentio-encoder-base", device: str = "cpu",
-                 dtype: str = "bf16", max_seq: int = 512, seed: int = 101):
+                 dtype: str = "bf16", max_seq: int = 512, seed: int = 101,
+                 cache_size: int = 0, cache_ttl: float = 3600.0):
         self.cfg = get_model_config(model)
         self.device = device
         self.max_seq = min(max_seq, self.cfg.max_seq)
@@ -28,17 +29,44 @@ class EncoderEngine:
         self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
         self.dim = self.cfg.dim
         self.calls = 0
+        self.cache = None
+        if cache_size > 0:
+            from sentio_amd.caching.memory import MemoryCache
+
+            self.cache = MemoryCache(max_size=cache_size, default_ttl=cache_ttl)
 
     @torch.inference_mode()
     def embed(self, texts: list[str], batch_size: int = 64) -> torch.Tensor:
-        """texts → [N, dim] fp32 L2-normalized embeddings (on self.device)."""
+        """texts → [N, dim] fp32 L2-normalized embeddings (on self.device).
+
+        With a cache configured, per-text lookup + merge semantics mirror the
+        reference's embedding cache (reference embeddings/base.py:23-106,
+        jina.py:217-248: hit texts skip the forward, misses are embedded in
+        one batch and written back)."""
         if not texts:
             return torch.empty(0, self.dim, device=self.device)
-        outs = []
-        for i in range(0, len(texts), batch_size):
-            outs.append(self._embed_batch(texts[i : i + batch_size]))
         self.calls += 1
-        return torch.cat(outs, dim=0)
+        if self.cache is None:
+            outs = []
+            for i in range(0, len(texts), batch_size):
+                outs.append(self._embed_batch(texts[i : i + batch_size]))
+            return torch.cat(outs, dim=0)
+
+        out = torch.empty(len(texts), self.dim, device=self.device)
+        miss_idx: list[int] = []
+        for i, t in enumerate(texts):
+            hit = self.cache.get_embedding(t)
+            if hit is not None:
+                out[i] = hit.to(self.device)
+            else:
+                miss_idx.append(i)
+        for i0 in range(0, len(miss_idx), batch_size):
+            idxs = miss_idx[i0 : i0 + batch_size]
+            vecs = self._embed_batch([texts[i] for i in idxs])
+            for j, i in enumerate(idxs):
+                out[i] = vecs[j]
+                self.cache.set_embedding(texts[i], vecs[j])
+        return out
 
     def _embed_batch(self, texts: list[str]) -> torch.Tensor:
         padded, lens = self.tokenizer.encode_batch(texts, self.max_seq)

@@ -56,7 +56,9 @@ class ServiceContainer:
             from sentio_amd.engines.encoder import EncoderEngine
 
             return EncoderEngine(self.settings.encoder_model, device=self.device,
-                                 dtype=self.settings.compute_dtype)
+                                 dtype=self.settings.compute_dtype,
+                                 cache_size=self.settings.embedding_cache_size,
+                                 cache_ttl=self.settings.embedding_cache_ttl_s)
 
         return self._get("encoder", make)
 

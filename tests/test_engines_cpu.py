@@ -69,3 +69,16 @@ def test_decode_matches_prefill_consistency():
     g.model.prefill(tokens[:, :-1], cache2)
     logits_decode = g.model.decode_step(tokens[:, -1:], cache2)
     torch.testing.assert_close(logits_prefill, logits_decode, rtol=1e-3, atol=1e-3)
+
+
+def test_encoder_embedding_cache():
+    from sentio_amd.engines.encoder import EncoderEngine
+
+    e = EncoderEngine("tiny-encoder", device="cpu", cache_size=16)
+    v1 = e.embed(["alpha", "beta"])
+    assert e.cache.hits == 0 and e.cache.misses >= 2
+    v2 = e.embed(["alpha", "gamma", "beta"])
+    assert e.cache.hits == 2
+    import torch
+    assert torch.allclose(v1[0], v2[0], atol=1e-6)
+    assert torch.allclose(v1[1], v2[2], atol=1e-6)
