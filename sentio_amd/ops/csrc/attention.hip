@@ -353,13 +353,16 @@ __global__ void decode_attn_split_kernel(
       }
     }
     __syncthreads();
-    // phase A2: thread-per-key dot vs all G query heads, K read from LDS
-    {
-      float sc[G];
+    // phase A2: thread-per-key dot vs all G query heads, K read from LDS;
+    // scores stay in registers (the softmax below is the same thread).
+    float sc[G];
 #pragma unroll
-      for (int g = 0; g < G; ++g) sc[g] = 0.f;
+    for (int g = 0; g < G; ++g) sc[g] = -INFINITY;
+    {
       const int row = threadIdx.x;
       if (row < chunk) {
+#pragma unroll
+        for (int g = 0; g < G; ++g) sc[g] = 0.f;
         for (int d = 0; d < D; d += 8) {
           bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
               k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
@@ -372,30 +375,34 @@ __global__ void decode_attn_split_kernel(
             sc[g] += acc;
           }
         }
+#pragma unroll
+        for (int g = 0; g < G; ++g) sc[g] *= scale;
       }
+    }
+    // batched per-head softmax: ONE reduction pair for all G maxima, one
+    // for all G sums (was 3 barriers × 2 reductions × G heads).
+    {
+      float mx[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) mx[g] = sc[g];
+      block_reduce_vec<G, true>(mx, red);
+      float sums[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        const float m_new = fmaxf(m_run[g], mx[g]);
+        float a = (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+        if (m_new == -INFINITY) a = 1.f;
+        alpha[g] = a;
+        const float p = (sc[g] != -INFINITY) ? __expf(sc[g] - m_new) : 0.f;
+        p_sh[g * DEC_CHUNK + threadIdx.x] = p;
+        sums[g] = p;
+        m_run[g] = m_new;
+      }
+      block_reduce_vec<G, false>(sums, red);  // barrier also publishes p_sh
 #pragma unroll
       for (int g = 0; g < G; ++g)
-        p_sh[g * DEC_CHUNK + threadIdx.x] =
-            (row < chunk) ? sc[g] * scale : -INFINITY;
+        l_run[g] = l_run[g] * alpha[g] + sums[g];
     }
-    __syncthreads();
-    // per-head block max/sum + in-place exp (scores → p in p_sh)
-#pragma unroll
-    for (int g = 0; g < G; ++g) {
-      const float sc = p_sh[g * DEC_CHUNK + threadIdx.x];
-      const bool ok = threadIdx.x < chunk;
-      float cmax = block_max(ok ? sc : -INFINITY, red);
-      float m_new = fmaxf(m_run[g], cmax);
-      float a = (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
-      if (m_new == -INFINITY) a = 1.f;
-      alpha[g] = a;
-      float p = (ok && sc != -INFINITY) ? __expf(sc - m_new) : 0.f;
-      p_sh[g * DEC_CHUNK + threadIdx.x] = p;
-      float csum = block_sum(p, red);
-      l_run[g] = l_run[g] * a + csum;
-      m_run[g] = m_new;
-    }
-    __syncthreads();
     // phase B: PV.  v loads are bf16x8 (16 B) and fully coalesced; the
     // p_sh reads broadcast (16 lanes share one address).
 #pragma unroll

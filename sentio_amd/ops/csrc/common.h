@@ -95,6 +95,37 @@ __device__ __forceinline__ float block_max(float v, float* scratch) {
   return out;
 }
 
+// Batched block reduction: G values per thread reduced across the block in
+// ONE barrier pair (vs 3 barriers × G calls of block_max/block_sum).
+// scratch must hold G * (blockDim/WAVE) floats.
+template <int G, bool IS_MAX>
+__device__ __forceinline__ void block_reduce_vec(float (&v)[G],
+                                                 float* scratch) {
+  const int wid = threadIdx.x / WAVE;
+  const int nw = blockDim.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      float o = __shfl_xor(v[g], off, WAVE);
+      v[g] = IS_MAX ? fmaxf(v[g], o) : v[g] + o;
+    }
+    if (lane == 0) scratch[g * nw + wid] = v[g];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    float acc = IS_MAX ? -INFINITY : 0.f;
+    for (int w = 0; w < nw; ++w) {
+      float x = scratch[g * nw + w];
+      acc = IS_MAX ? fmaxf(acc, x) : acc + x;
+    }
+    v[g] = acc;
+  }
+  __syncthreads();  // scratch reusable after return
+}
+
 // simple splitmix-style hash for on-device RNG (sampling kernel)
 __device__ __forceinline__ unsigned hash_u32(unsigned a, unsigned b, unsigned c) {
   unsigned h = a * 0x9E3779B9u ^ b * 0x85EBCA6Bu ^ c * 0xC2B2AE35u;
