@@ -30,10 +30,19 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 __device__ __forceinline__ int k_lds_off(int row, int d_byte, int Dbytes) {
   return (row * Dbytes + d_byte) ^ ((row & 7) << 4);
 }
-// V^T tile [D][KVBLK] bf16 with 8-byte row pad: row stride = 64+8 = 72 B
-#define VT_STRIDE 72
-__device__ __forceinline__ int vt_lds_off(int d, int key_byte) {
-  return d * VT_STRIDE + key_byte;
+// V tile image for ds_read_b64_tr_b16 (gfx950 hardware transpose-read,
+// guide §LDS): per 16-dim block db, 8 slots of 4(key)x16(dim) bf16 tiles,
+// evens-then-odds slot order so a wave's 16-lane group g finds its PV
+// fragment's keys 8g+0..3 at slot g and keys 8g+4..7 at slot g+4 — the
+// second read is one uniform +512 B offset.  Staging stays vectorized
+// 16 B row-major writes (the old V^T image needed 8 scalar stores per
+// load); the fragment read is 2 tr-reads instead of 8 scalar ds_reads.
+typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
+typedef __attribute__((address_space(3))) bf16x4_t* lds_v4_ptr;
+__device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
+  const int db = d >> 4, kb = key >> 2;
+  const int slot = (kb & 1) ? 4 + (kb >> 1) : (kb >> 1);
+  return (db * 8 + slot) * 64 + (key & 3) * 16 + (d & 15);
 }
 // P tile [QBLK][KVBLK] bf16 with 16-byte row pad: stride = 64+16 = 80 B
 #define P_STRIDE 80
@@ -54,11 +63,11 @@ __global__ void flash_attn_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // carve: K tile | V^T tile | per-wave P tiles
   char* k_lds = smem;                                   // KVBLK * D * 2
-  char* vt_lds = k_lds + KVBLK * D * 2;                 // D * VT_STRIDE
+  char* vt_lds = k_lds + KVBLK * D * 2;                 // KVBLK * D * 2 (tr image)
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  char* p_lds = vt_lds + D * VT_STRIDE + wid * QBLK * P_STRIDE;
+  char* p_lds = vt_lds + KVBLK * D * 2 + wid * QBLK * P_STRIDE;
 
   const int qt = blockIdx.x * FA_WAVES + wid;  // this wave's q tile
   const int h = blockIdx.y;
@@ -128,7 +137,8 @@ __global__ void flash_attn_kernel(
         }
         // swizzled K write: 16B aligned chunks keep XOR validity ((d*2)%16==0)
         *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) = val;
-        // V^T write: transpose — scalar stores
+        // V write: one 16 B store into the tr-read tile image (d..d+7 stays
+        // inside one 4x16 tile since d % 8 == 0)
         bf16x8_t vv;
         if (key < kvlen) {
           vv = *reinterpret_cast<const bf16x8_t*>(
@@ -137,9 +147,7 @@ __global__ void flash_attn_kernel(
           bf16x8_t z = {};
           vv = z;
         }
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<__bf16*>(vt_lds + vt_lds_off(d + j, row * 2)) = vv[j];
+        *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) = vv;
       }
     }
     __syncthreads();  // staging visible to every wave
@@ -232,12 +240,17 @@ __global__ void flash_attn_kernel(
           // rescale accumulator rows by alpha[r]
 #pragma unroll
           for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
-          // B frag: V^T[d = nb*16 + (l&15)][key = kofs + j]
+          // B frag via hardware transpose-read: group g = lane>>4 gets
+          // column (lane&15) of tiles at slot g (keys 8g+0..3) and slot
+          // g+4 (keys 8g+4..7, uniform +512 B)
+          char* vb_base = vt_lds + nb * 1024 + (long)lane * 8;
+          bf16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_v4_ptr)vb_base);
+          bf16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_v4_ptr)(vb_base + 512));
           bf16x8_t b_v;
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            b_v[j] = *reinterpret_cast<const __bf16*>(
-                vt_lds + vt_lds_off(nb * 16 + (lane & 15), (kofs + j) * 2));
+          for (int j = 0; j < 4; ++j) { b_v[j] = lo[j]; b_v[4 + j] = hi[j]; }
           o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_p, b_v, o_acc[nb], 0, 0, 0);
         }
@@ -491,7 +504,7 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
                              int H, int Hkv, int D, float scale, int causal,
                              hipStream_t stream) {
   if (D % 32 != 0 || D > MAXD) return hipErrorInvalidValue;
-  size_t lds = (size_t)KVBLK * D * 2 + (size_t)D * VT_STRIDE
+  size_t lds = (size_t)KVBLK * D * 2 * 2   // K (swizzled) + V (tr image)
                + 4 * QBLK * P_STRIDE;
   dim3 grid((S + 4 * QBLK - 1) / (4 * QBLK), H, B);
   hipLaunchKernelGGL(flash_attn_kernel, grid, dim3(256), lds, stream,
