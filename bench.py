@@ -62,6 +62,10 @@ def parse_args():
     p.add_argument("--vocab-terms", type=int, default=30000)
     p.add_argument("--seq-len", type=int, default=2048,
                    help="generation context budget (prompt cap)")
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree for the generator "
+                        "(must equal world size; config #5: --model "
+                        "llama3-70b --tp 8)")
     return p.parse_args()
 
 
@@ -205,12 +209,20 @@ def main():
     from sentio_amd.engines.encoder import EncoderEngine
     from sentio_amd.engines.generator import GeneratorEngine
     from sentio_amd.engines.reranker import RerankerEngine
+    from sentio_amd.parallel.tp import TPContext
+
+    tp_mode = args.tp > 1
+    if tp_mode:
+        assert world == args.tp, (
+            f"--tp {args.tp} requires world size {args.tp} (got {world}); "
+            "all ranks form one TP group")
 
     t_init = time.time()
     encoder = EncoderEngine(args.encoder, device=device, max_seq=128)
     reranker = RerankerEngine(args.reranker, device=device, max_seq=256)
     generator = GeneratorEngine(args.model, device=device,
-                                max_seq=args.seq_len + args.gen_tokens + 8)
+                                max_seq=args.seq_len + args.gen_tokens + 8,
+                                tp=TPContext.from_env() if tp_mode else None)
     dense, bm = build_synthetic_indexes(
         args.docs_per_gpu, encoder.dim, args.vocab_terms, device, rank)
     corpus = SyntheticCorpus(rank)
@@ -219,7 +231,9 @@ def main():
         torch.cuda.synchronize()
     init_s = time.time() - t_init
 
-    rng = np.random.RandomState(42 + rank)
+    # TP mode: every rank must build IDENTICAL queries (one shared batch
+    # through the TP group); DP mode: per-rank query streams.
+    rng = np.random.RandomState(42 if tp_mode else 42 + rank)
 
     def make_queries(step: int) -> list[str]:
         out = []
@@ -365,7 +379,7 @@ def main():
         tdist.all_reduce(e, op=tdist.ReduceOp.MAX)
     elapsed = float(e.item())
 
-    total_requests = args.batch * world * args.steps
+    total_requests = args.batch * args.steps * (1 if tp_mode else world)
     qps = total_requests / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
@@ -382,7 +396,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 1),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if tp_mode else "weak",
             "vs_baseline": None,
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
@@ -395,7 +409,8 @@ def main():
                 "gen_tokens": args.gen_tokens,
                 "docs_per_gpu": args.docs_per_gpu,
                 "total_docs": args.docs_per_gpu * world,
-                "parallelism": f"dp{world}+index-shard{world}",
+                "parallelism": (f"tp{world}+index-shard{world}" if tp_mode
+                                else f"dp{world}+index-shard{world}"),
                 "pipeline": "embed>hybrid(dense+bm25+rrf)>rerank>select>generate"
                             + (">verify" if args.verify else ""),
                 "p50_ms_per_request_batch": round(ms_per_step, 1),

@@ -101,6 +101,10 @@ class GeneratorEngine:
         use_graphs = (
             self._GRAPHS_ENABLED
             and self.device != "cpu"
+            # RCCL collectives inside hipGraph capture are not worth the
+            # risk for TP decode (the all-reduce graph-capture path varies
+            # by RCCL version); TP decode runs eager.
+            and not self.model.tp.enabled
             and os.environ.get("SENTIO_DISABLE_HIPGRAPH", "0") != "1"
         )
         key = (batch, cache_len)

@@ -202,3 +202,25 @@ def _broadcast_worker(rank, world):
 
 def test_broadcast_weights_syncs_replicas():
     _run_workers(_broadcast_worker, world=2)
+
+
+# ---------------- TP bench path (gloo, world 2) ----------------
+
+def _tp_generate_worker(rank, world):
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.parallel.tp import TPContext
+
+    g = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=64,
+                        tp=TPContext.from_env())
+    outs = g.generate(["hello world", "tensor parallel"],
+                      max_new_tokens=6, temperature=0.0, stop_on_eos=False)
+    assert len(outs) == 2
+    # both ranks must produce identical completions (shared batch)
+    import torch.distributed as dist
+    gathered = [None, None]
+    dist.all_gather_object(gathered, outs)
+    assert gathered[0] == gathered[1]
+
+
+def test_tp_generation_consistent_across_ranks():
+    _run_workers(_tp_generate_worker, world=2)
