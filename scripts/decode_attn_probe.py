@@ -1,0 +1,36 @@
+"""Isolated decode-attention microbenchmark (llama3-8b decode shape).
+Prints achieved µs + effective KV TB/s per splits setting.
+Run: gpurun -- 'python scripts/decode_attn_probe.py'"""
+import os, time
+import torch
+from sentio_amd import ops
+
+def bench(q, kc, vc, lens, iters=200):
+    for _ in range(20):
+        ops.decode_attention(q, kc, vc, lens)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ops.decode_attention(q, kc, vc, lens)
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+def main():
+    dev = "cuda:0"
+    for B, slen in [(16, 1600), (32, 1600), (32, 512)]:
+        H, Hkv, Smax, D = 32, 8, 2120, 128
+        torch.manual_seed(0)
+        q = torch.randn(B, H, D, dtype=torch.bfloat16, device=dev)
+        kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+        vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+        lens = torch.full((B,), slen, dtype=torch.int32, device=dev)
+        kv_bytes = 2 * B * Hkv * slen * D * 2
+        print(f"--- B={B} slen={slen} (KV {kv_bytes/1e6:.0f} MB)")
+        for splits in (0, 1, 2, 4, 8):
+            if splits: os.environ["SENTIO_DECODE_SPLITS"] = str(splits)
+            else: os.environ.pop("SENTIO_DECODE_SPLITS", None)
+            t = bench(q, kc, vc, lens)
+            print(f"splits={splits or 'auto'}: {t*1e6:7.1f}us  {kv_bytes/t/1e12:.2f} TB/s")
+
+if __name__ == "__main__":
+    main()

@@ -4,6 +4,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 
 extern "C" {
 hipError_t sentio_rmsnorm(const void*, const void*, void*, long, int, float,
@@ -253,6 +254,8 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   int splits = (int)((512 + (long)B * Hkv - 1) / ((long)B * Hkv));
   int max_splits = std::max(1, Smax / 256);
   splits = std::max(1, std::min(splits, max_splits));
+  if (const char* ov = std::getenv("SENTIO_DECODE_SPLITS"))
+    splits = std::max(1, std::min(atoi(ov), max_splits));
   auto ws_o = torch::empty({(long)B * Hkv * splits * G * D},
                            q.options().dtype(torch::kFloat));
   auto ws_ml = torch::empty({(long)B * Hkv * splits * G * 2},
