@@ -154,7 +154,10 @@ def dense_search_ids(dense: DenseIndex, q: torch.Tensor, k: int):
     return vals, idx
 
 
-def bm25_search_ids(bm: BM25Index, term_ids: np.ndarray, k: int, device: str):
+def bm25_search_batch(bm: BM25Index, term_id_lists: list[np.ndarray], k: int,
+                      device: str):
+    """Score a whole query batch with ONE host sync: launch every query's
+    scoring kernel back-to-back, stack, one batched top-k, one D2H copy."""
     import torch as T
 
     if device != "cpu":
@@ -169,24 +172,33 @@ def bm25_search_ids(bm: BM25Index, term_ids: np.ndarray, k: int, device: str):
                 "doc_len": T.from_numpy(bm.doc_len).to(device),
             }
         a = bm._device_arrays
-        tids = T.from_numpy(term_ids).to(device)
-        scores = ops.bm25_score(
-            tids, a["indptr"], a["post_doc"], a["post_tf"], a["idf"],
-            a["doc_len"], n_docs=len(bm.doc_len), k1=bm.k1, b=bm.b,
-            avgdl=float(bm.doc_len.mean()), plus_delta=0.0)
-        vals, idx = T.topk(scores, min(k, scores.shape[0]))
-        return vals.cpu().tolist(), idx.cpu().tolist()
+        avgdl = float(bm.doc_len.mean())
+        per_q = []
+        for term_ids in term_id_lists:
+            tids = T.from_numpy(term_ids).to(device)
+            per_q.append(ops.bm25_score(
+                tids, a["indptr"], a["post_doc"], a["post_tf"], a["idf"],
+                a["doc_len"], n_docs=len(bm.doc_len), k1=bm.k1, b=bm.b,
+                avgdl=avgdl, plus_delta=0.0))
+        S = T.stack(per_q)                       # [Q, N]
+        vals, idx = T.topk(S, min(k, S.shape[1]), dim=1)
+        vals_l = vals.cpu().tolist()
+        idx_l = idx.cpu().tolist()
+        return [list(zip(idx_l[i], vals_l[i])) for i in range(len(per_q))]
     # CPU fallback
-    scores = np.zeros(len(bm.doc_len), np.float32)
+    out = []
     avgdl = max(float(bm.doc_len.mean()), 1e-9)
     den = bm.k1 * (1 - bm.b + bm.b * bm.doc_len / avgdl)
-    for t in term_ids:
-        lo, hi = bm.indptr[t], bm.indptr[t + 1]
-        d = bm.post_doc[lo:hi]
-        tf = bm.post_tf[lo:hi]
-        scores[d] += bm.idf[t] * tf * (bm.k1 + 1) / (tf + den[d])
-    idx = np.argsort(-scores)[:k]
-    return scores[idx].tolist(), idx.tolist()
+    for term_ids in term_id_lists:
+        scores = np.zeros(len(bm.doc_len), np.float32)
+        for t in term_ids:
+            lo, hi = bm.indptr[t], bm.indptr[t + 1]
+            d = bm.post_doc[lo:hi]
+            tf = bm.post_tf[lo:hi]
+            scores[d] += bm.idf[t] * tf * (bm.k1 + 1) / (tf + den[d])
+        idx = np.argsort(-scores)[:k]
+        out.append(list(zip(idx.tolist(), scores[idx].tolist())))
+    return out
 
 
 def main():
@@ -273,14 +285,14 @@ def main():
             gathered_i = D.all_gather_objects(idx.numpy())
         else:
             gathered_v = [vals.numpy()]; gathered_i = [idx.numpy()]
-        # 3. sparse search (local shard) + gather
-        sparse_hits_local = []
-        for q in queries:
-            tids = np.array([int(tok[4:]) for tok in q.split()
-                             if tok.startswith("term") and tok[4:].isdigit()],
-                            np.int64)
-            sv, si = bm25_search_ids(bm, tids, args.top_k, device)
-            sparse_hits_local.append(list(zip(si, sv)))
+        # 3. sparse search (local shard, one batched sync) + gather
+        tid_lists = [
+            np.array([int(tok[4:]) for tok in q.split()
+                      if tok.startswith("term") and tok[4:].isdigit()],
+                     np.int64)
+            for q in queries
+        ]
+        sparse_hits_local = bm25_search_batch(bm, tid_lists, args.top_k, device)
         if world > 1:
             sparse_all = D.all_gather_objects(sparse_hits_local)
         else:

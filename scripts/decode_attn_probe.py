@@ -1,7 +1,8 @@
 """Isolated decode-attention microbenchmark (llama3-8b decode shape).
 Prints achieved µs + effective KV TB/s per splits setting.
 Run: gpurun -- 'python scripts/decode_attn_probe.py'"""
-import os, time
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from sentio_amd import ops
 

@@ -68,11 +68,31 @@ class EncoderEngine:
                 self.cache.set_embedding(texts[i], vecs[j])
         return out
 
+    def _graph_pool(self):
+        pool = getattr(self, "_graphs", None)
+        if pool is None:
+            from sentio_amd.engines.graphed import GraphedEnginePool
+
+            def fwd(tokens, kv_lens):
+                S = tokens.shape[1]
+                mask = (torch.arange(S, device=self.device).unsqueeze(0)
+                        < kv_lens.unsqueeze(1))
+                hidden = self.model.forward_hidden(tokens, kv_lens=kv_lens)
+                return ops.mean_pool_l2norm(hidden, mask)
+
+            pool = self._graphs = GraphedEnginePool(fwd, self.max_seq,
+                                                    self.device)
+        return pool
+
     def _embed_batch(self, texts: list[str]) -> torch.Tensor:
         padded, lens = self.tokenizer.encode_batch(texts, self.max_seq)
         tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
         B, S = tokens.shape
         kv_lens = torch.tensor(lens, dtype=torch.int32, device=self.device)
+        pool = self._graph_pool()
+        if pool.active:
+            # fixed-shape hipGraph replay (launch-bound at these sizes)
+            return pool.run(tokens, kv_lens)
         mask = torch.arange(S, device=self.device).unsqueeze(0) < kv_lens.unsqueeze(1)
         hidden = self.model.forward_hidden(tokens, kv_lens=kv_lens)
         return ops.mean_pool_l2norm(hidden, mask)

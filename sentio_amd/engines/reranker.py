@@ -26,10 +26,26 @@ class RerankerEngine:
         self.tokenizer = ByteTokenizer()
         self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
 
+    def _graph_pool(self):
+        pool = getattr(self, "_graphs", None)
+        if pool is None:
+            from sentio_amd.engines.graphed import GraphedEnginePool
+
+            def fwd(tokens, kv_lens):
+                hidden = self.model.forward_hidden(tokens, kv_lens=kv_lens)
+                pooled = hidden[:, 0, :]   # first-token pooled representation
+                s = torch.nn.functional.linear(pooled, self.model.w.head)
+                return torch.sigmoid(s.float()).squeeze(-1)
+
+            pool = self._graphs = GraphedEnginePool(fwd, self.max_seq,
+                                                    self.device)
+        return pool
+
     @torch.inference_mode()
     def score_pairs(self, query: str, texts: list[str], batch_size: int = 32) -> list[float]:
         if not texts:
             return []
+        pool = self._graph_pool()
         scores: list[float] = []
         for i in range(0, len(texts), batch_size):
             chunk = texts[i : i + batch_size]
@@ -37,10 +53,14 @@ class RerankerEngine:
             padded, lens = self.tokenizer.encode_batch(pair_texts, self.max_seq)
             tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
             kv_lens = torch.tensor(lens, dtype=torch.int32, device=self.device)
-            hidden = self.model.forward_hidden(tokens, kv_lens=kv_lens)
-            pooled = hidden[:, 0, :]  # first-token pooled representation
-            s = torch.nn.functional.linear(pooled, self.model.w.head).float().squeeze(-1)
-            scores.extend(torch.sigmoid(s).cpu().tolist())
+            if pool.active:
+                s = pool.run(tokens, kv_lens)
+            else:
+                hidden = self.model.forward_hidden(tokens, kv_lens=kv_lens)
+                pooled = hidden[:, 0, :]  # first-token pooled representation
+                s = torch.sigmoid(torch.nn.functional.linear(
+                    pooled, self.model.w.head).float()).squeeze(-1)
+            scores.extend(s.cpu().tolist())
         return scores
 
     @torch.inference_mode()
