@@ -296,14 +296,19 @@ __global__ void flash_attn_kernel(
 // UNNORMALIZED partials (o = sum exp(s-m) v, plus m and l) to the
 // workspace; a combine kernel reduces the splits.  SPLITS is chosen to
 // fill the 256 CUs (Hkv*B alone is 0.5 blocks/CU at common shapes).
-template <int G>
+// DT: compile-time head dim.  Every inner loop gets a compile-time trip
+// count so the compiler unrolls and BATCHES the global loads — the dynamic
+// version compiled to ONE load + s_waitcnt(0) per iteration (zero
+// memory-level parallelism; measured 87 us where streaming SoL is ~17 us).
+template <int G, int DT>
 __global__ void decode_attn_split_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc,
     float* __restrict__ ws_o,      // [B, Hkv, SPLITS, G, D]
     float* __restrict__ ws_ml,     // [B, Hkv, SPLITS, G, 2]
     const int* __restrict__ seq_lens,
-    int H, int Hkv, int Smax, int D, float scale, int splits) {
+    int H, int Hkv, int Smax, int D_, float scale, int splits) {
+  constexpr int D = DT;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* p_sh = reinterpret_cast<float*>(smem);            // [G][DEC_CHUNK]
   float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
@@ -350,11 +355,27 @@ __global__ void decode_attn_split_kernel(
   const int Dbytes = D * 2;
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    // phase A1: cooperative coalesced K staging into XOR-swizzled LDS
-    // (256 threads issue independent bf16x8 global loads — full lines and
-    // deep memory-level parallelism; the swizzle kills the 16-way bank
-    // conflict a row-major [key][D] layout would give phase A2).
-    {
+    // phase A1: cooperative coalesced K staging into XOR-swizzled LDS.
+    // Full chunks take the unrolled path: 8 loads in flight before any
+    // ds_write (the dynamic loop serializes one load per s_waitcnt).
+    if (chunk == DEC_CHUNK) {
+      constexpr int IT = DEC_CHUNK * D / (DEC_CHUNK * 8);  // = D/8 per thread
+      bf16x8 tmp[IT];
+#pragma unroll
+      for (int u = 0; u < IT; ++u) {
+        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
+        tmp[u] = nt_load8(
+            reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
+            + (i % D));
+      }
+#pragma unroll
+      for (int u = 0; u < IT; ++u) {
+        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
+        const int row = i / D, d = i % D;
+        *reinterpret_cast<bf16x8*>(
+            k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = tmp[u];
+      }
+    } else {
       const int elems = chunk * D;
       for (int i = threadIdx.x * 8; i < elems; i += DEC_CHUNK * 8) {
         const int row = i / D;
@@ -417,24 +438,48 @@ __global__ void decode_attn_split_kernel(
         l_run[g] = l_run[g] * alpha[g] + sums[g];
     }
     // phase B: PV.  v loads are bf16x8 (16 B) and fully coalesced; the
-    // p_sh reads broadcast (16 lanes share one address).
+    // p_sh reads broadcast (16 lanes share one address).  Full chunks:
+    // all 16 V loads issued up front (batched MLP), FMAs after.
 #pragma unroll
     for (int g = 0; g < G; ++g)
 #pragma unroll
       for (int e = 0; e < 8; ++e) o_part[g][e] *= alpha[g];
     if (dg_ok) {
-      for (int j = jslot; j < chunk; j += 16) {
-        bf16x8 v8 = nt_load8(
-            reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
-            dgroup * 8);
-        float vf[8];
+      if (chunk == DEC_CHUNK) {
+        constexpr int JT = DEC_CHUNK / 16;
+        bf16x8 v8[JT];
 #pragma unroll
-        for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[e]);
+        for (int u = 0; u < JT; ++u)
+          v8[u] = nt_load8(
+              reinterpret_cast<const short*>(
+                  vb + (long)(s0 + jslot + u * 16) * D) + dgroup * 8);
 #pragma unroll
-        for (int g = 0; g < G; ++g) {
-          const float p = p_sh[g * DEC_CHUNK + j];
+        for (int u = 0; u < JT; ++u) {
+          const int j = jslot + u * 16;
+          float vf[8];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = p_sh[g * DEC_CHUNK + j];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+          }
+        }
+      } else {
+        for (int j = jslot; j < chunk; j += 16) {
+          bf16x8 v8 = nt_load8(
+              reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
+              dgroup * 8);
+          float vf[8];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[e]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = p_sh[g * DEC_CHUNK + j];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+          }
         }
       }
     }
@@ -525,16 +570,22 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
   size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float)
                + (size_t)DEC_CHUNK * D * 2;   // swizzled K stage
   dim3 grid(Hkv, B, splits);
-#define DEC_CASE(GV)                                                          \
-  case GV:                                                                    \
-    hipLaunchKernelGGL((decode_attn_split_kernel<GV>), grid, dim3(DEC_CHUNK), \
-                       lds, stream, (const bf16*)q, (const bf16*)kc,          \
-                       (const bf16*)vc, ws_o, ws_ml, seq_lens, H, Hkv, Smax,  \
-                       D, scale, splits);                                     \
-    break;
-  switch (G) {
-    DEC_CASE(1) DEC_CASE(2) DEC_CASE(3) DEC_CASE(4)
-    DEC_CASE(5) DEC_CASE(6) DEC_CASE(7) DEC_CASE(8)
+  if (D != 64 && D != 128) return hipErrorInvalidValue;
+#define DEC_CASE(GV, DV)                                                      \
+  hipLaunchKernelGGL((decode_attn_split_kernel<GV, DV>), grid,                \
+                     dim3(DEC_CHUNK), lds, stream, (const bf16*)q,            \
+                     (const bf16*)kc, (const bf16*)vc, ws_o, ws_ml, seq_lens, \
+                     H, Hkv, Smax, D, scale, splits);                         \
+  break;
+  switch (G * 1000 + D) {
+    case 1064: DEC_CASE(1, 64) case 2064: DEC_CASE(2, 64)
+    case 3064: DEC_CASE(3, 64) case 4064: DEC_CASE(4, 64)
+    case 5064: DEC_CASE(5, 64) case 6064: DEC_CASE(6, 64)
+    case 7064: DEC_CASE(7, 64) case 8064: DEC_CASE(8, 64)
+    case 1128: DEC_CASE(1, 128) case 2128: DEC_CASE(2, 128)
+    case 3128: DEC_CASE(3, 128) case 4128: DEC_CASE(4, 128)
+    case 5128: DEC_CASE(5, 128) case 6128: DEC_CASE(6, 128)
+    case 7128: DEC_CASE(7, 128) case 8128: DEC_CASE(8, 128)
     default:
       return hipErrorInvalidValue;
   }
