@@ -52,6 +52,7 @@ __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
 // traffic than wave-private tiles, cooperative 256-thread staging.
 #define FA_WAVES 4
 
+template <int DT>
 __launch_bounds__(256)
 __global__ void flash_attn_kernel(
     const bf16* __restrict__ q,    // [B, S, H, D]
@@ -59,7 +60,9 @@ __global__ void flash_attn_kernel(
     const bf16* __restrict__ v,    // [B, S, Hkv, D]
     bf16* __restrict__ out,        // [B, S, H, D]
     const int* __restrict__ kv_lens,  // [B]
-    int B, int S, int H, int Hkv, int D, float scale, int causal) {
+    int B, int S, int H, int Hkv, int D_, float scale, int causal) {
+  constexpr int D = DT;            // compile-time: every staging/frag loop
+                                   // unrolls, loads batch before waits
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // carve: K tile | V^T tile | per-wave P tiles
   char* k_lds = smem;                                   // KVBLK * D * 2
@@ -78,33 +81,31 @@ __global__ void flash_attn_kernel(
   const bool active = q0 < S;           // inactive waves still hit barriers
   const int q_hi = min(q0 + QBLK - 1, S - 1);
 
-  const int DC = D / 32;                // feature chunks per mfma K-dim
-  const int NB = D / 16;                // output column blocks
+  constexpr int DC = D / 32;            // feature chunks per mfma K-dim
+  constexpr int NB = D / 16;            // output column blocks
 
   // ---- load Q fragments: a_q[dc] = Q[q0 + (l&15)][dc*32 + (l>>4)*8 + j]
   const int arow = lane & 15;
   const int kofs = (lane >> 4) * 8;
-  bf16x8_t a_q[4];
+  bf16x8_t a_q[DC];
   {
     const int qrow = q0 + arow;
     const bf16* qp = q + (((long)b * S + qrow) * H + h) * D + kofs;
 #pragma unroll
-    for (int dc = 0; dc < 4; ++dc) {
-      if (dc < DC) {
-        if (active && qrow < S) {
-          a_q[dc] = *reinterpret_cast<const bf16x8_t*>(qp + dc * 32);
-        } else {
-          bf16x8_t z = {};
-          a_q[dc] = z;
-        }
+    for (int dc = 0; dc < DC; ++dc) {
+      if (active && qrow < S) {
+        a_q[dc] = *reinterpret_cast<const bf16x8_t*>(qp + dc * 32);
+      } else {
+        bf16x8_t z = {};
+        a_q[dc] = z;
       }
     }
   }
 
   // ---- accumulators
-  f32x4_t o_acc[MAXD / 16];             // O in C-frag layout per 16-col block
+  f32x4_t o_acc[NB];                    // O in C-frag layout per 16-col block
 #pragma unroll
-  for (int nb = 0; nb < MAXD / 16; ++nb) o_acc[nb] = f32x4_t{};
+  for (int nb = 0; nb < NB; ++nb) o_acc[nb] = f32x4_t{};
   // per-lane row state: the C layout puts row = (l>>4)*4 + r; softmax rows
   // are shared by the 16 lanes of each group -> track per (group,reg)
   float m_run[4], l_run[4];
@@ -120,34 +121,33 @@ __global__ void flash_attn_kernel(
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int kv0 = kt * KVBLK;
-    // ---- stage K tile (swizzled) and V^T tile into LDS (256 threads)
+    // ---- stage K tile (swizzled) and V tr-image into LDS: all of this
+    // thread's K and V loads issue BEFORE any LDS store (batched MLP)
     {
-      const int elems = KVBLK * D;      // bf16 elements in tile
-      for (int i = threadIdx.x * 8; i < elems; i += FA_WAVES * WAVE * 8) {
-        const int row = i / D;
+      constexpr int IT = KVBLK * D / (FA_WAVES * WAVE * 8);
+      bf16x8_t kvals[IT], vvals[IT];
+#pragma unroll
+      for (int u = 0; u < IT; ++u) {
+        const int i = threadIdx.x * 8 + u * FA_WAVES * WAVE * 8;
+        const int key = kv0 + i / D;
         const int d = i % D;
-        const int key = kv0 + row;
-        bf16x8_t val;
         if (key < kvlen) {
-          val = *reinterpret_cast<const bf16x8_t*>(
-              k + (((long)b * S + key) * Hkv + hkv) * D + d);
+          const long base = (((long)b * S + key) * Hkv + hkv) * D + d;
+          kvals[u] = *reinterpret_cast<const bf16x8_t*>(k + base);
+          vvals[u] = *reinterpret_cast<const bf16x8_t*>(v + base);
         } else {
           bf16x8_t z = {};
-          val = z;
+          kvals[u] = z;
+          vvals[u] = z;
         }
-        // swizzled K write: 16B aligned chunks keep XOR validity ((d*2)%16==0)
-        *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) = val;
-        // V write: one 16 B store into the tr-read tile image (d..d+7 stays
-        // inside one 4x16 tile since d % 8 == 0)
-        bf16x8_t vv;
-        if (key < kvlen) {
-          vv = *reinterpret_cast<const bf16x8_t*>(
-              v + (((long)b * S + key) * Hkv + hkv) * D + d);
-        } else {
-          bf16x8_t z = {};
-          vv = z;
-        }
-        *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) = vv;
+      }
+#pragma unroll
+      for (int u = 0; u < IT; ++u) {
+        const int i = threadIdx.x * 8 + u * FA_WAVES * WAVE * 8;
+        const int row = i / D, d = i % D;
+        *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) =
+            kvals[u];
+        *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) = vvals[u];
       }
     }
     __syncthreads();  // staging visible to every wave
@@ -162,15 +162,13 @@ __global__ void flash_attn_kernel(
     for (int half = 0; half < 2; ++half) {
       s_acc[half] = f32x4_t{};
 #pragma unroll
-      for (int dc = 0; dc < 4; ++dc) {
-        if (dc < DC) {
-          // B frag: K_lds[half*16 + (l&15)][dc*32 + kofs + j]
-          const int krow = half * 16 + (lane & 15);
-          bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(
-              k_lds + k_lds_off(krow, (dc * 32 + kofs) * 2, D * 2));
-          s_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_q[dc], b_frag, s_acc[half], 0, 0, 0);
-        }
+      for (int dc = 0; dc < DC; ++dc) {
+        // B frag: K_lds[half*16 + (l&15)][dc*32 + kofs + j]
+        const int krow = half * 16 + (lane & 15);
+        bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(
+            k_lds + k_lds_off(krow, (dc * 32 + kofs) * 2, D * 2));
+        s_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_q[dc], b_frag, s_acc[half], 0, 0, 0);
       }
     }
 
@@ -235,8 +233,8 @@ __global__ void flash_attn_kernel(
 
       // ---- O = alpha*O + P V   (one mfma per 16-col block of V)
 #pragma unroll
-      for (int nb = 0; nb < MAXD / 16; ++nb) {
-        if (nb < NB) {
+      for (int nb = 0; nb < NB; ++nb) {
+        {
           // rescale accumulator rows by alpha[r]
 #pragma unroll
           for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
@@ -262,8 +260,8 @@ __global__ void flash_attn_kernel(
   // ---- epilogue: divide by l, store
   if (!active) return;
 #pragma unroll
-  for (int nb = 0; nb < MAXD / 16; ++nb) {
-    if (nb < NB) {
+  for (int nb = 0; nb < NB; ++nb) {
+    {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qrow_local = (lane >> 4) * 4 + r;
@@ -548,13 +546,23 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
                              void* out, const int* kv_lens, int B, int S,
                              int H, int Hkv, int D, float scale, int causal,
                              hipStream_t stream) {
-  if (D % 32 != 0 || D > MAXD) return hipErrorInvalidValue;
   size_t lds = (size_t)KVBLK * D * 2 * 2   // K (swizzled) + V (tr image)
                + 4 * QBLK * P_STRIDE;
   dim3 grid((S + 4 * QBLK - 1) / (4 * QBLK), H, B);
-  hipLaunchKernelGGL(flash_attn_kernel, grid, dim3(256), lds, stream,
-                     (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                     (bf16*)out, kv_lens, B, S, H, Hkv, D, scale, causal);
+  switch (D) {
+    case 64:
+      hipLaunchKernelGGL((flash_attn_kernel<64>), grid, dim3(256), lds, stream,
+                         (const bf16*)q, (const bf16*)k, (const bf16*)v,
+                         (bf16*)out, kv_lens, B, S, H, Hkv, D, scale, causal);
+      break;
+    case 128:
+      hipLaunchKernelGGL((flash_attn_kernel<128>), grid, dim3(256), lds, stream,
+                         (const bf16*)q, (const bf16*)k, (const bf16*)v,
+                         (bf16*)out, kv_lens, B, S, H, Hkv, D, scale, causal);
+      break;
+    default:
+      return hipErrorInvalidValue;
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
