@@ -299,6 +299,8 @@ __global__ void flash_attn_kernel(
 // version compiled to ONE load + s_waitcnt(0) per iteration (zero
 // memory-level parallelism; measured 87 us where streaming SoL is ~17 us).
 template <int G, int DT>
+__launch_bounds__(DEC_CHUNK, 1)   // LDS caps at 2 blocks/CU anyway: take the
+                                  // full VGPR budget, no scratch spills
 __global__ void decode_attn_split_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc,
@@ -357,21 +359,25 @@ __global__ void decode_attn_split_kernel(
     // Full chunks take the unrolled path: 8 loads in flight before any
     // ds_write (the dynamic loop serializes one load per s_waitcnt).
     if (chunk == DEC_CHUNK) {
-      constexpr int IT = DEC_CHUNK * D / (DEC_CHUNK * 8);  // = D/8 per thread
-      bf16x8 tmp[IT];
+      constexpr int IT = D / 8;          // loads per thread for a full chunk
+      constexpr int BATCH = 4;           // loads in flight (16-deep spilled)
 #pragma unroll
-      for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
-        tmp[u] = nt_load8(
-            reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
-            + (i % D));
-      }
+      for (int u0 = 0; u0 < IT; u0 += BATCH) {
+        bf16x8 tmp[BATCH];
 #pragma unroll
-      for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
-        const int row = i / D, d = i % D;
-        *reinterpret_cast<bf16x8*>(
-            k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = tmp[u];
+        for (int u = 0; u < BATCH; ++u) {
+          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
+          tmp[u] = nt_load8(
+              reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
+              + (i % D));
+        }
+#pragma unroll
+        for (int u = 0; u < BATCH; ++u) {
+          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
+          const int row = i / D, d = i % D;
+          *reinterpret_cast<bf16x8*>(
+              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = tmp[u];
+        }
       }
     } else {
       const int elems = chunk * D;
@@ -445,23 +451,27 @@ __global__ void decode_attn_split_kernel(
     if (dg_ok) {
       if (chunk == DEC_CHUNK) {
         constexpr int JT = DEC_CHUNK / 16;
-        bf16x8 v8[JT];
+        constexpr int JB = 4;            // V loads in flight per batch
 #pragma unroll
-        for (int u = 0; u < JT; ++u)
-          v8[u] = nt_load8(
-              reinterpret_cast<const short*>(
-                  vb + (long)(s0 + jslot + u * 16) * D) + dgroup * 8);
+        for (int u0 = 0; u0 < JT; u0 += JB) {
+          bf16x8 v8[JB];
 #pragma unroll
-        for (int u = 0; u < JT; ++u) {
-          const int j = jslot + u * 16;
-          float vf[8];
+          for (int u = 0; u < JB; ++u)
+            v8[u] = nt_load8(
+                reinterpret_cast<const short*>(
+                    vb + (long)(s0 + jslot + (u0 + u) * 16) * D) + dgroup * 8);
 #pragma unroll
-          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
+          for (int u = 0; u < JB; ++u) {
+            const int j = jslot + (u0 + u) * 16;
+            float vf[8];
 #pragma unroll
-          for (int g = 0; g < G; ++g) {
-            const float p = p_sh[g * DEC_CHUNK + j];
+            for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
 #pragma unroll
-            for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+            for (int g = 0; g < G; ++g) {
+              const float p = p_sh[g * DEC_CHUNK + j];
+#pragma unroll
+              for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+            }
           }
         }
       } else {
