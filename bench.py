@@ -358,6 +358,10 @@ def main():
         t0 = _mark("select", t0)
         answers = generator.generate(prompts, max_new_tokens=args.gen_tokens,
                                      temperature=0.3, stop_on_eos=False)
+        stage_t["gen.prefill"] = stage_t.get("gen.prefill", 0.0) + \
+            getattr(generator, "last_prefill_s", 0.0)
+        stage_t["gen.decode"] = stage_t.get("gen.decode", 0.0) + \
+            getattr(generator, "last_decode_s", 0.0)
         t0 = _mark("generate", t0)
         # 8. optional verify
         if args.verify:

@@ -135,6 +135,8 @@ class GeneratorEngine:
         """
         if not prompts:
             return []
+        import time as _time
+
         B = len(prompts)
         prompt_budget = self.max_seq - max_new_tokens - 1
         padded, lens = self.tokenizer.encode_batch(
@@ -143,7 +145,13 @@ class GeneratorEngine:
         tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
         S = tokens.shape[1]
         sess = self._decode_session(B, self.max_seq)
+        _t0 = _time.perf_counter()
         logits = sess.prefill(tokens)
+        if self.device != "cpu":
+            torch.cuda.synchronize()
+        self.last_prefill_s = _time.perf_counter() - _t0
+        self.last_prompt_tokens = B * S
+        _t0 = _time.perf_counter()
 
         # tokens accumulate on-device; the host syncs only for EOS checks
         # (every 16 steps) or the streaming callback — not per token.
@@ -164,6 +172,10 @@ class GeneratorEngine:
                 break
             logits = sess.decode_step(cur)
             cur = self._sample(logits, temperature)
+        if self.device != "cpu":
+            torch.cuda.synchronize()
+        self.last_decode_s = _time.perf_counter() - _t0
+        self.last_decode_steps = n_steps
         rows = toks_buf[:, :n_steps].cpu().tolist()
         out = []
         for row in rows:
