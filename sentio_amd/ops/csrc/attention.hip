@@ -124,14 +124,16 @@ __global__ void flash_attn_kernel(
     // ---- stage K tile (swizzled) and V tr-image into LDS: all of this
     // thread's K and V loads issue BEFORE any LDS store (batched MLP)
     {
-      constexpr int IT = KVBLK * D / (FA_WAVES * WAVE * 8);
+      constexpr int ELEMS = KVBLK * D;
+      constexpr int STRIDE = FA_WAVES * WAVE * 8;
+      constexpr int IT = (ELEMS + STRIDE - 1) / STRIDE;   // >= 1 for D >= 32
       bf16x8_t kvals[IT], vvals[IT];
 #pragma unroll
       for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * FA_WAVES * WAVE * 8;
+        const int i = threadIdx.x * 8 + u * STRIDE;
         const int key = kv0 + i / D;
         const int d = i % D;
-        if (key < kvlen) {
+        if (i < ELEMS && key < kvlen) {
           const long base = (((long)b * S + key) * Hkv + hkv) * D + d;
           kvals[u] = *reinterpret_cast<const bf16x8_t*>(k + base);
           vvals[u] = *reinterpret_cast<const bf16x8_t*>(v + base);
@@ -143,11 +145,14 @@ __global__ void flash_attn_kernel(
       }
 #pragma unroll
       for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * FA_WAVES * WAVE * 8;
-        const int row = i / D, d = i % D;
-        *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) =
-            kvals[u];
-        *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) = vvals[u];
+        const int i = threadIdx.x * 8 + u * STRIDE;
+        if (i < ELEMS) {
+          const int row = i / D, d = i % D;
+          *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) =
+              kvals[u];
+          *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) =
+              vvals[u];
+        }
       }
     }
     __syncthreads();  // staging visible to every wave
@@ -559,20 +564,19 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
   size_t lds = (size_t)KVBLK * D * 2 * 2   // K (swizzled) + V (tr image)
                + 4 * QBLK * P_STRIDE;
   dim3 grid((S + 4 * QBLK - 1) / (4 * QBLK), H, B);
+#define FA_CASE(DV)                                                          \
+  case DV:                                                                   \
+    hipLaunchKernelGGL((flash_attn_kernel<DV>), grid, dim3(256), lds,        \
+                       stream, (const bf16*)q, (const bf16*)k,               \
+                       (const bf16*)v, (bf16*)out, kv_lens, B, S, H, Hkv, D, \
+                       scale, causal);                                       \
+    break;
   switch (D) {
-    case 64:
-      hipLaunchKernelGGL((flash_attn_kernel<64>), grid, dim3(256), lds, stream,
-                         (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                         (bf16*)out, kv_lens, B, S, H, Hkv, D, scale, causal);
-      break;
-    case 128:
-      hipLaunchKernelGGL((flash_attn_kernel<128>), grid, dim3(256), lds, stream,
-                         (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                         (bf16*)out, kv_lens, B, S, H, Hkv, D, scale, causal);
-      break;
+    FA_CASE(32) FA_CASE(64) FA_CASE(96) FA_CASE(128)
     default:
       return hipErrorInvalidValue;
   }
+#undef FA_CASE
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
