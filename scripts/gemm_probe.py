@@ -54,3 +54,20 @@ def main():
 
 if __name__ == "__main__":
     main()
+    skinny_probe()
+
+
+def skinny_probe():
+    from sentio_amd import ops
+    dev = "cuda:0"
+    print("=== skinny_gemm vs F.linear (W [N,K]) ===")
+    for B in (16, 32):
+        for name, K, N in SHAPES:
+            x = torch.randn(B, K, dtype=torch.bfloat16, device=dev)
+            wt = torch.randn(N, K, dtype=torch.bfloat16, device=dev)
+            bytes_w = K * N * 2
+            t_lin = bench(lambda: torch.nn.functional.linear(x, wt))
+            t_sk = bench(lambda: ops.skinny_gemm(x, wt))
+            print(f"B={B} {name:8s} linear {t_lin*1e6:7.1f}us "
+                  f"({bytes_w/t_lin/1e12:.2f} TB/s)  skinny {t_sk*1e6:7.1f}us "
+                  f"({bytes_w/t_sk/1e12:.2f} TB/s)")

@@ -177,6 +177,15 @@ def sample_token(logits, temperature: float, seed: int = 0):
     return torch_ref.sample_token(logits, temperature, g)
 
 
+def skinny_gemm(x, w):
+    """Skinny decode GEMM: x [M<=32, K] @ W^T with W stored [N, K] row-major
+    (TN layout) → [M, N] bf16.  Streams W rows straight to MFMA A-fragments.
+    CPU path: fp32 linear."""
+    if _on_gpu(x):
+        return _require_hip().skinny_gemm(x.contiguous(), w.contiguous())
+    return torch.nn.functional.linear(x.float(), w.float()).to(x.dtype)
+
+
 def gemm_bf16(a, b):
     """Hand-written MFMA bf16 GEMM: a [M,K] @ b [K,N] → [M,N] bf16.
     CPU path: fp32 matmul."""
@@ -188,5 +197,5 @@ def gemm_bf16(a, b):
 __all__ = [
     "rmsnorm", "rmsnorm_residual", "rope_apply", "decode_qkv_prep", "swiglu", "swiglu_packed", "softmax",
     "attention", "decode_attention", "mean_pool_l2norm", "cosine_topk",
-    "bm25_score", "sample_token", "gemm_bf16", "hip_available", "torch_ref",
+    "bm25_score", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
 ]

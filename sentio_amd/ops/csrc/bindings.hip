@@ -38,6 +38,8 @@ hipError_t sentio_decode_attn(const void*, const void*, const void*, void*,
                               int, int, float, hipStream_t);
 hipError_t sentio_gemm_bf16(const void*, const void*, void*, int, int, int,
                             hipStream_t);
+hipError_t sentio_skinny_gemm(const void*, const void*, void*, int, int, int,
+                              hipStream_t);
 }
 
 namespace {
@@ -281,6 +283,18 @@ torch::Tensor gemm_bf16(torch::Tensor a, torch::Tensor b) {
 
 }  // namespace
 
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
+  check_bf16_cuda(x, "x");
+  check_bf16_cuda(w, "w");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "x [M,K], w [N,K]");
+  TORCH_CHECK(x.size(1) == w.size(1), "K mismatch");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  auto out = torch::empty({M, N}, x.options());
+  check_hip(sentio_skinny_gemm(x.data_ptr(), w.data_ptr(), out.data_ptr(),
+                               M, K, N, stream()), "skinny_gemm");
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_residual", &rmsnorm_residual);
@@ -296,4 +310,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn", &flash_attn);
   m.def("decode_attn", &decode_attn);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("skinny_gemm", &skinny_gemm);
 }

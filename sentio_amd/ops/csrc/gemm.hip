@@ -154,6 +154,116 @@ __global__ void gemm_bf16_kernel(const bf16* __restrict__ A,
   }
 }
 
+
+// ------------------------------------------------------ skinny decode GEMM
+// C[M,N] = x[M,K] @ W^T with W stored [N,K] row-major (the engine's TN
+// weight layout) and M <= 32 (decode batches).  Weight-bandwidth-bound:
+// W rows stream STRAIGHT to VGPRs as MFMA A-fragments (guide §6: for
+// M <= 32 decode weights "load straight to VGPRs, deep unroll, late
+// vmcnt" — an LDS round trip is pure overhead), x is tiny and L2-resident.
+// Computes C^T tiles: D-frag row = n, col = m.
+// Grid: (N/64); block 256 = 4 waves, wave w owns 16 consecutive n.
+// UNR k-steps (32 deep each) are unrolled with all A loads issued first.
+template <int MT>   // padded M: 16 or 32
+__launch_bounds__(256, 1)
+__global__ void skinny_gemm_kernel(const bf16* __restrict__ x,
+                                   const bf16* __restrict__ w,
+                                   bf16* __restrict__ out,
+                                   int M, int K, int N) {
+  constexpr int MTILES = MT / 16;
+  constexpr int UNR = 8;              // 8 x 32 = 256-deep k per iteration
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int n0 = (blockIdx.x * 4 + wid) * 16;
+  if (n0 >= N) return;
+
+  const int arow = lane & 15;         // n offset within the wave's strip
+  const int kofs = (lane >> 4) * 8;   // k octet of the fragment
+
+  f32x4_t acc[MTILES];
+#pragma unroll
+  for (int mt = 0; mt < MTILES; ++mt) acc[mt] = f32x4_t{};
+
+  const bf16* wrow = w + (long)(n0 + arow) * K;      // this lane's W row
+  // x fragment rows: col m = arow (same lane mapping); guard m < M
+  const bool m_ok0 = arow < M;
+  const bf16* xrow0 = x + (long)(m_ok0 ? arow : 0) * K;
+  const bool m_ok1 = MTILES > 1 && (16 + arow) < M;
+  const bf16* xrow1 = x + (long)(m_ok1 ? 16 + arow : 0) * K;
+
+  int k0 = 0;
+  for (; k0 + UNR * 32 <= K; k0 += UNR * 32) {
+    bf16x8_t a[UNR];
+#pragma unroll
+    for (int u = 0; u < UNR; ++u)
+      a[u] = *reinterpret_cast<const bf16x8_t*>(wrow + k0 + u * 32 + kofs);
+    bf16x8_t b0[UNR], b1[MTILES > 1 ? UNR : 1];
+#pragma unroll
+    for (int u = 0; u < UNR; ++u) {
+      b0[u] = *reinterpret_cast<const bf16x8_t*>(xrow0 + k0 + u * 32 + kofs);
+      if (MTILES > 1)
+        b1[u] = *reinterpret_cast<const bf16x8_t*>(xrow1 + k0 + u * 32 + kofs);
+    }
+    if (!m_ok0) {
+#pragma unroll
+      for (int u = 0; u < UNR; ++u) { bf16x8_t z = {}; b0[u] = z; }
+    }
+    if (MTILES > 1 && !m_ok1) {
+#pragma unroll
+      for (int u = 0; u < UNR; ++u) { bf16x8_t z = {}; b1[u] = z; }
+    }
+#pragma unroll
+    for (int u = 0; u < UNR; ++u) {
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b0[u], acc[0],
+                                                       0, 0, 0);
+      if (MTILES > 1)
+        acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u], b1[u], acc[1],
+                                                         0, 0, 0);
+    }
+  }
+  for (; k0 < K; k0 += 32) {          // K tail (K % 256)
+    bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(wrow + k0 + kofs);
+    bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(xrow0 + k0 + kofs);
+    if (!m_ok0) { bf16x8_t z = {}; b0 = z; }
+    acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc[0], 0, 0, 0);
+    if (MTILES > 1) {
+      bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(xrow1 + k0 + kofs);
+      if (!m_ok1) { bf16x8_t z = {}; b1 = z; }
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc[1], 0, 0, 0);
+    }
+  }
+
+  // D-frag: lane l reg r -> n = n0 + (l>>4)*4 + r, m = mt*16 + (l&15)
+#pragma unroll
+  for (int mt = 0; mt < MTILES; ++mt) {
+    const int m = mt * 16 + arow;
+    if (m >= M) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int n = n0 + (lane >> 4) * 4 + r;
+      if (n < N) out[(long)m * N + n] = f2bf(acc[mt][r]);
+    }
+  }
+}
+
+extern "C" {
+
+hipError_t sentio_skinny_gemm(const void* x, const void* w, void* out, int M,
+                              int K, int N, hipStream_t stream) {
+  if (M > 32 || (K % 32)) return hipErrorInvalidValue;
+  dim3 grid((N + 63) / 64);
+  if (M <= 16)
+    hipLaunchKernelGGL((skinny_gemm_kernel<16>), grid, dim3(256), 0, stream,
+                       (const bf16*)x, (const bf16*)w, (bf16*)out, M, K, N);
+  else
+    hipLaunchKernelGGL((skinny_gemm_kernel<32>), grid, dim3(256), 0, stream,
+                       (const bf16*)x, (const bf16*)w, (bf16*)out, M, K, N);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+}  // extern "C"
+
 extern "C" {
 
 hipError_t sentio_gemm_bf16(const void* A, const void* B, void* C, int M,

@@ -334,3 +334,16 @@ def test_swiglu_packed(dev):
     got = ops.swiglu_packed(gu)
     want = ops.torch_ref.swiglu_packed(gu.cpu().float())
     _cmp(got, want, rtol=2e-2, atol=2e-2)
+
+
+def test_skinny_gemm(dev):
+    from sentio_amd import ops
+
+    torch.manual_seed(9)
+    for M, K, N in [(16, 128, 256), (32, 256, 192), (7, 4096, 512),
+                    (32, 4096, 28672)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.5
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.5
+        got = ops.skinny_gemm(x, w)
+        want = torch.nn.functional.linear(x.cpu().float(), w.cpu().float())
+        _cmp(got, want, rtol=3e-2, atol=3e-1)
