@@ -52,11 +52,6 @@ def main():
                   f"x@Wt.t() {t_tv*1e6:7.1f}us ({bytes_w/t_tv/1e12:.2f} TB/s)")
 
 
-if __name__ == "__main__":
-    main()
-    skinny_probe()
-
-
 def skinny_probe():
     from sentio_amd import ops
     dev = "cuda:0"
@@ -71,3 +66,8 @@ def skinny_probe():
             print(f"B={B} {name:8s} linear {t_lin*1e6:7.1f}us "
                   f"({bytes_w/t_lin/1e12:.2f} TB/s)  skinny {t_sk*1e6:7.1f}us "
                   f"({bytes_w/t_sk/1e12:.2f} TB/s)")
+
+
+if __name__ == "__main__":
+    main()
+    skinny_probe()
