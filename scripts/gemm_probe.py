@@ -1,4 +1,6 @@
 """Probe decode-shaped skinny GEMM layouts on MI355X.
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 Decode GEMMs at batch B read the whole weight once per token — they are
 weight-bandwidth-bound.  Compares: (a) x @ W  ([K,N] row-major, the current

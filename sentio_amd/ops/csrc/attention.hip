@@ -325,7 +325,9 @@ __global__ void decode_attn_split_kernel(
   const int b = blockIdx.y;
   const int split = blockIdx.z;
   const int slen = seq_lens[b];
-  const int span = (Smax + splits - 1) / splits;
+  // split the VALID range (not Smax): splitting by the allocation left the
+  // low splits with full spans and the tail split nearly idle
+  const int span = (slen + splits - 1) / splits;
   const int s_begin = split * span;
   const int s_end = min(slen, s_begin + span);
   const bf16* kb = kc + ((long)b * Hkv + hkv) * Smax * (long)D;
@@ -365,7 +367,7 @@ __global__ void decode_attn_split_kernel(
     // ds_write (the dynamic loop serializes one load per s_waitcnt).
     if (chunk == DEC_CHUNK) {
       constexpr int IT = D / 8;          // loads per thread for a full chunk
-      constexpr int BATCH = 4;           // loads in flight (16-deep spilled)
+      constexpr int BATCH = 8;           // loads in flight
 #pragma unroll
       for (int u0 = 0; u0 < IT; u0 += BATCH) {
         bf16x8 tmp[BATCH];
@@ -456,7 +458,7 @@ __global__ void decode_attn_split_kernel(
     if (dg_ok) {
       if (chunk == DEC_CHUNK) {
         constexpr int JT = DEC_CHUNK / 16;
-        constexpr int JB = 4;            // V loads in flight per batch
+        constexpr int JB = 8;            // V loads in flight per batch
 #pragma unroll
         for (int u0 = 0; u0 < JT; u0 += JB) {
           bf16x8 v8[JB];
