@@ -1,6 +1,4 @@
 """Probe decode-shaped skinny GEMM layouts on MI355X.
-import sys, os
-sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 Decode GEMMs at batch B read the whole weight once per token — they are
 weight-bandwidth-bound.  Compares: (a) x @ W  ([K,N] row-major, the current
@@ -9,7 +7,11 @@ for the llama3-8b per-layer shapes + lm_head.  Prints achieved TB/s.
 Run: gpurun -- 'python scripts/gemm_probe.py'
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
