@@ -130,9 +130,15 @@ def create_generator_node(generator, mode: str = "balanced",
             from sentio_amd.engines.generator import MODE_TEMPERATURE
 
             temperature = MODE_TEMPERATURE.get(mode, 0.3)
-        prompt = builder.system_prompt() + "\n\n" + builder.build_qa_prompt(
-            state["query"], context, mode
+        history = state.get("metadata", {}).get("history") or []
+        hist_txt = "".join(
+            f"{turn.get('role', 'user')}: {turn.get('content', '')}\n"
+            for turn in history[-6:] if isinstance(turn, dict)
         )
+        prompt = builder.system_prompt() + "\n\n"
+        if hist_txt:
+            prompt += "Conversation so far:\n" + hist_txt + "\n"
+        prompt += builder.build_qa_prompt(state["query"], context, mode)
         try:
             answer = generator.generate(
                 [prompt], max_new_tokens=max_tokens, temperature=float(temperature)

@@ -167,7 +167,7 @@ def create_app(settings: Settings | None = None,
         with metrics_collector.track_request("/chat"):
             t0 = time.perf_counter()
             result = await _run_sync(chat_handler.process, req.question,
-                                     req.top_k, req.temperature)
+                                     req.top_k, req.temperature, req.history)
             performance_monitor.record_value(
                 "chat_latency_ms", (time.perf_counter() - t0) * 1e3)
             return result

@@ -25,13 +25,16 @@ class ChatHandler:
         self.cache: CacheManager = container.cache_manager()
 
     def process(self, question: str, top_k: int | None = None,
-                temperature: float | None = None) -> dict[str, Any]:
+                temperature: float | None = None,
+                history: list[dict[str, str]] | None = None) -> dict[str, Any]:
         query_id = str(uuid.uuid4())
         meta: dict[str, Any] = {"query_id": query_id}
         if top_k is not None:
             meta["user_top_k"] = int(top_k)
         if temperature is not None:
             meta["temperature"] = float(temperature)
+        if history:
+            meta["history"] = history
 
         cached = self.cache.l1.get_query(question)
         try:
