@@ -363,20 +363,27 @@ __global__ void decode_attn_split_kernel(
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
     // phase A1: cooperative coalesced K staging into XOR-swizzled LDS.
-    // Full chunks take the unrolled path: 8 loads in flight before any
-    // ds_write (the dynamic loop serializes one load per s_waitcnt).
-    if (chunk == DEC_CHUNK) {
+    // One predicated unrolled path for EVERY chunk (a separate dynamic tail
+    // loop serialized one load per s_waitcnt and dominated short spans):
+    // 8 loads in flight before any ds_write; out-of-range rows stage zeros.
+    {
       constexpr int IT = D / 8;          // loads per thread for a full chunk
       constexpr int BATCH = 8;           // loads in flight
+      const int lim = chunk * D;
 #pragma unroll
       for (int u0 = 0; u0 < IT; u0 += BATCH) {
         bf16x8 tmp[BATCH];
 #pragma unroll
         for (int u = 0; u < BATCH; ++u) {
           const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
-          tmp[u] = nt_load8(
-              reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
-              + (i % D));
+          if (i < lim) {
+            tmp[u] = nt_load8(
+                reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
+                + (i % D));
+          } else {
+            bf16x8 z = {};
+            tmp[u] = z;
+          }
         }
 #pragma unroll
         for (int u = 0; u < BATCH; ++u) {
@@ -385,16 +392,6 @@ __global__ void decode_attn_split_kernel(
           *reinterpret_cast<bf16x8*>(
               k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = tmp[u];
         }
-      }
-    } else {
-      const int elems = chunk * D;
-      for (int i = threadIdx.x * 8; i < elems; i += DEC_CHUNK * 8) {
-        const int row = i / D;
-        const int d = i % D;
-        bf16x8 v = nt_load8(
-            reinterpret_cast<const short*>(kb + (long)(s0 + row) * D) + d);
-        *reinterpret_cast<bf16x8*>(
-            k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = v;
       }
     }
     __syncthreads();
@@ -456,39 +453,30 @@ __global__ void decode_attn_split_kernel(
 #pragma unroll
       for (int e = 0; e < 8; ++e) o_part[g][e] *= alpha[g];
     if (dg_ok) {
-      if (chunk == DEC_CHUNK) {
-        constexpr int JT = DEC_CHUNK / 16;
-        constexpr int JB = 8;            // V loads in flight per batch
+      constexpr int JT = DEC_CHUNK / 16;
+      constexpr int JB = 8;              // V loads in flight per batch
 #pragma unroll
-        for (int u0 = 0; u0 < JT; u0 += JB) {
-          bf16x8 v8[JB];
+      for (int u0 = 0; u0 < JT; u0 += JB) {
+        bf16x8 v8[JB];
 #pragma unroll
-          for (int u = 0; u < JB; ++u)
+        for (int u = 0; u < JB; ++u) {
+          const int j = jslot + (u0 + u) * 16;
+          if (j < chunk) {
             v8[u] = nt_load8(
-                reinterpret_cast<const short*>(
-                    vb + (long)(s0 + jslot + (u0 + u) * 16) * D) + dgroup * 8);
-#pragma unroll
-          for (int u = 0; u < JB; ++u) {
-            const int j = jslot + (u0 + u) * 16;
-            float vf[8];
-#pragma unroll
-            for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
-#pragma unroll
-            for (int g = 0; g < G; ++g) {
-              const float p = p_sh[g * DEC_CHUNK + j];
-#pragma unroll
-              for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
-            }
+                reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
+                dgroup * 8);
+          } else {
+            bf16x8 z = {};
+            v8[u] = z;
           }
         }
-      } else {
-        for (int j = jslot; j < chunk; j += 16) {
-          bf16x8 v8 = nt_load8(
-              reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
-              dgroup * 8);
+#pragma unroll
+        for (int u = 0; u < JB; ++u) {
+          const int j = jslot + (u0 + u) * 16;
+          if (j >= chunk) continue;     // p is 0 there anyway
           float vf[8];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[e]);
+          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
 #pragma unroll
           for (int g = 0; g < G; ++g) {
             const float p = p_sh[g * DEC_CHUNK + j];
