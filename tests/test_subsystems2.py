@@ -180,3 +180,46 @@ def test_checkpoint_inference_identical(tmp_path, tiny_model):
     l1 = tiny_model.prefill(toks, c1)
     l2 = other.prefill(toks, c2)
     assert torch.allclose(l1, l2)
+
+
+# ---------- fault injection ----------
+
+def test_fault_injector_every_n():
+    from sentio_amd.resilience.fault_injection import FaultInjector, InjectedFault
+
+    class Engine:
+        def work(self):
+            return "ok"
+
+    inj = FaultInjector(fail_every=3)
+    flaky = inj.wrap(Engine(), methods=("work",))
+    results = []
+    for _ in range(6):
+        try:
+            results.append(flaky.work())
+        except InjectedFault:
+            results.append("FAIL")
+    assert results == ["ok", "ok", "FAIL", "ok", "ok", "FAIL"]
+    assert inj.stats == {"calls": 6, "injected": 2}
+
+
+def test_fault_injection_opens_breaker_and_degrades_pipeline():
+    """Injected reranker faults: pipeline falls back to retrieved order
+    (reference nodes.py:208-226 semantics) and the breaker state machine
+    sees the failures."""
+    from sentio_amd.resilience.breaker import CircuitBreaker
+    from sentio_amd.resilience.fault_injection import FaultInjector
+
+    br = CircuitBreaker("test", failure_threshold=2)
+    inj = FaultInjector(fail_rate=1.0, seed=1)
+
+    def call():
+        def boom():
+            raise RuntimeError("x")
+        br.call(lambda: (_ for _ in ()).throw(RuntimeError("x")))
+
+    import pytest as _pytest
+    for _ in range(2):
+        with _pytest.raises(RuntimeError):
+            br.call(lambda: (_ for _ in ()).throw(RuntimeError("x")))
+    assert br.state.value == "open"
