@@ -101,3 +101,20 @@ def main() -> None:
 
 if __name__ == "__main__":
     main()
+
+
+@app.command("run")
+def run_all(host: str = typer.Option(None), port: int = typer.Option(None)):
+    """Start the full stack: API server (the built-in /ui page is the UI —
+    reference `sentio run all` spawned API + Streamlit + Qdrant docker;
+    here everything is in-process)."""
+    import uvicorn
+
+    from sentio_amd.config import settings
+    from sentio_amd.serving.app import create_app
+
+    resolved_port = port or settings.api_port
+    typer.echo(f"serving API + UI on http://{host or settings.api_host}:"
+               f"{resolved_port} (UI at /ui)")
+    uvicorn.run(create_app(), host=host or settings.api_host,
+                port=resolved_port)

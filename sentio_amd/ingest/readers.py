@@ -37,6 +37,14 @@ def read_file(path: Path) -> Document | None:
 
             data = yaml.safe_load(path.read_text(errors="replace"))
             text = json.dumps(data, indent=1, default=str)
+        elif suffix == ".pdf":
+            text = _read_pdf(path)
+            if text is None:
+                return None
+        elif suffix == ".docx":
+            text = _read_docx(path)
+            if text is None:
+                return None
         else:
             return None
     except Exception as exc:
@@ -59,3 +67,36 @@ def read_directory(directory: str, recursive: bool = True) -> list[Document]:
             if doc is not None:
                 docs.append(doc)
     return docs
+
+
+def _read_pdf(path: Path) -> str | None:
+    """PDF text extraction (reference ingest.py:172-223 used PyPDF2).
+    PyPDF2 is absent from this deployment image; gated import keeps the
+    capability wired without a hard dependency."""
+    try:
+        import PyPDF2  # type: ignore
+    except ImportError:
+        logger.warning("PyPDF2 not installed; skipping PDF %s", path)
+        return None
+    try:
+        with open(path, "rb") as f:
+            reader = PyPDF2.PdfReader(f)
+            return "\n".join(page.extract_text() or "" for page in reader.pages)
+    except Exception as exc:
+        logger.warning("failed to read PDF %s: %s", path, exc)
+        return None
+
+
+def _read_docx(path: Path) -> str | None:
+    """DOCX extraction (reference ingest.py used python-docx); gated."""
+    try:
+        import docx  # type: ignore
+    except ImportError:
+        logger.warning("python-docx not installed; skipping DOCX %s", path)
+        return None
+    try:
+        d = docx.Document(str(path))
+        return "\n".join(p.text for p in d.paragraphs)
+    except Exception as exc:
+        logger.warning("failed to read DOCX %s: %s", path, exc)
+        return None

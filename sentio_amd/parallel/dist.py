@@ -52,6 +52,17 @@ def barrier() -> None:
         dist.barrier()
 
 
+def _account(op: str, nbytes: int) -> None:
+    """RCCL traffic counters (surfaced as rccl_bytes_total in /metrics —
+    the GPU-era analogue of the reference's HTTP client metrics)."""
+    try:
+        from sentio_amd.observability.metrics import metrics_collector
+
+        metrics_collector.inc("rccl_bytes_total", float(nbytes), op=op)
+    except Exception:
+        pass
+
+
 def all_gather_tensor(t: torch.Tensor) -> torch.Tensor:
     """Concatenate equal-shaped tensors from all ranks along dim 0."""
     if not is_distributed():
@@ -59,12 +70,14 @@ def all_gather_tensor(t: torch.Tensor) -> torch.Tensor:
     world = get_world_size()
     out = [torch.empty_like(t) for _ in range(world)]
     dist.all_gather(out, t.contiguous())
+    _account("all_gather", t.nelement() * t.element_size() * world)
     return torch.cat(out, dim=0)
 
 
 def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
     if is_distributed():
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        _account("all_reduce", t.nelement() * t.element_size())
     return t
 
 
