@@ -95,14 +95,6 @@ def chat_once(question: str = typer.Argument(...)):
     typer.echo(json.dumps(result, indent=2))
 
 
-def main() -> None:
-    app()
-
-
-if __name__ == "__main__":
-    main()
-
-
 @app.command("run")
 def run_all(host: str = typer.Option(None), port: int = typer.Option(None)):
     """Start the full stack: API server (the built-in /ui page is the UI —
@@ -118,3 +110,11 @@ def run_all(host: str = typer.Option(None), port: int = typer.Option(None)):
                f"{resolved_port} (UI at /ui)")
     uvicorn.run(create_app(), host=host or settings.api_host,
                 port=resolved_port)
+
+
+def main() -> None:
+    app()
+
+
+if __name__ == "__main__":
+    main()
