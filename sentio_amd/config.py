@@ -100,6 +100,8 @@ class Settings:
     index_shards: int = 1                   # = world size when distributed
     kv_cache_max_tokens: int = 8192
     max_batch_size: int = 32
+    dynamic_batching: bool = True           # batch concurrent /chat generations
+    batch_wait_ms: float = 8.0
     device: str = "auto"                    # auto | cuda | cpu
 
     # --- observability ---
@@ -145,6 +147,8 @@ class Settings:
         "index_shards": "INDEX_SHARDS",
         "kv_cache_max_tokens": "KV_CACHE_MAX_TOKENS",
         "max_batch_size": "MAX_BATCH_SIZE",
+        "dynamic_batching": "DYNAMIC_BATCHING",
+        "batch_wait_ms": "BATCH_WAIT_MS",
         "device": "SENTIO_DEVICE",
         "enable_metrics": "ENABLE_METRICS",
         "enable_tracing": "ENABLE_TRACING",

@@ -83,6 +83,9 @@ class GeneratorEngine:
         self.model = Transformer(self.cfg, device=device, dtype=dtype,
                                  seed=seed, tp=tp)
         self._step_seed = 0
+        import threading
+
+        self._gen_lock = threading.Lock()   # decode sessions hold static state
 
     def _new_cache(self, batch: int, max_seq: int) -> KVCache:
         return KVCache(self.cfg, batch, max_seq, self.device, self.model.dtype,
@@ -137,6 +140,15 @@ class GeneratorEngine:
             return []
         import time as _time
 
+        self._gen_lock.acquire()
+        try:
+            return self._generate_locked(prompts, max_new_tokens, temperature,
+                                         stop_on_eos, on_token, _time)
+        finally:
+            self._gen_lock.release()
+
+    def _generate_locked(self, prompts, max_new_tokens, temperature,
+                         stop_on_eos, on_token, _time) -> list[str]:
         B = len(prompts)
         prompt_budget = self.max_seq - max_new_tokens - 1
         padded, lens = self.tokenizer.encode_batch(

@@ -15,6 +15,7 @@ reference src/core/embeddings/providers/jina.py:165)."""
 from __future__ import annotations
 
 import os
+import threading
 from typing import Callable
 
 import torch
@@ -40,6 +41,7 @@ class GraphedForward:
     def __init__(self, fn: Callable, batch: int, seq: int, device: str):
         self.fn = fn
         self.device = device
+        self._lock = threading.Lock()   # static buffers: one replay at a time
         self.tokens = torch.zeros(batch, seq, dtype=torch.int64, device=device)
         self.kv_lens = torch.ones(batch, dtype=torch.int32, device=device)
         self.graph: torch.cuda.CUDAGraph | None = None
@@ -57,6 +59,7 @@ class GraphedForward:
             self.out = self.fn(self.tokens, self.kv_lens)
 
     def run(self, tokens: torch.Tensor, kv_lens: torch.Tensor) -> torch.Tensor:
+      with self._lock:
         if self.graph is None:
             self._capture()
         b = tokens.shape[0]
@@ -68,7 +71,8 @@ class GraphedForward:
             self.kv_lens[b:].fill_(1)            # keep padded rows valid
         self.kv_lens[:b].copy_(kv_lens)
         self.graph.replay()
-        return self.out[:b]
+        # clone: the caller may hold the result across the next replay
+        return self.out[:b].clone()
 
 
 class GraphedEnginePool:

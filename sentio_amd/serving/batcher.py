@@ -1,0 +1,156 @@
+"""Dynamic request batching for the generation engine.
+
+The reference served each request's LLM call as its own HTTP round trip to
+a provider that batched internally (reference src/core/llm/providers/
+openai.py:117).  On-device, concurrent `/chat` requests must be batched by
+US: decode is weight-bandwidth-bound, so a batch of 32 decodes costs barely
+more than a batch of 1 (the 16 GB of weights stream once per token either
+way).  This batcher turns per-request `generate()` calls into shared
+batches:
+
+* callers enqueue (prompt, params) and block on a future;
+* a worker thread drains the queue, groups requests by compatible sampling
+  params (temperature bucket, max_tokens), waits up to `max_wait_ms` for
+  stragglers, and runs ONE batched `generator.generate`;
+* results fan back out to the callers.
+
+Used by ChatHandler when `settings.dynamic_batching` is on; the pipeline
+path is unchanged (it already batches within one bench step).
+"""
+
+from __future__ import annotations
+
+import queue
+import threading
+import time
+from concurrent.futures import Future
+from dataclasses import dataclass, field
+from typing import Any
+
+
+@dataclass
+class _Item:
+    prompt: str
+    max_new_tokens: int
+    temperature: float
+    stop_on_eos: bool
+    future: Future = field(default_factory=Future)
+
+    @property
+    def group_key(self) -> tuple:
+        # temperature bucketed to 0.1 — sampling is per-batch in the engine
+        return (round(self.temperature, 1), self.max_new_tokens,
+                self.stop_on_eos)
+
+
+class DynamicBatcher:
+    def __init__(self, generator, max_batch: int = 32,
+                 max_wait_ms: float = 8.0):
+        self.generator = generator
+        self.max_batch = max_batch
+        self.max_wait_s = max_wait_ms / 1e3
+        self._q: queue.Queue[_Item] = queue.Queue()
+        self._stop = threading.Event()
+        self._thread: threading.Thread | None = None
+        self.stats = {"requests": 0, "batches": 0, "max_batch_seen": 0}
+
+    # ----- caller side -----
+    def generate(self, prompt: str, max_new_tokens: int = 128,
+                 temperature: float = 0.3, stop_on_eos: bool = True,
+                 timeout_s: float = 300.0) -> str:
+        self.start()
+        item = _Item(prompt, max_new_tokens, float(temperature), stop_on_eos)
+        self._q.put(item)
+        return item.future.result(timeout=timeout_s)
+
+    # ----- worker side -----
+    def start(self) -> None:
+        if self._thread is not None and self._thread.is_alive():
+            return
+        self._stop.clear()
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="sentio-batcher")
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=2.0)
+            self._thread = None
+
+    def _loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                first = self._q.get(timeout=0.1)
+            except queue.Empty:
+                continue
+            batch = [first]
+            deadline = time.monotonic() + self.max_wait_s
+            leftovers: list[_Item] = []
+            while len(batch) < self.max_batch:
+                remaining = deadline - time.monotonic()
+                if remaining <= 0:
+                    break
+                try:
+                    nxt = self._q.get(timeout=remaining)
+                except queue.Empty:
+                    break
+                if nxt.group_key == first.group_key:
+                    batch.append(nxt)
+                else:
+                    leftovers.append(nxt)   # different params: next batch
+            for item in leftovers:
+                self._q.put(item)
+            self._run_batch(batch)
+
+    def _run_batch(self, batch: list[_Item]) -> None:
+        self.stats["requests"] += len(batch)
+        self.stats["batches"] += 1
+        self.stats["max_batch_seen"] = max(self.stats["max_batch_seen"],
+                                           len(batch))
+        try:
+            outs = self.generator.generate(
+                [it.prompt for it in batch],
+                max_new_tokens=batch[0].max_new_tokens,
+                temperature=batch[0].temperature,
+                stop_on_eos=batch[0].stop_on_eos,
+            )
+            for it, out in zip(batch, outs):
+                it.future.set_result(out)
+        except Exception as exc:
+            for it in batch:
+                if not it.future.done():
+                    it.future.set_exception(exc)
+
+    def health(self) -> dict[str, Any]:
+        return {"queued": self._q.qsize(),
+                "running": self._thread is not None and self._thread.is_alive(),
+                **self.stats}
+
+
+class BatchedGenerator:
+    """Drop-in generator frontend: single-prompt calls route through the
+    shared DynamicBatcher; already-batched calls pass straight through."""
+
+    def __init__(self, raw, max_batch: int = 32, max_wait_ms: float = 8.0):
+        self.raw = raw
+        self.batcher = DynamicBatcher(raw, max_batch=max_batch,
+                                      max_wait_ms=max_wait_ms)
+
+    def generate(self, prompts: list[str], max_new_tokens: int = 128,
+                 temperature: float = 0.3, stop_on_eos: bool = True,
+                 **kwargs) -> list[str]:
+        if len(prompts) != 1 or kwargs.get("on_token") is not None:
+            return self.raw.generate(prompts, max_new_tokens=max_new_tokens,
+                                     temperature=temperature,
+                                     stop_on_eos=stop_on_eos, **kwargs)
+        return [self.batcher.generate(prompts[0],
+                                      max_new_tokens=max_new_tokens,
+                                      temperature=temperature,
+                                      stop_on_eos=stop_on_eos)]
+
+    def stream(self, *args, **kwargs):
+        return self.raw.stream(*args, **kwargs)
+
+    def __getattr__(self, name):
+        return getattr(self.raw, name)

@@ -89,6 +89,23 @@ class ServiceContainer:
 
         return self._get("generator", make)
 
+    def generator_frontend(self):
+        """The generator the pipeline sees: wraps the raw engine in the
+        dynamic batcher so concurrent /chat requests share decode batches
+        (weight-bandwidth amortization; no reference counterpart — the
+        reference's provider batched server-side)."""
+        def make():
+            raw = self.generator()
+            if not self.settings.dynamic_batching:
+                return raw
+            from sentio_amd.serving.batcher import BatchedGenerator
+
+            return BatchedGenerator(raw,
+                                    max_batch=self.settings.max_batch_size,
+                                    max_wait_ms=self.settings.batch_wait_ms)
+
+        return self._get("generator_frontend", make)
+
     # --- indexes ---
     def dense_index(self) -> DenseIndex:
         return self._get("dense_index", lambda: DenseIndex(
@@ -116,7 +133,7 @@ class ServiceContainer:
                 self.settings,
                 retriever=self.retriever(),
                 reranker=self.reranker() if self.settings.use_reranker else None,
-                generator=self.generator(),
+                generator=self.generator_frontend(),
                 verifier=self.verifier() if self.settings.use_verifier else None,
             )
             return build_basic_graph(cfg)

@@ -126,3 +126,52 @@ def test_ui_page(client):
     r = client.get("/ui")
     assert r.status_code == 200
     assert "sentio-amd" in r.text and "/chat" in r.text
+
+
+def test_dynamic_batcher_coalesces_concurrent_requests():
+    """Concurrent generate() calls share one engine batch
+    (weight-bandwidth amortization on device)."""
+    import threading
+
+    from sentio_amd.serving.batcher import DynamicBatcher
+
+    class SlowMock:
+        def __init__(self):
+            self.calls = []
+
+        def generate(self, prompts, **kw):
+            import time
+            time.sleep(0.02)
+            self.calls.append(len(prompts))
+            return [f"ans:{p}" for p in prompts]
+
+    eng = SlowMock()
+    b = DynamicBatcher(eng, max_batch=8, max_wait_ms=40)
+    results = {}
+
+    def worker(i):
+        results[i] = b.generate(f"q{i}", max_new_tokens=8, temperature=0.3)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=10)
+    b.stop()
+    assert all(results[i] == f"ans:q{i}" for i in range(8))
+    assert len(eng.calls) < 8          # fewer engine calls than requests
+    assert max(eng.calls) > 1          # at least one real batch
+    assert b.stats["requests"] == 8
+
+
+def test_batched_generator_passthrough_for_batches():
+    from sentio_amd.serving.batcher import BatchedGenerator
+
+    class Mock:
+        def generate(self, prompts, **kw):
+            return [p.upper() for p in prompts]
+
+    bg = BatchedGenerator(Mock(), max_batch=4, max_wait_ms=5)
+    assert bg.generate(["a", "b"]) == ["A", "B"]      # passthrough
+    assert bg.generate(["solo"]) == ["SOLO"]          # via batcher
+    bg.batcher.stop()
