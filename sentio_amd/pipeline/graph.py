@@ -17,6 +17,7 @@ from typing import Callable
 
 from sentio_amd.config import Settings
 from sentio_amd.pipeline import nodes as N
+from sentio_amd.observability.tracing import trace_operation
 from sentio_amd.pipeline.state import RAGState, add_metadata
 
 logger = logging.getLogger(__name__)
@@ -62,12 +63,15 @@ class RagPipeline:
 
     def invoke(self, state: RAGState) -> RAGState:
         t0 = time.perf_counter()
-        for name, fn in self.stages:
-            try:
-                state = fn(state)
-            except Exception as exc:
-                logger.error("stage %s raised: %s", name, exc)
-                add_metadata(state, f"{name}_error", str(exc))
+        with trace_operation("pipeline.invoke",
+                             query_id=state.get("metadata", {}).get("query_id")):
+            for name, fn in self.stages:
+                try:
+                    with trace_operation(f"stage.{name}"):
+                        state = fn(state)
+                except Exception as exc:
+                    logger.error("stage %s raised: %s", name, exc)
+                    add_metadata(state, f"{name}_error", str(exc))
         add_metadata(state, "pipeline_ms", (time.perf_counter() - t0) * 1e3)
         return state
 

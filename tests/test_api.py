@@ -175,3 +175,29 @@ def test_batched_generator_passthrough_for_batches():
     assert bg.generate(["a", "b"]) == ["A", "B"]      # passthrough
     assert bg.generate(["solo"]) == ["SOLO"]          # via batcher
     bg.batcher.stop()
+
+
+def test_auth_protected_endpoint_flow():
+    """With DISABLE_AUTH off, the auth manager guards scopes end to end
+    (reference auth.py:444-470 dependency-guard capability)."""
+    from sentio_amd.utils.auth import AuthManager, AuthScope, AuthError, UserRole
+
+    mgr = AuthManager(secret="test-secret")
+    token = mgr.issue_token("alice", role=UserRole.WRITER)
+    data = mgr.require_scopes(token, AuthScope.CHAT, AuthScope.EMBED)
+    assert data.subject == "alice"
+    reader_token = mgr.issue_token("bob", role=UserRole.READER)
+    import pytest as _pytest
+    with _pytest.raises(AuthError):
+        mgr.require_scopes(reader_token, AuthScope.EMBED)
+
+
+def test_tracing_spans_recorded_on_chat(client):
+    from sentio_amd.observability import tracing
+
+    tracing.clear_spans()
+    r = client.post("/chat", json={"question": "what is a span?"})
+    assert r.status_code == 200
+    names = [s["name"] for s in tracing.recent_spans()]
+    assert "pipeline.invoke" in names
+    assert any(n.startswith("stage.") for n in names)
