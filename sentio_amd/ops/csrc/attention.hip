@@ -22,7 +22,7 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 //   C/D[16m x 16n]: lane l, reg r -> col = l&15, row = (l>>4)*4 + r
 
 #define QBLK 16
-#define KVBLK 32
+#define KVBLK 64   // keys per staged tile: amortizes softmax/barrier/staging
 #define MAXD 128
 
 // LDS byte-offset helpers
@@ -39,13 +39,14 @@ __device__ __forceinline__ int k_lds_off(int row, int d_byte, int Dbytes) {
 // load); the fragment read is 2 tr-reads instead of 8 scalar ds_reads.
 typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
 typedef __attribute__((address_space(3))) bf16x4_t* lds_v4_ptr;
+#define V_SLOTS (KVBLK / 4)
 __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
   const int db = d >> 4, kb = key >> 2;
-  const int slot = (kb & 1) ? 4 + (kb >> 1) : (kb >> 1);
-  return (db * 8 + slot) * 64 + (key & 3) * 16 + (d & 15);
+  const int slot = (kb & 1) ? (KVBLK / 8) + (kb >> 1) : (kb >> 1);
+  return (db * V_SLOTS + slot) * 64 + (key & 3) * 16 + (d & 15);
 }
-// P tile [QBLK][KVBLK] bf16 with 16-byte row pad: stride = 64+16 = 80 B
-#define P_STRIDE 80
+// P tile [QBLK][KVBLK] bf16 with 16-byte row pad
+#define P_STRIDE (KVBLK * 2 + 16)
 
 // 4 waves per block: waves handle 4 consecutive 16-row Q tiles of ONE
 // (batch, head) and SHARE the K/V LDS staging — 4x less HBM/LDS staging
@@ -161,10 +162,11 @@ __global__ void flash_attn_kernel(
     // still execute every barrier (uniform control flow)
     const bool compute = active && (!causal || kv0 <= q_hi);
 
-    // ---- S = scale * Q K^T for the two 16-key halves
-    f32x4_t s_acc[2];
+    // ---- S = scale * Q K^T for the KVBLK/16 16-key halves
+    constexpr int HALVES = KVBLK / 16;
+    f32x4_t s_acc[HALVES];
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < HALVES; ++half) {
       s_acc[half] = f32x4_t{};
 #pragma unroll
       for (int dc = 0; dc < DC; ++dc) {
@@ -177,9 +179,9 @@ __global__ void flash_attn_kernel(
       }
     }
 
-    // ---- online softmax on the 16x32 score tile
+    // ---- online softmax on the 16 x KVBLK score tile
     // lane holds: col = l&15 (+16*half), rows = (l>>4)*4 + r
-    float p_val[2][4];
+    float p_val[HALVES][4];
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) alpha[r] = 1.f;
@@ -188,29 +190,34 @@ __global__ void flash_attn_kernel(
     for (int r = 0; r < 4; ++r) {
       const int qrow_local = (lane >> 4) * 4 + r;
       const int qrow = q0 + qrow_local;
-      float s0 = s_acc[0][r] * scale;
-      float s1 = s_acc[1][r] * scale;
-      const int key0 = kv0 + (lane & 15);
-      const int key1 = key0 + 16;
-      bool ok0 = key0 < kvlen && (!causal || key0 <= qrow);
-      bool ok1 = key1 < kvlen && (!causal || key1 <= qrow);
-      s0 = ok0 ? s0 : -INFINITY;
-      s1 = ok1 ? s1 : -INFINITY;
-      // row max across the 16 lanes of the group (both halves)
-      float rmax = group16_max(fmaxf(s0, s1));
+      float sv[HALVES];
+      float rmax_l = -INFINITY;
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half) {
+        const int key = kv0 + half * 16 + (lane & 15);
+        const bool ok = key < kvlen && (!causal || key <= qrow);
+        sv[half] = ok ? s_acc[half][r] * scale : -INFINITY;
+        rmax_l = fmaxf(rmax_l, sv[half]);
+      }
+      // row max across the 16 lanes of the group (all halves)
+      float rmax = group16_max(rmax_l);
       float m_new = fmaxf(m_run[r], rmax);
       // guard: fully-masked row keeps m=-inf; exp(-inf - -inf) handled below
       float a = (m_run[r] == -INFINITY) ? 0.f
                 : __expf(m_run[r] - m_new);
       if (m_new == -INFINITY) a = 1.f;  // nothing seen yet at all
       alpha[r] = a;
-      float p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
-      float p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
-      float rsum = group16_sum(p0 + p1);
+      float psum = 0.f;
+#pragma unroll
+      for (int half = 0; half < HALVES; ++half) {
+        const float pv = (sv[half] == -INFINITY) ? 0.f
+                                                 : __expf(sv[half] - m_new);
+        p_val[half][r] = pv;
+        psum += pv;
+      }
+      float rsum = group16_sum(psum);
       l_run[r] = l_run[r] * a + rsum;
       m_run[r] = m_new;
-      p_val[0][r] = p0;
-      p_val[1][r] = p1;
     }
 
     // ---- write P to LDS (wave-private buffer) in C layout, reread as A
@@ -218,7 +225,7 @@ __global__ void flash_attn_kernel(
     // the block barrier's lgkmcnt drain; barriers stay uniform across waves.
     if (compute) {
 #pragma unroll
-      for (int half = 0; half < 2; ++half) {
+      for (int half = 0; half < HALVES; ++half) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qrow_local = (lane >> 4) * 4 + r;
@@ -231,31 +238,35 @@ __global__ void flash_attn_kernel(
     }
     __syncthreads();
     if (compute) {
-      // A frag for PV: P[row = l&15][k = key = kofs + j] — one frag covers
-      // all 32 keys of the tile (k = (l>>4)*8 + j spans 0..31)
-      const bf16x8_t a_p = *reinterpret_cast<const bf16x8_t*>(
-          p_lds + (lane & 15) * P_STRIDE + kofs * 2);
+      // A frags for PV: P[row = l&15][k = kc*32 + kofs + j] per 32-key chunk
+      bf16x8_t a_p[KVBLK / 32];
+#pragma unroll
+      for (int kc = 0; kc < KVBLK / 32; ++kc)
+        a_p[kc] = *reinterpret_cast<const bf16x8_t*>(
+            p_lds + (lane & 15) * P_STRIDE + (kc * 32 + kofs) * 2);
 
-      // ---- O = alpha*O + P V   (one mfma per 16-col block of V)
+      // ---- O = alpha*O + P V   (KVBLK/32 mfmas per 16-col block of V)
 #pragma unroll
       for (int nb = 0; nb < NB; ++nb) {
-        {
-          // rescale accumulator rows by alpha[r]
+        // rescale accumulator rows by alpha[r]
 #pragma unroll
-          for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
-          // B frag via hardware transpose-read: group g = lane>>4 gets
-          // column (lane&15) of tiles at slot g (keys 8g+0..3) and slot
-          // g+4 (keys 8g+4..7, uniform +512 B)
-          char* vb_base = vt_lds + nb * 1024 + (long)lane * 8;
+        for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
+        // B frags via hardware transpose-read: for key chunk kc, group
+        // g = lane>>4 needs keys kc*32 + 8g + 0..7 = tile slots
+        // {kc*(KVBLK/16) extra offset}: lo at +kc*512 B, hi at
+        // +kc*512 + (KVBLK/8)*128 B (evens-then-odds slot order)
+        char* vb_base = vt_lds + nb * (V_SLOTS * 128) + (long)lane * 8;
+#pragma unroll
+        for (int kc = 0; kc < KVBLK / 32; ++kc) {
           bf16x4_t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-              (lds_v4_ptr)vb_base);
+              (lds_v4_ptr)(vb_base + kc * 512));
           bf16x4_t hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-              (lds_v4_ptr)(vb_base + 512));
+              (lds_v4_ptr)(vb_base + kc * 512 + (KVBLK / 8) * 128));
           bf16x8_t b_v;
 #pragma unroll
           for (int j = 0; j < 4; ++j) { b_v[j] = lo[j]; b_v[4 + j] = hi[j]; }
           o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_p, b_v, o_acc[nb], 0, 0, 0);
+              a_p[kc], b_v, o_acc[nb], 0, 0, 0);
         }
       }
     }
