@@ -17,7 +17,6 @@ from __future__ import annotations
 import threading
 from typing import Any
 
-import numpy as np
 import torch
 
 from sentio_amd.models.document import Document

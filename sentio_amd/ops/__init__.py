@@ -9,8 +9,6 @@ hermetic config).
 
 from __future__ import annotations
 
-import os
-
 import torch
 
 from sentio_amd.ops import torch_ref

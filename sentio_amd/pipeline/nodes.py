@@ -11,7 +11,7 @@ from typing import Callable
 from sentio_amd.models.document import Document
 from sentio_amd.observability.metrics import metrics_collector
 from sentio_amd.pipeline.context import numbered_context, prepare_context
-from sentio_amd.pipeline.prompt_builder import MODE_INSTRUCTIONS, PromptBuilder
+from sentio_amd.pipeline.prompt_builder import PromptBuilder
 from sentio_amd.pipeline.state import (
     RAGState,
     add_metadata,

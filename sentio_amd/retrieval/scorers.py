@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import time
 
-import torch
 
 from sentio_amd.models.document import Document
 

@@ -11,7 +11,6 @@ from typing import Any
 
 from sentio_amd.caching.manager import CacheManager
 from sentio_amd.models.document import Document
-from sentio_amd.observability.metrics import metrics_collector
 from sentio_amd.pipeline.state import create_initial_state
 from sentio_amd.resilience.fallbacks import fallback_manager, llm_fallback
 from sentio_amd.serving.container import ServiceContainer
