@@ -92,9 +92,7 @@ def build_basic_graph(cfg: GraphConfig) -> RagPipeline:
                                                         cfg.generation_mode,
                                                         cfg.llm_max_tokens)))
     if cfg.use_verifier and cfg.verifier is not None:
-            verifier = cfg.verifier if not callable(getattr(cfg.verifier, "verify", None)) \
-            else cfg.verifier
-        stages.append(("verifier", N.create_verifier_node(verifier)))
+        stages.append(("verifier", N.create_verifier_node(cfg.verifier)))
     return RagPipeline(stages)
 
 
