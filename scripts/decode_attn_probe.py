@@ -35,3 +35,25 @@ def main():
 
 if __name__ == "__main__":
     main()
+    flash_probe()
+
+
+def flash_probe():
+    """Prefill flash-attention microbench (llama3-8b shape)."""
+    dev = "cuda:0"
+    for B, S in [(16, 2048), (32, 2048)]:
+        H, Hkv, D = 32, 8, 128
+        q = torch.randn(B, S, H, D, dtype=torch.bfloat16, device=dev)
+        k = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device=dev)
+        v = torch.randn(B, S, Hkv, D, dtype=torch.bfloat16, device=dev)
+        for _ in range(5):
+            ops.attention(q, k, v, causal=True)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        iters = 20
+        for _ in range(iters):
+            ops.attention(q, k, v, causal=True)
+        torch.cuda.synchronize()
+        t = (time.perf_counter() - t0) / iters
+        flops = 2.0 * B * H * S * S * D * 2 / 2  # causal half
+        print(f"flash B={B} S={S}: {t*1e3:.2f} ms  {flops/t/1e12:.0f} TFLOP/s")
