@@ -432,6 +432,25 @@ __global__ void decode_attn_split_kernel(
         for (int g = 0; g < G; ++g) sc[g] *= scale;
       }
     }
+    // issue ALL of this thread's V loads for the chunk NOW — they do not
+    // depend on the softmax, so the HBM pipe stays busy while the block
+    // reductions run (otherwise it idles through both reduction barriers)
+    constexpr int JT = DEC_CHUNK / 16;
+    bf16x8 v8[JT];
+    if (dg_ok) {
+#pragma unroll
+      for (int u = 0; u < JT; ++u) {
+        const int j = jslot + u * 16;
+        if (j < chunk) {
+          v8[u] = nt_load8(
+              reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
+              dgroup * 8);
+        } else {
+          bf16x8 z = {};
+          v8[u] = z;
+        }
+      }
+    }
     // batched per-head softmax: ONE reduction pair for all G maxima, one
     // for all G sums (was 3 barriers × 2 reductions × G heads).
     {
@@ -456,44 +475,24 @@ __global__ void decode_attn_split_kernel(
       for (int g = 0; g < G; ++g)
         l_run[g] = l_run[g] * alpha[g] + sums[g];
     }
-    // phase B: PV.  v loads are bf16x8 (16 B) and fully coalesced; the
-    // p_sh reads broadcast (16 lanes share one address).  Full chunks:
-    // all 16 V loads issued up front (batched MLP), FMAs after.
+    // phase B: PV consume — V already in flight since before the softmax.
 #pragma unroll
     for (int g = 0; g < G; ++g)
 #pragma unroll
       for (int e = 0; e < 8; ++e) o_part[g][e] *= alpha[g];
     if (dg_ok) {
-      constexpr int JT = DEC_CHUNK / 16;
-      constexpr int JB = 8;              // V loads in flight per batch
 #pragma unroll
-      for (int u0 = 0; u0 < JT; u0 += JB) {
-        bf16x8 v8[JB];
+      for (int u = 0; u < JT; ++u) {
+        const int j = jslot + u * 16;
+        if (j >= chunk) continue;       // p is 0 there anyway
+        float vf[8];
 #pragma unroll
-        for (int u = 0; u < JB; ++u) {
-          const int j = jslot + (u0 + u) * 16;
-          if (j < chunk) {
-            v8[u] = nt_load8(
-                reinterpret_cast<const short*>(vb + (long)(s0 + j) * D) +
-                dgroup * 8);
-          } else {
-            bf16x8 z = {};
-            v8[u] = z;
-          }
-        }
+        for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
 #pragma unroll
-        for (int u = 0; u < JB; ++u) {
-          const int j = jslot + (u0 + u) * 16;
-          if (j >= chunk) continue;     // p is 0 there anyway
-          float vf[8];
+        for (int g = 0; g < G; ++g) {
+          const float p = p_sh[g * DEC_CHUNK + j];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
-#pragma unroll
-          for (int g = 0; g < G; ++g) {
-            const float p = p_sh[g * DEC_CHUNK + j];
-#pragma unroll
-            for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
-          }
+          for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
         }
       }
     }
