@@ -33,9 +33,7 @@ def main():
             t = bench(q, kc, vc, lens)
             print(f"splits={splits or 'auto'}: {t*1e6:7.1f}us  {kv_bytes/t/1e12:.2f} TB/s")
 
-if __name__ == "__main__":
-    main()
-    flash_probe()
+
 
 
 def flash_probe():
@@ -57,3 +55,8 @@ def flash_probe():
         t = (time.perf_counter() - t0) / iters
         flops = 2.0 * B * H * S * S * D * 2 / 2  # causal half
         print(f"flash B={B} S={S}: {t*1e3:.2f} ms  {flops/t/1e12:.0f} TFLOP/s")
+
+
+if __name__ == "__main__":
+    main()
+    flash_probe()

@@ -165,6 +165,40 @@ def bm25_score(term_ids, indptr, post_doc, post_tf, idf, doc_len, *,
 
 # ---------------- sampling / GEMM ----------------
 
+_FUSE_METHODS = {"rrf": 0, "weighted_rrf": 1, "comb_sum": 2}
+
+
+def fuse_topk(d_ids, d_scores, s_ids, s_scores, *, method: str = "rrf",
+              top_k: int = 10, rrf_k: float = 60.0, dense_weight: float = 0.7,
+              sparse_weight: float = 0.3):
+    """Batched device fusion (K4): per-query dense+sparse candidate lists
+    (int64 ids, -1 pad; rank order) → fused top-k (ids, scores).
+    CPU path defers to index.fusion.fuse (the semantics oracle)."""
+    if method not in _FUSE_METHODS:
+        raise ValueError(f"Unknown fusion_method: {method}")
+    if _on_gpu(d_ids):
+        return _require_hip().fuse_topk(
+            d_ids, d_scores, s_ids, s_scores, int(top_k),
+            _FUSE_METHODS[method], float(rrf_k), float(dense_weight),
+            float(sparse_weight))
+    from sentio_amd.index import fusion as F
+
+    B = d_ids.shape[0]
+    out_i = torch.full((B, top_k), -1, dtype=torch.int64)
+    out_s = torch.zeros(B, top_k)
+    for q in range(B):
+        dh = [(str(int(i)), float(s)) for i, s in zip(d_ids[q], d_scores[q])
+              if int(i) >= 0]
+        sh = [(str(int(i)), float(s)) for i, s in zip(s_ids[q], s_scores[q])
+              if int(i) >= 0]
+        fused = F.fuse(dh, sh, method=method, top_k=top_k, rrf_k=int(rrf_k),
+                       dense_weight=dense_weight, sparse_weight=sparse_weight)
+        for k, (doc, sc) in enumerate(fused):
+            out_i[q, k] = int(doc)
+            out_s[q, k] = sc
+    return out_i, out_s
+
+
 def sample_token(logits, temperature: float, seed: int = 0):
     if _on_gpu(logits):
         return _require_hip().sample_token(
@@ -195,5 +229,5 @@ def gemm_bf16(a, b):
 __all__ = [
     "rmsnorm", "rmsnorm_residual", "rope_apply", "decode_qkv_prep", "swiglu", "swiglu_packed", "softmax",
     "attention", "decode_attention", "mean_pool_l2norm", "cosine_topk",
-    "bm25_score", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
+    "bm25_score", "fuse_topk", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
 ]

@@ -40,6 +40,9 @@ hipError_t sentio_gemm_bf16(const void*, const void*, void*, int, int, int,
                             hipStream_t);
 hipError_t sentio_skinny_gemm(const void*, const void*, void*, int, int, int,
                               hipStream_t);
+hipError_t sentio_fuse_topk(const long*, const float*, const long*,
+                            const float*, long*, float*, int, int, int, int,
+                            int, float, float, float, hipStream_t);
 }
 
 namespace {
@@ -295,6 +298,27 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
   return out;
 }
 
+std::vector<torch::Tensor> fuse_topk(torch::Tensor d_ids, torch::Tensor d_scores,
+                                     torch::Tensor s_ids, torch::Tensor s_scores,
+                                     int64_t top_k, int64_t method,
+                                     double rrf_k, double dw, double sw) {
+  TORCH_CHECK(d_ids.is_cuda() && d_ids.scalar_type() == torch::kLong, "d_ids i64 GPU");
+  TORCH_CHECK(s_ids.scalar_type() == torch::kLong, "s_ids i64");
+  const int B = d_ids.size(0), Kd = d_ids.size(1), Ks = s_ids.size(1);
+  auto out_ids = torch::empty({B, top_k}, d_ids.options());
+  auto out_scores = torch::empty({B, top_k},
+                                 d_scores.options().dtype(torch::kFloat));
+  check_hip(sentio_fuse_topk(
+      d_ids.contiguous().data_ptr<int64_t>(),
+      d_scores.to(torch::kFloat).contiguous().data_ptr<float>(),
+      s_ids.contiguous().data_ptr<int64_t>(),
+      s_scores.to(torch::kFloat).contiguous().data_ptr<float>(),
+      out_ids.data_ptr<int64_t>(), out_scores.data_ptr<float>(), B, Kd, Ks,
+      (int)top_k, (int)method, (float)rrf_k, (float)dw, (float)sw,
+      stream()), "fuse_topk");
+  return {out_ids, out_scores};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_residual", &rmsnorm_residual);
@@ -311,4 +335,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attn", &decode_attn);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("fuse_topk", &fuse_topk);
 }

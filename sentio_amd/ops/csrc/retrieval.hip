@@ -113,6 +113,117 @@ __global__ void bm25_kernel(const long* __restrict__ term_ids,
   }
 }
 
+
+// ------------------------------------------------------------- fusion (K4)
+// Batched device fusion of per-source top-k candidate lists
+// (reference hybrid.py:204-259 semantics; host fusion.py is the oracle).
+// One block per query; union <= FUSE_MAX entries; ids are int64 (-1 = pad).
+// method: 0=rrf, 1=weighted_rrf, 2=comb_sum.
+#define FUSE_MAX 128
+__global__ void fuse_topk_kernel(
+    const long* __restrict__ d_ids, const float* __restrict__ d_scores,
+    const long* __restrict__ s_ids, const float* __restrict__ s_scores,
+    long* __restrict__ out_ids, float* __restrict__ out_scores,
+    int Kd, int Ks, int top_k, int method, float rrf_k, float dw, float sw) {
+  __shared__ long ids[FUSE_MAX];
+  __shared__ float score[FUSE_MAX];
+  __shared__ unsigned char owner[FUSE_MAX];
+  __shared__ float red[2];
+  const int q = blockIdx.x;
+  const int n = Kd + Ks;
+  const int t = threadIdx.x;
+
+  // load candidates in insertion order: dense ranks then sparse ranks
+  for (int i = t; i < n; i += blockDim.x) {
+    ids[i] = (i < Kd) ? d_ids[(long)q * Kd + i] : s_ids[(long)q * Ks + (i - Kd)];
+    score[i] = 0.f;
+    owner[i] = 0;
+  }
+  __syncthreads();
+  // owner = first slot with this id (stable insertion order)
+  for (int i = t; i < n; i += blockDim.x) {
+    if (ids[i] < 0) continue;
+    int first = i;
+    for (int j = 0; j < i; ++j)
+      if (ids[j] == ids[i]) { first = j; break; }
+    owner[i] = (first == i);
+  }
+  __syncthreads();
+
+  // per-source normalization constants (comb_sum only)
+  float dmin = 0.f, dscale = 0.f, smin = 0.f, sscale = 0.f;
+  if (method == 2) {
+    if (t == 0) {
+      float lo = INFINITY, hi = -INFINITY;
+      for (int i = 0; i < Kd; ++i) {
+        const long id = d_ids[(long)q * Kd + i];
+        if (id < 0) continue;
+        const float v = d_scores[(long)q * Kd + i];
+        lo = fminf(lo, v); hi = fmaxf(hi, v);
+      }
+      red[0] = lo; red[1] = hi;
+    }
+    __syncthreads();
+    dmin = red[0]; dscale = red[1] - red[0];
+    __syncthreads();
+    if (t == 0) {
+      float lo = INFINITY, hi = -INFINITY;
+      for (int i = 0; i < Ks; ++i) {
+        const long id = s_ids[(long)q * Ks + i];
+        if (id < 0) continue;
+        const float v = s_scores[(long)q * Ks + i];
+        lo = fminf(lo, v); hi = fmaxf(hi, v);
+      }
+      red[0] = lo; red[1] = hi;
+    }
+    __syncthreads();
+    smin = red[0]; sscale = red[1] - red[0];
+  }
+
+  // accumulate each entry's contribution into its owner slot
+  for (int i = t; i < n; i += blockDim.x) {
+    if (ids[i] < 0) continue;
+    const bool is_dense = i < Kd;
+    const int rank = is_dense ? i : (i - Kd);
+    float c;
+    if (method == 2) {
+      const float raw = is_dense ? d_scores[(long)q * Kd + rank]
+                                 : s_scores[(long)q * Ks + rank];
+      const float mn = is_dense ? dmin : smin;
+      const float sc = is_dense ? dscale : sscale;
+      const float norm = (sc > 0.f) ? (raw - mn) / sc : 1.f;
+      c = (is_dense ? dw : sw) * norm;
+    } else {
+      const float w = (method == 0) ? 1.f : (is_dense ? dw : sw);
+      c = w / (rrf_k + (float)rank);
+    }
+    // find owner slot
+    int o = i;
+    for (int j = 0; j < i; ++j)
+      if (ids[j] == ids[i]) { o = j; break; }
+    atomicAdd(&score[o], c);
+  }
+  __syncthreads();
+
+  // top-k selection: thread 0 scans (n <= 128, k <= 20 — trivial)
+  if (t == 0) {
+    for (int k = 0; k < top_k; ++k) {
+      float best = -INFINITY;
+      int bi = -1;
+      for (int i = 0; i < n; ++i)
+        if (owner[i] && score[i] > best) { best = score[i]; bi = i; }
+      if (bi < 0) {
+        out_ids[(long)q * top_k + k] = -1;
+        out_scores[(long)q * top_k + k] = 0.f;
+      } else {
+        out_ids[(long)q * top_k + k] = ids[bi];
+        out_scores[(long)q * top_k + k] = best;
+        owner[bi] = 0;
+      }
+    }
+  }
+}
+
 extern "C" {
 
 hipError_t sentio_cosine_scores_f16(const void* mat, const void* q,
@@ -139,6 +250,19 @@ hipError_t sentio_bm25(const long* term_ids, const long* qoff,
   hipLaunchKernelGGL(bm25_kernel, dim3((unsigned)blocks), dim3(256), 0, stream,
                      term_ids, qoff, starts, post_doc, post_tf, idf, doc_len,
                      scores, T, total, k1, b, 1.f / avgdl, delta);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sentio_fuse_topk(const long* d_ids, const float* d_scores,
+                            const long* s_ids, const float* s_scores,
+                            long* out_ids, float* out_scores, int B, int Kd,
+                            int Ks, int top_k, int method, float rrf_k,
+                            float dw, float sw, hipStream_t stream) {
+  if (Kd + Ks > FUSE_MAX || top_k > FUSE_MAX) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(fuse_topk_kernel, dim3(B), dim3(128), 0, stream, d_ids,
+                     d_scores, s_ids, s_scores, out_ids, out_scores, Kd, Ks,
+                     top_k, method, rrf_k, dw, sw);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

@@ -347,3 +347,33 @@ def test_skinny_gemm(dev):
         got = ops.skinny_gemm(x, w)
         want = torch.nn.functional.linear(x.cpu().float(), w.cpu().float())
         _cmp(got, want, rtol=3e-2, atol=3e-1)
+
+
+@pytest.mark.parametrize("method", ["rrf", "weighted_rrf", "comb_sum"])
+def test_fuse_topk_matches_host(dev, method):
+    from sentio_amd import ops
+    from sentio_amd.index import fusion
+
+    torch.manual_seed(11)
+    B, Kd, Ks, top_k = 4, 10, 10, 6
+    d_ids = torch.randint(0, 40, (B, Kd), dtype=torch.int64)
+    s_ids = torch.randint(0, 40, (B, Ks), dtype=torch.int64)
+    d_scores = torch.rand(B, Kd).sort(dim=1, descending=True).values
+    s_scores = torch.rand(B, Ks).sort(dim=1, descending=True).values
+    d_ids[0, -1] = -1  # padding case
+    got_i, got_s = ops.fuse_topk(
+        d_ids.to(dev), d_scores.to(dev), s_ids.to(dev), s_scores.to(dev),
+        method=method, top_k=top_k)
+    for q in range(B):
+        dh = [(str(int(i)), float(s)) for i, s in zip(d_ids[q], d_scores[q])
+              if int(i) >= 0]
+        sh = [(str(int(i)), float(s)) for i, s in zip(s_ids[q], s_scores[q])
+              if int(i) >= 0]
+        want = fusion.fuse(dh, sh, method=method, top_k=top_k)
+        got = [(str(int(i)), float(s))
+               for i, s in zip(got_i[q].cpu(), got_s[q].cpu()) if int(i) >= 0]
+        assert len(got) == len(want)
+        for (gi, gs), (wi, ws) in zip(got, want):
+            assert abs(gs - ws) < 1e-4
+        # same id SET at equal scores (ties may reorder)
+        assert {g[0] for g in got} == {w[0] for w in want}
