@@ -266,7 +266,12 @@ def main():
         stage_t[name] = stage_t.get(name, 0.0) + (t1 - t0)
         return t1
 
-    def chat_step(step: int) -> None:
+    # The per-step pipeline is split into a retrieval phase (stages 1-6,
+    # ends fully host-synced) and a generation phase (7-8).  Steps are
+    # independent requests, so step N+1's retrieval runs on a side thread +
+    # side HIP stream UNDER step N's generation — steady-state serving
+    # overlap; ms_per_step stays wall-clock / completed batches.
+    def retrieval_phase(step: int):
         t0 = time.perf_counter()
         queries = make_queries(step)
         # 1. embed queries (one encoder batch)
@@ -354,8 +359,22 @@ def main():
             ctx = prepare_context(docs)
             prompts.append(builder.system_prompt() + "\n\n" +
                            builder.build_qa_prompt(queries[qi], ctx))
-        # 7. batched generation
         t0 = _mark("select", t0)
+        return queries, reranked, prompts
+
+    ret_stream = torch.cuda.Stream() if on_gpu else None
+
+    def retrieval_phase_streamed(step: int):
+        # side stream: retrieval kernels fill gaps under decode; the phase
+        # returns host data only (every GPU result is .cpu()'d inside)
+        if on_gpu:
+            with torch.cuda.stream(ret_stream):
+                return retrieval_phase(step)
+        return retrieval_phase(step)
+
+    def generate_phase(payload) -> None:
+        queries, reranked, prompts = payload
+        t0 = time.perf_counter()
         answers = generator.generate(prompts, max_new_tokens=args.gen_tokens,
                                      temperature=0.3, stop_on_eos=False)
         stage_t["gen.prefill"] = stage_t.get("gen.prefill", 0.0) + \
@@ -363,7 +382,7 @@ def main():
         stage_t["gen.decode"] = stage_t.get("gen.decode", 0.0) + \
             getattr(generator, "last_decode_s", 0.0)
         t0 = _mark("generate", t0)
-        # 8. optional verify
+        # optional verify
         if args.verify:
             vprompts = [builder.build_verify_prompt(
                 query=queries[qi], context=prepare_context(reranked[qi]),
@@ -371,17 +390,39 @@ def main():
             generator.generate(vprompts, max_new_tokens=args.verify_tokens,
                                temperature=0.0, stop_on_eos=False)
 
+    from concurrent.futures import ThreadPoolExecutor
+
+    ret_ex = ThreadPoolExecutor(max_workers=1)
+
+    # TP mode interleaves generation all-reduces with retrieval all-gathers
+    # from two threads — collective-order hazard across ranks — so TP runs
+    # the phases back-to-back.  DP keeps ALL collectives on the retrieval
+    # thread in deterministic step order.
+    pipelined = not tp_mode
+
+    def run_pipelined(base: int, count: int) -> None:
+        if count <= 0:
+            return
+        if not pipelined:
+            for i in range(count):
+                generate_phase(retrieval_phase_streamed(base + i))
+            return
+        fut = ret_ex.submit(retrieval_phase_streamed, base)
+        for i in range(count):
+            payload = fut.result()
+            if i + 1 < count:
+                fut = ret_ex.submit(retrieval_phase_streamed, base + i + 1)
+            generate_phase(payload)
+
     # ---- warmup ----
-    for w in range(args.warmup):
-        chat_step(w)
+    run_pipelined(0, args.warmup)
     D.barrier()
     if on_gpu:
         torch.cuda.synchronize()
 
     # ---- timed ----
     t0 = time.perf_counter()
-    for s in range(args.steps):
-        chat_step(1000 + s)
+    run_pipelined(1000, args.steps)
     D.barrier()
     if on_gpu:
         torch.cuda.synchronize()
@@ -428,7 +469,9 @@ def main():
                 "parallelism": (f"tp{world}+index-shard{world}" if tp_mode
                                 else f"dp{world}+index-shard{world}"),
                 "pipeline": "embed>hybrid(dense+bm25+rrf)>rerank>select>generate"
-                            + (">verify" if args.verify else ""),
+                            + (">verify" if args.verify else "")
+                            + ("|retrieval pipelined under prior step's "
+                               "generation" if pipelined else ""),
                 "p50_ms_per_request_batch": round(ms_per_step, 1),
                 "init_s": round(init_s, 1),
                 "device": "cuda" if on_gpu else "cpu-plumbing",
