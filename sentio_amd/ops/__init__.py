@@ -172,7 +172,8 @@ def fuse_topk(d_ids, d_scores, s_ids, s_scores, *, method: str = "rrf",
               top_k: int = 10, rrf_k: float = 60.0, dense_weight: float = 0.7,
               sparse_weight: float = 0.3):
     """Batched device fusion (K4): per-query dense+sparse candidate lists
-    (int64 ids, -1 pad; rank order) → fused top-k (ids, scores).
+    (int64 ids, -1 pad; rank order; ids unique within a list — a top-k from
+    one source never repeats a doc) → fused top-k (ids, scores).
     CPU path defers to index.fusion.fuse (the semantics oracle)."""
     if method not in _FUSE_METHODS:
         raise ValueError(f"Unknown fusion_method: {method}")

@@ -356,8 +356,10 @@ def test_fuse_topk_matches_host(dev, method):
 
     torch.manual_seed(11)
     B, Kd, Ks, top_k = 4, 10, 10, 6
-    d_ids = torch.randint(0, 40, (B, Kd), dtype=torch.int64)
-    s_ids = torch.randint(0, 40, (B, Ks), dtype=torch.int64)
+    # ids are unique WITHIN each list (a top-k from one source never
+    # repeats a doc) but overlap across lists — the kernel's contract
+    d_ids = torch.stack([torch.randperm(40)[:Kd] for _ in range(B)]).long()
+    s_ids = torch.stack([torch.randperm(40)[:Ks] for _ in range(B)]).long()
     d_scores = torch.rand(B, Kd).sort(dim=1, descending=True).values
     s_scores = torch.rand(B, Ks).sort(dim=1, descending=True).values
     d_ids[0, -1] = -1  # padding case
