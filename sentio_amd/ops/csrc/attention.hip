@@ -53,6 +53,11 @@ __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
 // traffic than wave-private tiles, cooperative 256-thread staging.
 #define FA_WAVES 4
 
+// QT q-tiles (16 rows each) per wave: the K fragment loads and the V
+// transpose-reads are shared across the wave's QT row-tiles, and every
+// barrier/staging pass serves QT x 16 rows — per-row overhead halves at
+// QT=2 vs the one-tile version.
+#define FA_QT 2
 template <int DT>
 __launch_bounds__(256)
 __global__ void flash_attn_kernel(
@@ -64,58 +69,60 @@ __global__ void flash_attn_kernel(
     int B, int S, int H, int Hkv, int D_, float scale, int causal) {
   constexpr int D = DT;            // compile-time: every staging/frag loop
                                    // unrolls, loads batch before waits
+  constexpr int QT = FA_QT;
+  constexpr int WROWS = QT * QBLK; // q rows per wave
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // carve: K tile | V^T tile | per-wave P tiles
+  // carve: K tile | V tr image | per-wave P tiles
   char* k_lds = smem;                                   // KVBLK * D * 2
-  char* vt_lds = k_lds + KVBLK * D * 2;                 // KVBLK * D * 2 (tr image)
+  char* vt_lds = k_lds + KVBLK * D * 2;                 // KVBLK * D * 2
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  char* p_lds = vt_lds + KVBLK * D * 2 + wid * QBLK * P_STRIDE;
+  char* p_lds = vt_lds + KVBLK * D * 2 + wid * WROWS * P_STRIDE;
 
-  const int qt = blockIdx.x * FA_WAVES + wid;  // this wave's q tile
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hkv = h / (H / Hkv);
-  const int q0 = qt * QBLK;
+  const int q0 = (blockIdx.x * FA_WAVES + wid) * WROWS;
   const int kvlen = min(kv_lens[b], S);
   const bool active = q0 < S;           // inactive waves still hit barriers
-  const int q_hi = min(q0 + QBLK - 1, S - 1);
 
   constexpr int DC = D / 32;            // feature chunks per mfma K-dim
   constexpr int NB = D / 16;            // output column blocks
 
-  // ---- load Q fragments: a_q[dc] = Q[q0 + (l&15)][dc*32 + (l>>4)*8 + j]
+  // ---- load Q fragments per tile t:
+  // a_q[t][dc] = Q[q0 + t*16 + (l&15)][dc*32 + (l>>4)*8 + j]
   const int arow = lane & 15;
   const int kofs = (lane >> 4) * 8;
-  bf16x8_t a_q[DC];
-  {
-    const int qrow = q0 + arow;
+  bf16x8_t a_q[QT][DC];
+#pragma unroll
+  for (int t = 0; t < QT; ++t) {
+    const int qrow = q0 + t * QBLK + arow;
     const bf16* qp = q + (((long)b * S + qrow) * H + h) * D + kofs;
 #pragma unroll
     for (int dc = 0; dc < DC; ++dc) {
-      if (active && qrow < S) {
-        a_q[dc] = *reinterpret_cast<const bf16x8_t*>(qp + dc * 32);
+      if (qrow < S) {
+        a_q[t][dc] = *reinterpret_cast<const bf16x8_t*>(qp + dc * 32);
       } else {
         bf16x8_t z = {};
-        a_q[dc] = z;
+        a_q[t][dc] = z;
       }
     }
   }
 
-  // ---- accumulators
-  f32x4_t o_acc[NB];                    // O in C-frag layout per 16-col block
+  // ---- accumulators (per tile)
+  f32x4_t o_acc[QT][NB];
+  float m_run[QT][4], l_run[QT][4];
 #pragma unroll
-  for (int nb = 0; nb < NB; ++nb) o_acc[nb] = f32x4_t{};
-  // per-lane row state: the C layout puts row = (l>>4)*4 + r; softmax rows
-  // are shared by the 16 lanes of each group -> track per (group,reg)
-  float m_run[4], l_run[4];
+  for (int t = 0; t < QT; ++t) {
 #pragma unroll
-  for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+    for (int nb = 0; nb < NB; ++nb) o_acc[t][nb] = f32x4_t{};
+#pragma unroll
+    for (int r = 0; r < 4; ++r) { m_run[t][r] = -INFINITY; l_run[t][r] = 0.f; }
+  }
 
-  // block-level kv bound: the LAST wave's causal horizon (waves skip their
-  // own out-of-horizon tiles compute-side)
-  const int block_q_hi = min(blockIdx.x * FA_WAVES * QBLK + FA_WAVES * QBLK,
+  // block-level kv bound: the LAST wave's causal horizon
+  const int block_q_hi = min(blockIdx.x * FA_WAVES * WROWS + FA_WAVES * WROWS,
                              S);
   const int kv_hi = causal ? min(kvlen, block_q_hi) : kvlen;
   const int n_kv_tiles = (kv_hi + KVBLK - 1) / KVBLK;
@@ -160,101 +167,91 @@ __global__ void flash_attn_kernel(
 
     // waves whose causal horizon ends before this kv tile skip compute but
     // still execute every barrier (uniform control flow)
-    const bool compute = active && (!causal || kv0 <= q_hi);
+    const int wave_q_hi = min(q0 + WROWS - 1, S - 1);
+    const bool compute = active && (!causal || kv0 <= wave_q_hi);
 
-    // ---- S = scale * Q K^T for the KVBLK/16 16-key halves
     constexpr int HALVES = KVBLK / 16;
-    f32x4_t s_acc[HALVES];
-#pragma unroll
-    for (int half = 0; half < HALVES; ++half) {
-      s_acc[half] = f32x4_t{};
-#pragma unroll
-      for (int dc = 0; dc < DC; ++dc) {
-        // B frag: K_lds[half*16 + (l&15)][dc*32 + kofs + j]
-        const int krow = half * 16 + (lane & 15);
-        bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(
-            k_lds + k_lds_off(krow, (dc * 32 + kofs) * 2, D * 2));
-        s_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_q[dc], b_frag, s_acc[half], 0, 0, 0);
-      }
-    }
-
-    // ---- online softmax on the 16 x KVBLK score tile
-    // lane holds: col = l&15 (+16*half), rows = (l>>4)*4 + r
-    float p_val[HALVES][4];
-    float alpha[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) alpha[r] = 1.f;
-    if (compute)
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow_local = (lane >> 4) * 4 + r;
-      const int qrow = q0 + qrow_local;
-      float sv[HALVES];
-      float rmax_l = -INFINITY;
-#pragma unroll
-      for (int half = 0; half < HALVES; ++half) {
-        const int key = kv0 + half * 16 + (lane & 15);
-        const bool ok = key < kvlen && (!causal || key <= qrow);
-        sv[half] = ok ? s_acc[half][r] * scale : -INFINITY;
-        rmax_l = fmaxf(rmax_l, sv[half]);
-      }
-      // row max across the 16 lanes of the group (all halves)
-      float rmax = group16_max(rmax_l);
-      float m_new = fmaxf(m_run[r], rmax);
-      // guard: fully-masked row keeps m=-inf; exp(-inf - -inf) handled below
-      float a = (m_run[r] == -INFINITY) ? 0.f
-                : __expf(m_run[r] - m_new);
-      if (m_new == -INFINITY) a = 1.f;  // nothing seen yet at all
-      alpha[r] = a;
-      float psum = 0.f;
-#pragma unroll
-      for (int half = 0; half < HALVES; ++half) {
-        const float pv = (sv[half] == -INFINITY) ? 0.f
-                                                 : __expf(sv[half] - m_new);
-        p_val[half][r] = pv;
-        psum += pv;
-      }
-      float rsum = group16_sum(psum);
-      l_run[r] = l_run[r] * a + rsum;
-      m_run[r] = m_new;
-    }
-
-    // ---- write P to LDS (wave-private buffer) in C layout, reread as A
-    // fragments.  The in-wave ds-write -> ds-read ordering is provided by
-    // the block barrier's lgkmcnt drain; barriers stay uniform across waves.
+    float alpha[QT][4];
     if (compute) {
+      // ---- S = scale * Q K^T: K B-frag loaded ONCE per (half, dc),
+      // reused by every q tile
+      f32x4_t s_acc[QT][HALVES];
+#pragma unroll
+      for (int t = 0; t < QT; ++t)
+#pragma unroll
+        for (int half = 0; half < HALVES; ++half) s_acc[t][half] = f32x4_t{};
 #pragma unroll
       for (int half = 0; half < HALVES; ++half) {
+#pragma unroll
+        for (int dc = 0; dc < DC; ++dc) {
+          const int krow = half * 16 + (lane & 15);
+          bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(
+              k_lds + k_lds_off(krow, (dc * 32 + kofs) * 2, D * 2));
+#pragma unroll
+          for (int t = 0; t < QT; ++t)
+            s_acc[t][half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_q[t][dc], b_frag, s_acc[t][half], 0, 0, 0);
+        }
+      }
+
+      // ---- online softmax per tile; P written to the wave's LDS buffer
+#pragma unroll
+      for (int t = 0; t < QT; ++t) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qrow_local = (lane >> 4) * 4 + r;
-          const int key_local = half * 16 + (lane & 15);
-          *reinterpret_cast<__bf16*>(
-              p_lds + qrow_local * P_STRIDE + key_local * 2) =
-              (__bf16)p_val[half][r];
+          const int qrow = q0 + t * QBLK + qrow_local;
+          float sv[HALVES];
+          float rmax_l = -INFINITY;
+#pragma unroll
+          for (int half = 0; half < HALVES; ++half) {
+            const int key = kv0 + half * 16 + (lane & 15);
+            const bool ok = key < kvlen && (!causal || key <= qrow);
+            sv[half] = ok ? s_acc[t][half][r] * scale : -INFINITY;
+            rmax_l = fmaxf(rmax_l, sv[half]);
+          }
+          float rmax = group16_max(rmax_l);
+          float m_new = fmaxf(m_run[t][r], rmax);
+          float a = (m_run[t][r] == -INFINITY) ? 0.f
+                    : __expf(m_run[t][r] - m_new);
+          if (m_new == -INFINITY) a = 1.f;  // nothing seen yet at all
+          alpha[t][r] = a;
+          float psum = 0.f;
+#pragma unroll
+          for (int half = 0; half < HALVES; ++half) {
+            const float pv = (sv[half] == -INFINITY)
+                                 ? 0.f : __expf(sv[half] - m_new);
+            psum += pv;
+            *reinterpret_cast<__bf16*>(
+                p_lds + (t * QBLK + qrow_local) * P_STRIDE +
+                (half * 16 + (lane & 15)) * 2) = (__bf16)pv;
+          }
+          float rsum = group16_sum(psum);
+          l_run[t][r] = l_run[t][r] * a + rsum;
+          m_run[t][r] = m_new;
         }
       }
     }
     __syncthreads();
     if (compute) {
-      // A frags for PV: P[row = l&15][k = kc*32 + kofs + j] per 32-key chunk
-      bf16x8_t a_p[KVBLK / 32];
+      // A frags for PV per tile and 32-key chunk
+      bf16x8_t a_p[QT][KVBLK / 32];
 #pragma unroll
-      for (int kc = 0; kc < KVBLK / 32; ++kc)
-        a_p[kc] = *reinterpret_cast<const bf16x8_t*>(
-            p_lds + (lane & 15) * P_STRIDE + (kc * 32 + kofs) * 2);
+      for (int t = 0; t < QT; ++t)
+#pragma unroll
+        for (int kc = 0; kc < KVBLK / 32; ++kc)
+          a_p[t][kc] = *reinterpret_cast<const bf16x8_t*>(
+              p_lds + (t * QBLK + (lane & 15)) * P_STRIDE +
+              (kc * 32 + kofs) * 2);
 
-      // ---- O = alpha*O + P V   (KVBLK/32 mfmas per 16-col block of V)
+      // ---- O = alpha*O + P V: V tr-reads loaded once per (nb, kc),
+      // reused by every q tile
 #pragma unroll
       for (int nb = 0; nb < NB; ++nb) {
-        // rescale accumulator rows by alpha[r]
 #pragma unroll
-        for (int r = 0; r < 4; ++r) o_acc[nb][r] *= alpha[r];
-        // B frags via hardware transpose-read: for key chunk kc, group
-        // g = lane>>4 needs keys kc*32 + 8g + 0..7 = tile slots
-        // {kc*(KVBLK/16) extra offset}: lo at +kc*512 B, hi at
-        // +kc*512 + (KVBLK/8)*128 B (evens-then-odds slot order)
+        for (int t = 0; t < QT; ++t)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) o_acc[t][nb][r] *= alpha[t][r];
         char* vb_base = vt_lds + nb * (V_SLOTS * 128) + (long)lane * 8;
 #pragma unroll
         for (int kc = 0; kc < KVBLK / 32; ++kc) {
@@ -265,8 +262,10 @@ __global__ void flash_attn_kernel(
           bf16x8_t b_v;
 #pragma unroll
           for (int j = 0; j < 4; ++j) { b_v[j] = lo[j]; b_v[4 + j] = hi[j]; }
-          o_acc[nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_p[kc], b_v, o_acc[nb], 0, 0, 0);
+#pragma unroll
+          for (int t = 0; t < QT; ++t)
+            o_acc[t][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_p[t][kc], b_v, o_acc[t][nb], 0, 0, 0);
         }
       }
     }
@@ -276,17 +275,17 @@ __global__ void flash_attn_kernel(
   // ---- epilogue: divide by l, store
   if (!active) return;
 #pragma unroll
-  for (int nb = 0; nb < NB; ++nb) {
-    {
+  for (int t = 0; t < QT; ++t) {
+#pragma unroll
+    for (int nb = 0; nb < NB; ++nb) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow_local = (lane >> 4) * 4 + r;
-        const int qrow = q0 + qrow_local;
+        const int qrow = q0 + t * QBLK + (lane >> 4) * 4 + r;
         if (qrow < S) {
-          const float denom = l_run[r] > 0.f ? l_run[r] : 1.f;
+          const float denom = l_run[t][r] > 0.f ? l_run[t][r] : 1.f;
           const int d = nb * 16 + (lane & 15);
           out[(((long)b * S + qrow) * H + h) * D + d] =
-              f2bf(o_acc[nb][r] / denom);
+              f2bf(o_acc[t][nb][r] / denom);
         }
       }
     }
@@ -562,8 +561,9 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
                              int H, int Hkv, int D, float scale, int causal,
                              hipStream_t stream) {
   size_t lds = (size_t)KVBLK * D * 2 * 2   // K (swizzled) + V (tr image)
-               + 4 * QBLK * P_STRIDE;
-  dim3 grid((S + 4 * QBLK - 1) / (4 * QBLK), H, B);
+               + 4 * FA_QT * QBLK * P_STRIDE;
+  const int wave_rows = FA_QT * QBLK;
+  dim3 grid((S + 4 * wave_rows - 1) / (4 * wave_rows), H, B);
 #define FA_CASE(DV)                                                          \
   case DV:                                                                   \
     hipLaunchKernelGGL((flash_attn_kernel<DV>), grid, dim3(256), lds,        \
