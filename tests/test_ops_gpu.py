@@ -379,3 +379,32 @@ def test_fuse_topk_matches_host(dev, method):
             assert abs(gs - ws) < 1e-4
         # same id SET at equal scores (ties may reorder)
         assert {g[0] for g in got} == {w[0] for w in want}
+
+
+def test_batched_generator_concurrent_gpu(dev):
+    """Serving hot path on device: concurrent single-prompt requests share
+    one decode batch through the dynamic batcher."""
+    import threading
+
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.serving.batcher import BatchedGenerator
+
+    eng = GeneratorEngine("llama3-1b", device=dev, max_seq=256)
+    bg = BatchedGenerator(eng, max_batch=4, max_wait_ms=60)
+    results = {}
+
+    def worker(i):
+        results[i] = bg.generate([f"question number {i} about GPUs?"],
+                                 max_new_tokens=8, temperature=0.0,
+                                 stop_on_eos=False)[0]
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    bg.batcher.stop()
+    assert len(results) == 4
+    assert all(isinstance(v, str) for v in results.values())
+    assert bg.batcher.stats["requests"] == 4
+    assert bg.batcher.stats["batches"] <= 3  # at least one shared batch
