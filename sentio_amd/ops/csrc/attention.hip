@@ -57,7 +57,7 @@ __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
 // transpose-reads are shared across the wave's QT row-tiles, and every
 // barrier/staging pass serves QT x 16 rows — per-row overhead halves at
 // QT=2 vs the one-tile version.
-#define FA_QT 2
+#define FA_QT 1   // measured: QT=2 amortizes fragments but costs occupancy (124 vs 156 TF/s) — QT=1 wins
 template <int DT>
 __launch_bounds__(256)
 __global__ void flash_attn_kernel(
