@@ -372,24 +372,35 @@ __global__ void decode_attn_split_kernel(
   const int Dbytes = D * 2;
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    // phase A1: K staging by LDS-DMA (`global_load_lds`, nt policy): the
-    // transfers land in LDS without consuming VGPRs, so ALL of the chunk's
-    // loads are in flight at once — register-staged 16 B loads capped
-    // in-flight bytes far below the per-CU HBM share (guide ldsdma-fill).
-    // The __syncthreads() drains them (its fence emits vmcnt(0)).
+    // phase A1: cooperative coalesced K staging into XOR-swizzled LDS.
+    // One predicated unrolled path for EVERY chunk (a separate dynamic tail
+    // loop serialized one load per s_waitcnt and dominated short spans):
+    // 8 loads in flight before any ds_write; out-of-range rows stage zeros.
     {
-      constexpr int IT = D / 8;
+      constexpr int IT = D / 8;          // loads per thread for a full chunk
+      constexpr int BATCH = 8;           // loads in flight
       const int lim = chunk * D;
 #pragma unroll
-      for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
-        if (i < lim) {
+      for (int u0 = 0; u0 < IT; u0 += BATCH) {
+        bf16x8 tmp[BATCH];
+#pragma unroll
+        for (int u = 0; u < BATCH; ++u) {
+          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
+          if (i < lim) {
+            tmp[u] = nt_load8(
+                reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
+                + (i % D));
+          } else {
+            bf16x8 z = {};
+            tmp[u] = z;
+          }
+        }
+#pragma unroll
+        for (int u = 0; u < BATCH; ++u) {
+          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
           const int row = i / D, d = i % D;
-          auto gp = (const __attribute__((address_space(1))) unsigned int*)(
-              kb + (long)(s0 + row) * D + d);
-          auto lp = (__attribute__((address_space(3))) unsigned int*)(
-              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
-          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 2 /* nt */);
+          *reinterpret_cast<bf16x8*>(
+              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = tmp[u];
         }
       }
     }
