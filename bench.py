@@ -461,7 +461,7 @@ def main():
                 "model": args.model,
                 "encoder": args.encoder,
                 "reranker": args.reranker,
-                "global_batch": args.batch * world,
+                "global_batch": args.batch * (1 if tp_mode else world),
                 "seq_len": args.seq_len,
                 "gen_tokens": args.gen_tokens,
                 "docs_per_gpu": args.docs_per_gpu,
