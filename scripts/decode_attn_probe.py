@@ -17,6 +17,10 @@ def bench(q, kc, vc, lens, iters=200):
     return (time.perf_counter() - t0) / iters
 
 def main():
+    if not torch.cuda.is_available():
+        print(f"{__file__}: needs a GPU (MI355X) — skipping")
+        return
+
     dev = "cuda:0"
     for B, slen in [(16, 1600), (32, 1600), (32, 512)]:
         H, Hkv, Smax, D = 32, 8, 2120, 128

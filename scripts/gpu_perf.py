@@ -128,6 +128,10 @@ def bench_generator(name="llama3-1b", B=8, S=512, new=32):
 
 
 def main():
+    if not torch.cuda.is_available():
+        print(f"{__file__}: needs a GPU (MI355X) — skipping")
+        return
+
     os.makedirs("gpurun_out", exist_ok=True)
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     steps = [

@@ -13,6 +13,10 @@ from sentio_amd import ops
 
 
 def main():
+    if not torch.cuda.is_available():
+        print(f"{__file__}: needs a GPU (MI355X) — skipping")
+        return
+
     dev = "cuda:0"
     M = K = N = 128
     a = torch.zeros(M, K, dtype=torch.bfloat16, device=dev)
