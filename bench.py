@@ -400,19 +400,27 @@ def main():
     # thread in deterministic step order.
     pipelined = not tp_mode
 
-    def run_pipelined(base: int, count: int) -> None:
+    step_times: list[float] = []
+
+    def run_pipelined(base: int, count: int, record: bool = False) -> None:
         if count <= 0:
             return
         if not pipelined:
             for i in range(count):
+                ts = time.perf_counter()
                 generate_phase(retrieval_phase_streamed(base + i))
+                if record:
+                    step_times.append(time.perf_counter() - ts)
             return
         fut = ret_ex.submit(retrieval_phase_streamed, base)
         for i in range(count):
+            ts = time.perf_counter()
             payload = fut.result()
             if i + 1 < count:
                 fut = ret_ex.submit(retrieval_phase_streamed, base + i + 1)
             generate_phase(payload)
+            if record:
+                step_times.append(time.perf_counter() - ts)
 
     # ---- warmup ----
     run_pipelined(0, args.warmup)
@@ -422,7 +430,7 @@ def main():
 
     # ---- timed ----
     t0 = time.perf_counter()
-    run_pipelined(1000, args.steps)
+    run_pipelined(1000, args.steps, record=True)
     D.barrier()
     if on_gpu:
         torch.cuda.synchronize()
@@ -472,7 +480,13 @@ def main():
                             + (">verify" if args.verify else "")
                             + ("|retrieval pipelined under prior step's "
                                "generation" if pipelined else ""),
-                "p50_ms_per_request_batch": round(ms_per_step, 1),
+                "p50_ms_per_request_batch": round(
+                    sorted(step_times)[len(step_times) // 2] * 1e3, 1)
+                    if step_times else round(ms_per_step, 1),
+                "p95_ms_per_request_batch": round(
+                    sorted(step_times)[min(len(step_times) - 1,
+                                           int(len(step_times) * 0.95))] * 1e3,
+                    1) if step_times else None,
                 "init_s": round(init_s, 1),
                 "device": "cuda" if on_gpu else "cpu-plumbing",
             },
