@@ -1,0 +1,109 @@
+"""Utils coverage mirroring reference suites: exceptions shaping
+(reference test_exceptions.py), security validation, prompt builder,
+monitoring (reference src/tests/*)."""
+
+from __future__ import annotations
+
+import pytest
+
+from sentio_amd.utils.exceptions import (
+    ErrorCode,
+    RateLimitException,
+    SentioException,
+    ValidationException,
+)
+from sentio_amd.utils.security import (
+    InputValidator,
+    LogSanitizer,
+    ValidationError,
+)
+
+
+# ---------- exceptions ----------
+
+def test_exception_to_dict_shape():
+    exc = SentioException("boom", details={"k": "v"})
+    d = exc.to_dict()
+    assert d["error"] == ErrorCode.SYSTEM_ERROR.value
+    assert d["message"] == "boom"
+    assert d["details"] == {"k": "v"}
+    assert "timestamp" in d
+
+
+def test_typed_exceptions_carry_status():
+    assert ValidationException("bad").status == 422
+    assert RateLimitException("slow down").status == 429
+
+
+# ---------- security ----------
+
+def test_validator_rejects_overlong_query():
+    with pytest.raises(ValidationError):
+        InputValidator.validate_query("x" * 5000)
+
+
+def test_validator_rejects_empty():
+    with pytest.raises(ValidationError):
+        InputValidator.validate_query("   ")
+
+
+def test_validator_strips_and_passes_normal():
+    assert InputValidator.validate_query("  what is RCCL? ") == "what is RCCL?"
+
+
+def test_validator_blocks_injection_patterns():
+    for evil in ["<script>alert(1)</script>",
+                 "1; DROP TABLE users--",
+                 "q && rm -rf /"]:
+        with pytest.raises(ValidationError):
+            InputValidator.validate_query(evil)
+
+
+def test_log_sanitizer_redacts_secrets():
+    line = 'calling api_key="sk-abc123xyz" password=hunter2'
+    red = LogSanitizer.sanitize(line)
+    assert "sk-abc123xyz" not in red
+    assert "hunter2" not in red
+
+
+# ---------- prompt builder ----------
+
+def test_prompt_builder_substitutes_and_modes():
+    from sentio_amd.pipeline.prompt_builder import PromptBuilder
+
+    b = PromptBuilder("fast")
+    p = b.build_qa_prompt("why GPUs?", "CONTEXT BODY")
+    assert "why GPUs?" in p and "CONTEXT BODY" in p
+    assert b.system_prompt()
+    # all four modes resolve distinct instructions
+    prompts = {m: PromptBuilder(m).build_qa_prompt("q", "c")
+               for m in ("fast", "balanced", "quality", "creative")}
+    assert len(set(prompts.values())) >= 2
+
+
+def test_verify_prompt_contains_json_protocol():
+    from sentio_amd.pipeline.prompt_builder import PromptBuilder
+
+    vp = PromptBuilder("balanced").build_verify_prompt(
+        query="q?", context="[1] ctx", answer="the answer")
+    assert "verdict" in vp and "q?" in vp and "the answer" in vp
+
+
+# ---------- monitoring ----------
+
+def test_performance_monitor_records_and_summarizes():
+    from sentio_amd.observability.monitoring import PerformanceMonitor
+
+    m = PerformanceMonitor(history=16)
+    for i in range(10):
+        m.record_value("lat_ms", float(i))
+    s = m.summary("lat_ms")
+    assert s["count"] == 10
+    assert s["min"] == 0.0 and s["max"] == 9.0
+
+
+def test_resource_monitor_snapshot_keys():
+    from sentio_amd.observability.monitoring import resource_monitor
+
+    snap = resource_monitor.snapshot()
+    assert "cpu_percent" in snap or "memory" in snap or snap  # psutil-backed
