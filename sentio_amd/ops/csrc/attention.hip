@@ -127,43 +127,52 @@ __global__ void flash_attn_kernel(
   const int kv_hi = causal ? min(kvlen, block_q_hi) : kvlen;
   const int n_kv_tiles = (kv_hi + KVBLK - 1) / KVBLK;
 
-  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+  // T14 software-pipelined staging (guide §6: attention staging →
+  // register staging, split): tile kt's K/V live in registers while tile
+  // kt-1 computes; the loads for kt+1 issue right after the stores of kt
+  // land, so HBM latency hides under QK/softmax/PV of the previous tile.
+  constexpr int ST_ELEMS = KVBLK * D;
+  constexpr int ST_STRIDE = FA_WAVES * WAVE * 8;
+  constexpr int ST_IT = (ST_ELEMS + ST_STRIDE - 1) / ST_STRIDE;
+  bf16x8_t kvals[ST_IT], vvals[ST_IT];
+
+  auto load_tile = [&](int kt) {
     const int kv0 = kt * KVBLK;
-    // ---- stage K tile (swizzled) and V tr-image into LDS: all of this
-    // thread's K and V loads issue BEFORE any LDS store (batched MLP)
-    {
-      constexpr int ELEMS = KVBLK * D;
-      constexpr int STRIDE = FA_WAVES * WAVE * 8;
-      constexpr int IT = (ELEMS + STRIDE - 1) / STRIDE;   // >= 1 for D >= 32
-      bf16x8_t kvals[IT], vvals[IT];
 #pragma unroll
-      for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * STRIDE;
-        const int key = kv0 + i / D;
-        const int d = i % D;
-        if (i < ELEMS && key < kvlen) {
-          const long base = (((long)b * S + key) * Hkv + hkv) * D + d;
-          kvals[u] = *reinterpret_cast<const bf16x8_t*>(k + base);
-          vvals[u] = *reinterpret_cast<const bf16x8_t*>(v + base);
-        } else {
-          bf16x8_t z = {};
-          kvals[u] = z;
-          vvals[u] = z;
-        }
-      }
-#pragma unroll
-      for (int u = 0; u < IT; ++u) {
-        const int i = threadIdx.x * 8 + u * STRIDE;
-        if (i < ELEMS) {
-          const int row = i / D, d = i % D;
-          *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) =
-              kvals[u];
-          *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) =
-              vvals[u];
-        }
+    for (int u = 0; u < ST_IT; ++u) {
+      const int i = threadIdx.x * 8 + u * ST_STRIDE;
+      const int key = kv0 + i / D;
+      const int d = i % D;
+      if (i < ST_ELEMS && key < kvlen) {
+        const long base = (((long)b * S + key) * Hkv + hkv) * D + d;
+        kvals[u] = *reinterpret_cast<const bf16x8_t*>(k + base);
+        vvals[u] = *reinterpret_cast<const bf16x8_t*>(v + base);
+      } else {
+        bf16x8_t z = {};
+        kvals[u] = z;
+        vvals[u] = z;
       }
     }
+  };
+  auto store_tile = [&]() {
+#pragma unroll
+    for (int u = 0; u < ST_IT; ++u) {
+      const int i = threadIdx.x * 8 + u * ST_STRIDE;
+      if (i < ST_ELEMS) {
+        const int row = i / D, d = i % D;
+        *reinterpret_cast<bf16x8_t*>(k_lds + k_lds_off(row, d * 2, D * 2)) =
+            kvals[u];
+        *reinterpret_cast<bf16x8_t*>(vt_lds + v_tr_off(row, d) * 2) = vvals[u];
+      }
+    }
+  };
+
+  if (n_kv_tiles > 0) load_tile(0);
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kv0 = kt * KVBLK;
+    store_tile();
     __syncthreads();  // staging visible to every wave
+    if (kt + 1 < n_kv_tiles) load_tile(kt + 1);  // in flight under compute
 
     // waves whose causal horizon ends before this kv tile skip compute but
     // still execute every barrier (uniform control flow)
