@@ -144,7 +144,18 @@ def mean_pool_l2norm(hidden, mask):
 
 def cosine_topk(q, mat, k: int):
     if _on_gpu(q, mat):
-        scores = _require_hip().cosine_scores(q.contiguous(), mat.contiguous())
+        hip = _require_hip()
+        # the kernel stages the query block in LDS (B*D*4 bytes ≤ 160 KB):
+        # chunk large query batches
+        max_q = max(1, (150 * 1024) // (q.shape[1] * 4))
+        mat_c = mat.contiguous()
+        if q.shape[0] <= max_q:
+            scores = hip.cosine_scores(q.contiguous(), mat_c)
+        else:
+            scores = torch.cat([
+                hip.cosine_scores(q[i:i + max_q].contiguous(), mat_c)
+                for i in range(0, q.shape[0], max_q)
+            ], dim=0)
         return torch.topk(scores, k, dim=1)
     return torch_ref.cosine_topk(q, mat, k)
 

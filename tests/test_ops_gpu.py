@@ -408,3 +408,20 @@ def test_batched_generator_concurrent_gpu(dev):
     assert all(isinstance(v, str) for v in results.values())
     assert bg.batcher.stats["requests"] == 4
     assert bg.batcher.stats["batches"] <= 3  # at least one shared batch
+
+
+def test_cosine_topk_large_query_batch(dev):
+    """Query batches beyond the kernel's LDS staging limit chunk correctly
+    (B=48 at dim 1024 would need 196 KB of LDS)."""
+    from sentio_amd import ops
+
+    torch.manual_seed(2)
+    q = torch.randn(48, 1024, dtype=torch.float16, device=dev)
+    mat = torch.randn(5000, 1024, dtype=torch.float16, device=dev)
+    vals, idx = ops.cosine_topk(q, mat, 5)
+    wv, wi = ops.torch_ref.cosine_topk(q.cpu().float(), mat.cpu().float(), 5)
+    assert vals.shape == (48, 5)
+    torch.testing.assert_close(vals.cpu().float(), wv, rtol=2e-2, atol=2e-2)
+    overlap = sum(len(set(idx[i].cpu().tolist()) & set(wi[i].tolist()))
+                  for i in range(48))
+    assert overlap >= 48 * 4   # ties may swap the tail
