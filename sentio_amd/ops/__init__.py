@@ -143,21 +143,25 @@ def mean_pool_l2norm(hidden, mask):
 
 
 def cosine_topk(q, mat, k: int):
+    """Batched cosine top-k over an [N, D] row-normalized index.
+
+    GPU path: the score matrix IS a TN GEMM — mat's [N, D] row-major layout
+    is exactly F.linear's weight layout, so hipBLASLt streams the index on
+    MFMA at memory rate (measured 0.20 TB/s for the hand wave-per-row scan
+    vs ~5 TB/s through the library GEMM at 10M docs — GEMM-shaped work
+    belongs on the GEMM path).  The hand kernel remains exposed as
+    ops.cosine_scores for small/irregular scans."""
     if _on_gpu(q, mat):
-        hip = _require_hip()
-        # the kernel stages the query block in LDS (B*D*4 bytes ≤ 160 KB):
-        # chunk large query batches
-        max_q = max(1, (150 * 1024) // (q.shape[1] * 4))
-        mat_c = mat.contiguous()
-        if q.shape[0] <= max_q:
-            scores = hip.cosine_scores(q.contiguous(), mat_c)
-        else:
-            scores = torch.cat([
-                hip.cosine_scores(q[i:i + max_q].contiguous(), mat_c)
-                for i in range(0, q.shape[0], max_q)
-            ], dim=0)
+        scores = torch.nn.functional.linear(q, mat).float()
         return torch.topk(scores, k, dim=1)
     return torch_ref.cosine_topk(q, mat, k)
+
+
+def cosine_scores(q, mat):
+    """Hand wave-per-row scan (LDS-staged queries); B*D*4 must fit in LDS."""
+    if _on_gpu(q, mat):
+        return _require_hip().cosine_scores(q.contiguous(), mat.contiguous())
+    return (q.float() @ mat.float().T)
 
 
 def bm25_score(term_ids, indptr, post_doc, post_tf, idf, doc_len, *,
@@ -241,5 +245,5 @@ def gemm_bf16(a, b):
 __all__ = [
     "rmsnorm", "rmsnorm_residual", "rope_apply", "decode_qkv_prep", "swiglu", "swiglu_packed", "softmax",
     "attention", "decode_attention", "mean_pool_l2norm", "cosine_topk",
-    "bm25_score", "fuse_topk", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
+    "bm25_score", "cosine_scores", "fuse_topk", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
 ]
