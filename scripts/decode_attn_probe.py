@@ -40,6 +40,45 @@ def main():
 
 
 
+def bmm_probe():
+    """Compare the hand decode-attention kernel against a batched-GEMM
+    formulation (rocBLAS bmm): scores = Q·K^T and O = P·V per (b, kv-head)
+    group, full-Smax with masking."""
+    dev = "cuda:0"
+    for B, slen in [(16, 1600), (32, 1600)]:
+        H, Hkv, Smax, D = 32, 8, 2120, 128
+        G = H // Hkv
+        torch.manual_seed(0)
+        q = torch.randn(B, H, D, dtype=torch.bfloat16, device=dev)
+        kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+        vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+        lens = torch.full((B,), slen, dtype=torch.int32, device=dev)
+        mask = (torch.arange(Smax, device=dev)[None, :]
+                >= lens[:, None]).view(B, 1, 1, Smax)
+
+        def bmm_path():
+            qg = q.view(B * Hkv, G, D)
+            kg = kc.view(B * Hkv, Smax, D)
+            s_ = torch.bmm(qg, kg.transpose(1, 2)) * (D ** -0.5)
+            s_ = s_.view(B, Hkv, G, Smax).masked_fill(mask, float("-inf"))
+            p_ = torch.softmax(s_.float(), dim=-1).to(torch.bfloat16)
+            return torch.bmm(p_.view(B * Hkv, G, Smax),
+                             vc.view(B * Hkv, Smax, D))
+
+        for name, fn in [("hand", lambda: ops.decode_attention(q, kc, vc, lens)),
+                         ("bmm", bmm_path)]:
+            for _ in range(10):
+                fn()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(100):
+                fn()
+            torch.cuda.synchronize()
+            t = (time.perf_counter() - t0) / 100
+            kv_bytes = 2 * B * Hkv * slen * D * 2
+            print(f"B={B} {name}: {t*1e6:7.1f}us  {kv_bytes/t/1e12:.2f} TB/s")
+
+
 def flash_probe():
     """Prefill flash-attention microbench (llama3-8b shape)."""
     dev = "cuda:0"
@@ -63,4 +102,5 @@ def flash_probe():
 
 if __name__ == "__main__":
     main()
+    bmm_probe()
     flash_probe()
