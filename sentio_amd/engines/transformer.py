@@ -223,17 +223,37 @@ class Transformer:
             normed, x = ops.rmsnorm_residual(f, x, w_next, eps)
         return normed
 
+    # Decode-attention path choice: the hand split-S kernel reads only the
+    # VALID cache rows; the bmm path streams the full allocation through
+    # rocBLAS ~1.4x faster.  Crossover ≈ 55% full (env override:
+    # SENTIO_DECODE_ATTN=hand|bmm).
+    _BMM_FILL_THRESHOLD = 0.55
+
+    def _decode_attn_impl(self, fill_ratio: float):
+        import os
+
+        mode = os.environ.get("SENTIO_DECODE_ATTN", "auto")
+        if mode == "hand":
+            return ops.decode_attention
+        if mode == "bmm":
+            return ops.decode_attention_bmm
+        return (ops.decode_attention_bmm
+                if fill_ratio >= self._BMM_FILL_THRESHOLD
+                else ops.decode_attention)
+
     # ----- decode fast path: fused RoPE + KV-cache write + fused norms -----
     def _attn_decode(self, normed: torch.Tensor, layer: dict, cache: KVCache,
-                     layer_idx: int, attn_lens: torch.Tensor) -> torch.Tensor:
+                     layer_idx: int, attn_lens: torch.Tensor,
+                     attn_fn=None) -> torch.Tensor:
         B = normed.shape[0]
         d = self.cfg.dim
         hd = self.cfg.head_dim
         qkv = torch.nn.functional.linear(normed.view(B, d), layer["wqkv"])
         q = ops.decode_qkv_prep(qkv, cache.k[layer_idx], cache.v[layer_idx],
                                 self.rope_cos, self.rope_sin, cache.seq_lens)
-        out = ops.decode_attention(q, cache.k[layer_idx], cache.v[layer_idx],
-                                   attn_lens, self.scale)
+        attn = attn_fn or ops.decode_attention
+        out = attn(q, cache.k[layer_idx], cache.v[layer_idx],
+                   attn_lens, self.scale)
         out = torch.nn.functional.linear(out.view(B, self.h_local * hd), layer["wo"])
         out = self.tp.all_reduce(out)
         return out.view(B, 1, d)
@@ -248,9 +268,14 @@ class Transformer:
         layers = self.w.layers
         n = len(layers)
         attn_lens = cache.seq_lens + 1
+        # hand vs bmm kernel chosen at prefill time (set by the caller via
+        # self.decode_attn_fn — no device sync here: this path is captured
+        # into hipGraphs)
+        attn_fn = getattr(self, "decode_attn_fn", None)
         normed = ops.rmsnorm(x, layers[0]["attn_norm"], eps)
         for i, layer in enumerate(layers):
-            a = self._attn_decode(normed, layer, cache, i, attn_lens)
+            a = self._attn_decode(normed, layer, cache, i, attn_lens,
+                                  attn_fn=attn_fn)
             normed, x = ops.rmsnorm_residual(a, x, layer["ffn_norm"], eps)
             f = self._ffn(normed, layer)
             w_next = layers[i + 1]["attn_norm"] if i + 1 < n else self.w.final_norm
@@ -263,10 +288,16 @@ class Transformer:
         return torch.nn.functional.linear(hidden[:, -1, :], self.w.lm_head).float()  # [B, V]
 
     def prefill(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
-        """Prefill the cache; returns last-position logits [B, V]."""
+        """Prefill the cache; returns last-position logits [B, V].  Also
+        picks the decode-attention implementation for the upcoming decode
+        steps from the cache fill ratio (prompt length is host-side here —
+        no sync)."""
         B, S = tokens.shape
         hidden = self.forward_hidden(tokens, cache=cache)
         cache.seq_lens[:] = S
+        if self.device != "cpu":
+            self.decode_attn_fn = self._decode_attn_impl(
+                S / max(cache.max_seq, 1))
         return self.logits(hidden)
 
     def decode_step(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:

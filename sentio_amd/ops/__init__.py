@@ -134,6 +134,34 @@ def decode_attention(q, k_cache, v_cache, seq_lens, scale: float | None = None):
     return torch_ref.decode_attention(q, k_cache, v_cache, seq_lens, scale)
 
 
+def decode_attention_bmm(q, k_cache, v_cache, seq_lens,
+                         scale: float | None = None):
+    """Decode attention as two rocBLAS batched GEMMs over the FULL cache
+    extent (invalid keys masked): scores = Q·K^T per (b, kv-head) group,
+    softmax, O = P·V.  Reads Smax rows instead of slen, but the library
+    GEMM streams them ~1.4x faster than the hand split-S kernel when the
+    cache is mostly full (measured 108 vs 153 us at B=32, slen/Smax≈0.75) —
+    the hand kernel (ops.decode_attention) wins for mostly-empty caches.
+    All ops are tensor-graph-capturable (the mask reads seq_lens in-graph).
+    q: [B, H, D]; caches [B, Hkv, Smax, D]; returns [B, H, D]."""
+    import math
+
+    B, H, D = q.shape
+    Hkv, Smax = k_cache.shape[1], k_cache.shape[2]
+    G = H // Hkv
+    s = scale if scale is not None else 1.0 / math.sqrt(D)
+    if not _on_gpu(q):
+        return torch_ref.decode_attention(q, k_cache, v_cache, seq_lens, scale)
+    mask = (torch.arange(Smax, device=q.device)[None, :]
+            >= seq_lens[:, None]).view(B, 1, 1, Smax)
+    qg = q.view(B * Hkv, G, D)
+    sc = torch.bmm(qg, k_cache.view(B * Hkv, Smax, D).transpose(1, 2)) * s
+    sc = sc.view(B, Hkv, G, Smax).masked_fill(mask, float("-inf"))
+    p = torch.softmax(sc.float(), dim=-1).to(q.dtype)
+    out = torch.bmm(p.view(B * Hkv, G, Smax), v_cache.view(B * Hkv, Smax, D))
+    return out.view(B, H, D)
+
+
 # ---------------- pooling / retrieval ----------------
 
 def mean_pool_l2norm(hidden, mask):
@@ -244,6 +272,6 @@ def gemm_bf16(a, b):
 
 __all__ = [
     "rmsnorm", "rmsnorm_residual", "rope_apply", "decode_qkv_prep", "swiglu", "swiglu_packed", "softmax",
-    "attention", "decode_attention", "mean_pool_l2norm", "cosine_topk",
+    "attention", "decode_attention", "decode_attention_bmm", "mean_pool_l2norm", "cosine_topk",
     "bm25_score", "cosine_scores", "fuse_topk", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
 ]

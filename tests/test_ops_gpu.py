@@ -425,3 +425,18 @@ def test_cosine_topk_large_query_batch(dev):
     overlap = sum(len(set(idx[i].cpu().tolist()) & set(wi[i].tolist()))
                   for i in range(48))
     assert overlap >= 48 * 4   # ties may swap the tail
+
+
+def test_decode_attention_bmm_matches_ref(dev):
+    from sentio_amd import ops
+
+    B, H, Hkv, Smax, D = 3, 8, 2, 300, 128
+    torch.manual_seed(4)
+    q = torch.randn(B, H, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    lens = torch.tensor([300, 257, 1], dtype=torch.int32, device=dev)
+    got = ops.decode_attention_bmm(q, kc, vc, lens)
+    want = ops.torch_ref.decode_attention(q.cpu().float(), kc.cpu().float(),
+                                          vc.cpu().float(), lens.cpu())
+    _cmp(got, want, rtol=3e-2, atol=3e-2)
