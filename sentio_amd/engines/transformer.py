@@ -223,11 +223,12 @@ class Transformer:
             normed, x = ops.rmsnorm_residual(f, x, w_next, eps)
         return normed
 
-    # Decode-attention path choice: the hand split-S kernel reads only the
-    # VALID cache rows; the bmm path streams the full allocation through
-    # rocBLAS ~1.4x faster.  Crossover ≈ 55% full (env override:
-    # SENTIO_DECODE_ATTN=hand|bmm).
-    _BMM_FILL_THRESHOLD = 0.55
+    # Decode-attention path choice (env override SENTIO_DECODE_ATTN=hand|bmm):
+    # the bmm formulation looked ~1.4x faster in an isolated cold-box probe,
+    # but IN CONTEXT the extra P-matrix traffic + per-layer kernel tail made
+    # the end-to-end bench 7% slower (16.8 vs 18.1 QPS) — the hand split-S
+    # kernel with fused online softmax stays the default at any fill.
+    _BMM_FILL_THRESHOLD = 2.0   # auto never picks bmm
 
     def _decode_attn_impl(self, fill_ratio: float):
         import os
