@@ -52,6 +52,14 @@ MODEL_CONFIGS: dict[str, ModelConfig] = {
         name="llama3-70b", dim=8192, n_layers=80, n_heads=64, n_kv_heads=8,
         ffn_dim=28672, vocab_size=128256, max_seq=8192,
     ),
+    "qwen2-7b": ModelConfig(   # Qwen-2 family shape (GQA 28/4, 3584 dim)
+        name="qwen2-7b", dim=3584, n_layers=28, n_heads=28, n_kv_heads=4,
+        ffn_dim=18944, vocab_size=152064, max_seq=8192, rope_base=1000000.0,
+    ),
+    "mistral-7b": ModelConfig(  # Mistral-7B family shape (GQA 32/8)
+        name="mistral-7b", dim=4096, n_layers=32, n_heads=32, n_kv_heads=8,
+        ffn_dim=14336, vocab_size=32768, max_seq=8192, rope_base=1000000.0,
+    ),
     "llama3-1b": ModelConfig(  # small real decoder for quick GPU checks
         name="llama3-1b", dim=2048, n_layers=16, n_heads=32, n_kv_heads=8,
         ffn_dim=8192, vocab_size=128256, max_seq=8192,
