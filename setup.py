@@ -39,6 +39,12 @@ def build(verbose: bool = True) -> Path:
         "hipcc", f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
         "-DNDEBUG",
     ]
+    # SENTIO_SANITIZE=1: host-side ASan+UBSan test build (SURVEY §5 race/
+    # sanitizer row).  Run with LD_PRELOAD of the ASan runtime, e.g.
+    #   LD_PRELOAD=$(hipcc -print-file-name=libclang_rt.asan-x86_64.so) pytest
+    if os.environ.get("SENTIO_SANITIZE") == "1":
+        common += ["-fsanitize=address,undefined",
+                   "-fno-omit-frame-pointer", "-shared-libasan"]
     objs = []
 
     def compile_tu(src: Path, extra: list[str]) -> Path:
