@@ -440,3 +440,33 @@ def test_decode_attention_bmm_matches_ref(dev):
     want = ops.torch_ref.decode_attention(q.cpu().float(), kc.cpu().float(),
                                           vc.cpu().float(), lens.cpu())
     _cmp(got, want, rtol=3e-2, atol=3e-2)
+
+
+def test_full_pipeline_with_verifier_gpu(dev):
+    """Config #4 semantics on device: full graph incl. verifier."""
+    from sentio_amd.config import Settings
+    from sentio_amd.models.document import Document
+    from sentio_amd.serving.container import ServiceContainer
+    from sentio_amd.serving.handlers import ChatHandler
+
+    s = Settings()
+    s.device = "cuda"
+    s.mock_compute = False
+    s.encoder_model = "sentio-encoder-small"
+    s.generator_model = "llama3-1b"
+    s.reranker_model = "sentio-reranker-base"
+    s.use_reranker = True
+    s.use_verifier = True
+    s.llm_max_tokens = 12
+    s.verifier_max_tokens = 12
+    s.dynamic_batching = False
+    c = ServiceContainer(s)
+    c.initialize_all()
+    c.ingestor().ingest_documents([
+        Document(text=f"verified doc {i}: xGMI links carry RCCL traffic",
+                 id=f"v{i}") for i in range(6)
+    ])
+    out = ChatHandler(c).process("what carries RCCL traffic?")
+    assert out["answer"]
+    v = out["metadata"].get("verification")
+    assert v is not None and v.get("verdict") in ("pass", "warn", "fail")
