@@ -191,11 +191,18 @@ class Transformer:
     def _ffn(self, x: torch.Tensor, layer: dict) -> torch.Tensor:
         B, S, d = x.shape
         # measured: hipBLASLt's TN path beats the hand skinny_gemm kernel on
-        # the decode gate_up shape in context (63 vs ~87 us at B=32) — the
-        # library call stays; ops.skinny_gemm remains available standalone
-        gu = torch.nn.functional.linear(x.view(B * S, d), layer["w_gate_up"])
-        y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
-        out = torch.nn.functional.linear(y, layer["w_down"])
+        # the decode gate_up shape in context (63 vs ~87 us at B=32); decode
+        # rows additionally go through the autotuned-algorithm binding
+        rows = B * S
+        xx = x.view(rows, d)
+        if rows <= 64 and self.device != "cpu":
+            gu = ops.lt_linear(xx, layer["w_gate_up"])
+            y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
+            out = ops.lt_linear(y, layer["w_down"])
+        else:
+            gu = torch.nn.functional.linear(xx, layer["w_gate_up"])
+            y = ops.swiglu_packed(gu)
+            out = torch.nn.functional.linear(y, layer["w_down"])
         out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
         return out.view(B, S, d)
 
@@ -249,13 +256,13 @@ class Transformer:
         B = normed.shape[0]
         d = self.cfg.dim
         hd = self.cfg.head_dim
-        qkv = torch.nn.functional.linear(normed.view(B, d), layer["wqkv"])
+        qkv = ops.lt_linear(normed.view(B, d), layer["wqkv"])
         q = ops.decode_qkv_prep(qkv, cache.k[layer_idx], cache.v[layer_idx],
                                 self.rope_cos, self.rope_sin, cache.seq_lens)
         attn = attn_fn or ops.decode_attention
         out = attn(q, cache.k[layer_idx], cache.v[layer_idx],
                    attn_lens, self.scale)
-        out = torch.nn.functional.linear(out.view(B, self.h_local * hd), layer["wo"])
+        out = ops.lt_linear(out.view(B, self.h_local * hd), layer["wo"])
         out = self.tp.all_reduce(out)
         return out.view(B, 1, d)
 
@@ -286,7 +293,10 @@ class Transformer:
     # ----- decoder-specific -----
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
         B, S, d = hidden.shape
-        return torch.nn.functional.linear(hidden[:, -1, :], self.w.lm_head).float()  # [B, V]
+        last = hidden[:, -1, :]
+        if B <= 64 and self.device != "cpu":
+            return ops.lt_linear(last.contiguous(), self.w.lm_head).float()
+        return torch.nn.functional.linear(last, self.w.lm_head).float()  # [B, V]
 
     def prefill(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
         """Prefill the cache; returns last-position logits [B, V].  Also

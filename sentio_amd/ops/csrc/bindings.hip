@@ -40,6 +40,8 @@ hipError_t sentio_gemm_bf16(const void*, const void*, void*, int, int, int,
                             hipStream_t);
 hipError_t sentio_skinny_gemm(const void*, const void*, void*, int, int, int,
                               hipStream_t);
+int sentio_lt_gemm_tn(const void*, const void*, void*, int, int, int,
+                      hipStream_t);
 hipError_t sentio_fuse_topk(const long*, const float*, const long*,
                             const float*, long*, float*, int, int, int, int,
                             int, float, float, float, hipStream_t);
@@ -319,6 +321,19 @@ std::vector<torch::Tensor> fuse_topk(torch::Tensor d_ids, torch::Tensor d_scores
   return {out_ids, out_scores};
 }
 
+torch::Tensor lt_gemm_tn(torch::Tensor x, torch::Tensor w) {
+  check_bf16_cuda(x, "x");
+  check_bf16_cuda(w, "w");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+              "x [M,K], w [N,K]");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  auto out = torch::empty({M, N}, x.options());
+  int rc = sentio_lt_gemm_tn(x.data_ptr(), w.data_ptr(), out.data_ptr(), M, N,
+                             K, stream());
+  TORCH_CHECK(rc == 0, "sentio_lt_gemm_tn failed rc=", rc);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_residual", &rmsnorm_residual);
@@ -336,4 +351,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16", &gemm_bf16);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("fuse_topk", &fuse_topk);
+  m.def("lt_gemm_tn", &lt_gemm_tn);
 }

@@ -1,0 +1,167 @@
+// hipBLASLt TN GEMM with per-shape algorithm autotuning.
+//
+// torch's F.linear lets hipBLASLt's heuristic pick ONE algorithm; for the
+// skinny decode projections (M <= 64 rows vs multi-thousand-column weights)
+// the heuristic's pick measured ~35% off the best available tile
+// (profiles/README).  This TU asks the heuristic for its top candidates,
+// times each on the live stream ONCE per (M, N, K) shape (outside graph
+// capture — the generator's eager warmup runs before its hipGraph is
+// cut), caches the winner, and replays it thereafter — capture-safe, no
+// syncs after the first call.
+//
+// out[M, N] = x[M, K] @ w[N, K]^T   (all bf16 row-major, fp32 accumulate)
+// Column-major mapping: C_cm[N, M] = A^T (A = w_cm[K, N], opT) * B
+// (B = x_cm[K, M], opN).
+#include <hip/hip_runtime.h>
+#include <hipblaslt/hipblaslt.h>
+
+#include <cstdio>
+#include <map>
+#include <mutex>
+#include <tuple>
+#include <vector>
+
+namespace {
+
+constexpr size_t kWorkspaceBytes = 128ull << 20;
+constexpr int kHeuristicCandidates = 24;
+constexpr int kTimingIters = 8;
+
+struct LtState {
+  hipblasLtHandle_t handle = nullptr;
+  void* workspace = nullptr;
+  std::mutex mu;
+  std::map<std::tuple<int, int, int>, hipblasLtMatmulAlgo_t> algo_cache;
+};
+
+LtState& state() {
+  static LtState s;
+  static std::once_flag once;
+  std::call_once(once, [] {
+    hipblasLtCreate(&s.handle);
+    hipMalloc(&s.workspace, kWorkspaceBytes);
+  });
+  return s;
+}
+
+struct Descs {
+  hipblasLtMatmulDesc_t op = nullptr;
+  hipblasLtMatrixLayout_t a = nullptr, b = nullptr, c = nullptr;
+
+  bool init(int M, int N, int K) {
+    if (hipblasLtMatmulDescCreate(&op, HIPBLAS_COMPUTE_32F, HIP_R_32F) !=
+        HIPBLAS_STATUS_SUCCESS)
+      return false;
+    hipblasOperation_t opT = HIPBLAS_OP_T, opN = HIPBLAS_OP_N;
+    hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSA, &opT,
+                                    sizeof(opT));
+    hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSB, &opN,
+                                    sizeof(opN));
+    // A = w as column-major [K, N] (lda = K), transposed -> [N, K]
+    // B = x as column-major [K, M] (ldb = K)
+    // C = out as column-major [N, M] (ldc = N)
+    if (hipblasLtMatrixLayoutCreate(&a, HIP_R_16BF, K, N, K) !=
+            HIPBLAS_STATUS_SUCCESS ||
+        hipblasLtMatrixLayoutCreate(&b, HIP_R_16BF, K, M, K) !=
+            HIPBLAS_STATUS_SUCCESS ||
+        hipblasLtMatrixLayoutCreate(&c, HIP_R_16BF, N, M, N) !=
+            HIPBLAS_STATUS_SUCCESS)
+      return false;
+    return true;
+  }
+  ~Descs() {
+    if (op) hipblasLtMatmulDescDestroy(op);
+    if (a) hipblasLtMatrixLayoutDestroy(a);
+    if (b) hipblasLtMatrixLayoutDestroy(b);
+    if (c) hipblasLtMatrixLayoutDestroy(c);
+  }
+};
+
+bool autotune(LtState& s, Descs& d, int M, int N, int K, const void* x,
+              const void* w, void* out, hipStream_t stream,
+              hipblasLtMatmulAlgo_t* best_out) {
+  hipblasLtMatmulPreference_t pref;
+  if (hipblasLtMatmulPreferenceCreate(&pref) != HIPBLAS_STATUS_SUCCESS)
+    return false;
+  size_t ws = kWorkspaceBytes;
+  hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws));
+  hipblasLtMatmulHeuristicResult_t results[kHeuristicCandidates];
+  int n_results = 0;
+  hipblasStatus_t st = hipblasLtMatmulAlgoGetHeuristic(
+      s.handle, d.op, d.a, d.b, d.c, d.c, pref, kHeuristicCandidates, results,
+      &n_results);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  if (st != HIPBLAS_STATUS_SUCCESS || n_results == 0) return false;
+
+  const float alpha = 1.f, beta = 0.f;
+  hipEvent_t ev0, ev1;
+  hipEventCreate(&ev0);
+  hipEventCreate(&ev1);
+  float best_ms = 1e30f;
+  int best_i = -1;
+  for (int i = 0; i < n_results; ++i) {
+    if (results[i].state != HIPBLAS_STATUS_SUCCESS) continue;
+    // warm
+    if (hipblasLtMatmul(s.handle, d.op, &alpha, w, d.a, x, d.b, &beta, out,
+                        d.c, out, d.c, &results[i].algo, s.workspace,
+                        kWorkspaceBytes, stream) != HIPBLAS_STATUS_SUCCESS)
+      continue;
+    hipEventRecord(ev0, stream);
+    for (int it = 0; it < kTimingIters; ++it)
+      hipblasLtMatmul(s.handle, d.op, &alpha, w, d.a, x, d.b, &beta, out, d.c,
+                      out, d.c, &results[i].algo, s.workspace, kWorkspaceBytes,
+                      stream);
+    hipEventRecord(ev1, stream);
+    hipEventSynchronize(ev1);
+    float ms = 0.f;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    if (ms < best_ms) {
+      best_ms = ms;
+      best_i = i;
+    }
+  }
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
+  if (best_i < 0) return false;
+  *best_out = results[best_i].algo;
+  return true;
+}
+
+}  // namespace
+
+extern "C" {
+
+// Returns 0 on success.  First call for a shape autotunes (SYNCS the
+// stream); later calls replay the cached algorithm (capture-safe).
+int sentio_lt_gemm_tn(const void* x, const void* w, void* out, int M, int N,
+                      int K, hipStream_t stream) {
+  LtState& s = state();
+  if (!s.handle || !s.workspace) return 1;
+  Descs d;
+  if (!d.init(M, N, K)) return 2;
+
+  hipblasLtMatmulAlgo_t algo;
+  bool have = false;
+  {
+    std::lock_guard<std::mutex> lk(s.mu);
+    auto it = s.algo_cache.find({M, N, K});
+    if (it != s.algo_cache.end()) {
+      algo = it->second;
+      have = true;
+    }
+  }
+  if (!have) {
+    if (!autotune(s, d, M, N, K, x, w, out, stream, &algo)) return 3;
+    std::lock_guard<std::mutex> lk(s.mu);
+    s.algo_cache[{M, N, K}] = algo;
+  }
+  const float alpha = 1.f, beta = 0.f;
+  if (hipblasLtMatmul(s.handle, d.op, &alpha, w, d.a, x, d.b, &beta, out, d.c,
+                      out, d.c, &algo, s.workspace, kWorkspaceBytes,
+                      stream) != HIPBLAS_STATUS_SUCCESS)
+    return 4;
+  return 0;
+}
+
+}  // extern "C"

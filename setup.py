@@ -19,7 +19,7 @@ CSRC = ROOT / "sentio_amd" / "ops" / "csrc"
 OUT = ROOT / "sentio_amd" / "ops" / "_sentio_hip.so"
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-KERNEL_TUS = ["elementwise.hip", "retrieval.hip", "attention.hip", "gemm.hip"]
+KERNEL_TUS = ["elementwise.hip", "retrieval.hip", "attention.hip", "gemm.hip", "blaslt.hip"]
 BINDING_TU = "bindings.hip"
 
 
@@ -75,7 +75,7 @@ def build(verbose: bool = True) -> Path:
         + ["-shared", "-o", str(OUT)]
         + [str(o) for o in objs]
         + [f"-L{d}" for d in libdirs]
-        + ["-ltorch", "-ltorch_hip", "-lc10", "-lc10_hip", "-ltorch_python"]
+        + ["-ltorch", "-ltorch_hip", "-lc10", "-lc10_hip", "-ltorch_python", "-lhipblaslt"]
         + [f"-Wl,-rpath,{d}" for d in libdirs]
     )
     if verbose:

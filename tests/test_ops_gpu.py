@@ -470,3 +470,15 @@ def test_full_pipeline_with_verifier_gpu(dev):
     assert out["answer"]
     v = out["metadata"].get("verification")
     assert v is not None and v.get("verdict") in ("pass", "warn", "fail")
+
+
+def test_lt_gemm_tn_matches_linear(dev):
+    from sentio_amd import ops
+
+    torch.manual_seed(6)
+    for M, K, N in [(16, 4096, 6144), (32, 4096, 4096), (7, 512, 384)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.3
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.3
+        got = ops.lt_linear(x, w)
+        want = torch.nn.functional.linear(x.float(), w.float())
+        _cmp(got, want, rtol=3e-2, atol=3e-1)
