@@ -482,3 +482,20 @@ def test_lt_gemm_tn_matches_linear(dev):
         got = ops.lt_linear(x, w)
         want = torch.nn.functional.linear(x.float(), w.float())
         _cmp(got, want, rtol=3e-2, atol=3e-1)
+
+
+def test_decode_attn_gqa7(dev):
+    """G = H/Hkv = 7 (Qwen-2 family group size) exercises the odd-G
+    template instantiations."""
+    from sentio_amd import ops
+
+    B, H, Hkv, Smax, D = 2, 14, 2, 200, 128
+    torch.manual_seed(8)
+    q = torch.randn(B, H, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    lens = torch.tensor([200, 63], dtype=torch.int32, device=dev)
+    got = ops.decode_attention(q, kc, vc, lens)
+    want = ops.torch_ref.decode_attention(q.cpu().float(), kc.cpu().float(),
+                                          vc.cpu().float(), lens.cpu())
+    _cmp(got, want, rtol=3e-2, atol=3e-2)
