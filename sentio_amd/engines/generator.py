@@ -41,6 +41,19 @@ class _DecodeSession:
         self.cache.seq_lens.zero_()
         return self.engine.model.prefill(tokens, self.cache)
 
+    def prefill_with_prefix(self, prefix_ids: list[int],
+                            suffix_tokens: torch.Tensor) -> torch.Tensor:
+        """Copy the (cached) prefix KV into every batch slot, then prefill
+        only the suffix (prefix-KV caching: the shared system-prompt +
+        instruction header is prefilled ONCE per prefix, not per request)."""
+        engine = self.engine
+        pk, pv = engine._prefix_kv(tuple(prefix_ids))
+        P = len(prefix_ids)
+        for i in range(engine.cfg.n_layers):
+            self.cache.k[i][:, :, :P].copy_(pk[i])   # [1,...] broadcasts
+            self.cache.v[i][:, :, :P].copy_(pv[i])
+        return engine.model.prefill_suffix(suffix_tokens, self.cache, P)
+
     def _capture(self):
         model = self.engine.model
         dev = self.engine.device
@@ -90,6 +103,30 @@ class GeneratorEngine:
     def _new_cache(self, batch: int, max_seq: int) -> KVCache:
         return KVCache(self.cfg, batch, max_seq, self.device, self.model.dtype,
                        n_kv_heads=self.model.hkv_local)
+
+    # ----- prefix-KV cache (shared prompt prefixes prefilled once) -----
+    _PREFIX_MIN_TOKENS = 64
+
+    def _prefix_kv(self, prefix_ids: tuple):
+        """Per-prefix KV, computed once with a batch-1 forward and kept in
+        HBM (a few tens of MB per distinct prefix; LRU of 4)."""
+        store = getattr(self, "_prefix_store", None)
+        if store is None:
+            store = self._prefix_store = {}
+        hit = store.get(prefix_ids)
+        if hit is not None:
+            return hit
+        P = len(prefix_ids)
+        tmp = KVCache(self.cfg, 1, P, self.device, self.model.dtype,
+                      n_kv_heads=self.model.hkv_local)
+        toks = torch.tensor([list(prefix_ids)], dtype=torch.int64,
+                            device=self.device)
+        self.model.forward_hidden(toks, cache=tmp)
+        entry = ([k.clone() for k in tmp.k], [v.clone() for v in tmp.v])
+        if len(store) >= 4:
+            store.pop(next(iter(store)))
+        store[prefix_ids] = entry
+        return entry
 
     # ----- hipGraph-captured decode session -----
     # The per-token decode step launches ~10 kernels per layer; on MI355X the
@@ -151,14 +188,37 @@ class GeneratorEngine:
                          stop_on_eos, on_token, _time) -> list[str]:
         B = len(prompts)
         prompt_budget = self.max_seq - max_new_tokens - 1
-        padded, lens = self.tokenizer.encode_batch(
-            [p[-4 * prompt_budget:] for p in prompts], prompt_budget
-        )
-        tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
-        S = tokens.shape[1]
+        # prefix-KV caching: requests share the system-prompt + instruction
+        # header; prefill it once per distinct prefix and only forward each
+        # request's suffix (byte tokenizer → char-exact prefix alignment)
+        import os as _os
+
+        clipped = [p[-4 * prompt_budget:] for p in prompts]
+        prefix_txt = _os.path.commonprefix(clipped) if B > 1 else ""
+        # keep at least one suffix token per prompt
+        P = min(len(prefix_txt), prompt_budget - 8,
+                min((len(c) for c in clipped), default=0) - 1)
+        use_prefix = (self.device != "cpu" and P >= self._PREFIX_MIN_TOKENS
+                      and _os.environ.get("SENTIO_PREFIX_KV", "1") != "0")
         sess = self._decode_session(B, self.max_seq)
         _t0 = _time.perf_counter()
-        logits = sess.prefill(tokens)
+        if use_prefix:
+            prefix_ids = self.tokenizer.encode(prefix_txt[:P], P)
+            P = len(prefix_ids)
+            suffixes = [p[len(prefix_txt[:P]):] for p in clipped]
+            padded, lens = self.tokenizer.encode_batch(
+                suffixes, prompt_budget - P)
+            tokens = torch.tensor(padded, dtype=torch.int64,
+                                  device=self.device)
+            sess.cache.seq_lens.zero_()
+            logits = sess.prefill_with_prefix(prefix_ids, tokens)
+            S = P + tokens.shape[1]
+        else:
+            padded, lens = self.tokenizer.encode_batch(clipped, prompt_budget)
+            tokens = torch.tensor(padded, dtype=torch.int64,
+                                  device=self.device)
+            logits = sess.prefill(tokens)
+            S = tokens.shape[1]
         if self.device != "cpu":
             torch.cuda.synchronize()
         self.last_prefill_s = _time.perf_counter() - _t0

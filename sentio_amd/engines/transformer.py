@@ -150,7 +150,7 @@ class Transformer:
     # ----- core blocks -----
     def _attn(self, x: torch.Tensor, layer: dict, pos: torch.Tensor,
               cache: KVCache | None, layer_idx: int,
-              kv_lens: torch.Tensor | None) -> torch.Tensor:
+              kv_lens: torch.Tensor | None, q_off: int = 0) -> torch.Tensor:
         cfg = self.cfg
         B, S, d = x.shape
         hd = cfg.head_dim
@@ -181,6 +181,14 @@ class Transformer:
                 q.view(B, H, hd), cache.k[layer_idx],
                 cache.v[layer_idx], seq_lens, self.scale,
             ).view(B, 1, H, hd)
+        elif cache is not None and q_off > 0:
+            # prefix-KV-cached suffix prefill: attend to cache rows
+            # [0, q_off + S) — the prefix KV was computed once and copied in
+            lens_abs = torch.full((B,), q_off + S, dtype=torch.int32,
+                                  device=x.device)
+            out = ops.attention_cache(q, cache.k[layer_idx],
+                                      cache.v[layer_idx], lens_abs,
+                                      q_off, self.scale)
         else:
             out = ops.attention(q, k, v, causal=cfg.causal, scale=self.scale,
                                 kv_lens=kv_lens)
@@ -209,6 +217,7 @@ class Transformer:
     def forward_hidden(
         self, tokens: torch.Tensor, pos: torch.Tensor | None = None,
         cache: KVCache | None = None, kv_lens: torch.Tensor | None = None,
+        q_off: int = 0,
     ) -> torch.Tensor:
         """tokens: [B, S] int64 → hidden [B, S, dim] (after final norm).
         kv_lens: right-padding valid lengths for non-causal batches.
@@ -223,7 +232,7 @@ class Transformer:
         n = len(layers)
         normed = ops.rmsnorm(x, layers[0]["attn_norm"], eps)
         for i, layer in enumerate(layers):
-            a = self._attn(normed, layer, pos, cache, i, kv_lens)
+            a = self._attn(normed, layer, pos, cache, i, kv_lens, q_off=q_off)
             normed, x = ops.rmsnorm_residual(a, x, layer["ffn_norm"], eps)
             f = self._ffn(normed, layer)
             w_next = layers[i + 1]["attn_norm"] if i + 1 < n else self.w.final_norm
@@ -309,6 +318,22 @@ class Transformer:
         if self.device != "cpu":
             self.decode_attn_fn = self._decode_attn_impl(
                 S / max(cache.max_seq, 1))
+        return self.logits(hidden)
+
+    def prefill_suffix(self, suffix_tokens: torch.Tensor, cache: KVCache,
+                       prefix_len: int) -> torch.Tensor:
+        """Prefill only the suffix against a cache whose rows [0, prefix_len)
+        already hold the shared prefix's KV (prefix-KV caching).  Returns
+        last-position logits."""
+        B, S = suffix_tokens.shape
+        pos = (prefix_len + torch.arange(S, device=suffix_tokens.device)
+               ).unsqueeze(0).expand(B, S)
+        hidden = self.forward_hidden(suffix_tokens, pos=pos, cache=cache,
+                                     q_off=prefix_len)
+        cache.seq_lens[:] = prefix_len + S
+        if self.device != "cpu":
+            self.decode_attn_fn = self._decode_attn_impl(
+                (prefix_len + S) / max(cache.max_seq, 1))
         return self.logits(hidden)
 
     def decode_step(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:

@@ -123,6 +123,22 @@ def attention(q, k, v, causal: bool = True, scale: float | None = None,
     return torch_ref.attention(q, k, v, causal, scale, kv_lens)
 
 
+def attention_cache(q, k_cache, v_cache, kv_lens, q_off: int,
+                    scale: float | None = None):
+    """Causal prefill attention for a SUFFIX of queries against the KV
+    cache (prefix-KV caching): q [B, S_suf, H, D] at absolute positions
+    q_off..q_off+S_suf-1; caches [B, Hkv, Smax, D]; kv_lens [B] absolute
+    valid lengths (prefix + suffix)."""
+    import math
+
+    s = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return _require_hip().flash_attn_cache(
+            q.contiguous(), k_cache, v_cache,
+            kv_lens.to(torch.int32).contiguous(), float(s), int(q_off))
+    return torch_ref.attention_cache(q, k_cache, v_cache, kv_lens, q_off, s)
+
+
 def decode_attention(q, k_cache, v_cache, seq_lens, scale: float | None = None):
     if _on_gpu(q):
         import math
@@ -292,6 +308,6 @@ def gemm_bf16(a, b):
 
 __all__ = [
     "rmsnorm", "rmsnorm_residual", "rope_apply", "decode_qkv_prep", "swiglu", "swiglu_packed", "softmax",
-    "attention", "decode_attention", "decode_attention_bmm", "mean_pool_l2norm", "cosine_topk",
+    "attention", "attention_cache", "decode_attention", "decode_attention_bmm", "mean_pool_l2norm", "cosine_topk",
     "bm25_score", "cosine_scores", "fuse_topk", "lt_linear", "sample_token", "gemm_bf16", "skinny_gemm", "hip_available", "torch_ref",
 ]

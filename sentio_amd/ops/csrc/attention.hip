@@ -58,15 +58,20 @@ __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
 // barrier/staging pass serves QT x 16 rows — per-row overhead halves at
 // QT=2 vs the one-tile version.
 #define FA_QT 1   // measured: QT=2 amortizes fragments but costs occupancy (124 vs 156 TF/s) — QT=1 wins
-template <int DT>
+// CACHE_SRC=false: K/V are activation tensors [B, S, Hkv, D] (plain
+// prefill).  CACHE_SRC=true: K/V are KV-cache tensors [B, Hkv, Smax, D]
+// and the queries are a SUFFIX starting at absolute position q_off —
+// prefix-KV-cached prefill attends to cache rows [0, kv_lens[b]).
+template <int DT, bool CACHE_SRC>
 __launch_bounds__(256)
 __global__ void flash_attn_kernel(
-    const bf16* __restrict__ q,    // [B, S, H, D]
-    const bf16* __restrict__ k,    // [B, S, Hkv, D]
-    const bf16* __restrict__ v,    // [B, S, Hkv, D]
+    const bf16* __restrict__ q,    // [B, S, H, D] (S = suffix len if CACHE_SRC)
+    const bf16* __restrict__ k,
+    const bf16* __restrict__ v,
     bf16* __restrict__ out,        // [B, S, H, D]
-    const int* __restrict__ kv_lens,  // [B]
-    int B, int S, int H, int Hkv, int D_, float scale, int causal) {
+    const int* __restrict__ kv_lens,  // [B] (absolute KV lengths)
+    int B, int S, int H, int Hkv, int D_, float scale, int causal,
+    int Smax, int q_off) {
   constexpr int D = DT;            // compile-time: every staging/frag loop
                                    // unrolls, loads batch before waits
   constexpr int QT = FA_QT;
@@ -84,7 +89,7 @@ __global__ void flash_attn_kernel(
   const int b = blockIdx.z;
   const int hkv = h / (H / Hkv);
   const int q0 = (blockIdx.x * FA_WAVES + wid) * WROWS;
-  const int kvlen = min(kv_lens[b], S);
+  const int kvlen = CACHE_SRC ? min(kv_lens[b], Smax) : min(kv_lens[b], S);
   const bool active = q0 < S;           // inactive waves still hit barriers
 
   constexpr int DC = D / 32;            // feature chunks per mfma K-dim
@@ -124,7 +129,7 @@ __global__ void flash_attn_kernel(
   // block-level kv bound: the LAST wave's causal horizon
   const int block_q_hi = min(blockIdx.x * FA_WAVES * WROWS + FA_WAVES * WROWS,
                              S);
-  const int kv_hi = causal ? min(kvlen, block_q_hi) : kvlen;
+  const int kv_hi = causal ? min(kvlen, q_off + block_q_hi) : kvlen;
   const int n_kv_tiles = (kv_hi + KVBLK - 1) / KVBLK;
 
   // T14 software-pipelined staging (guide §6: attention staging →
@@ -177,7 +182,7 @@ __global__ void flash_attn_kernel(
     // waves whose causal horizon ends before this kv tile skip compute but
     // still execute every barrier (uniform control flow)
     const int wave_q_hi = min(q0 + WROWS - 1, S - 1);
-    const bool compute = active && (!causal || kv0 <= wave_q_hi);
+    const bool compute = active && (!causal || kv0 <= q_off + wave_q_hi);
 
     constexpr int HALVES = KVBLK / 16;
     float alpha[QT][4];
@@ -215,7 +220,7 @@ __global__ void flash_attn_kernel(
 #pragma unroll
           for (int half = 0; half < HALVES; ++half) {
             const int key = kv0 + half * 16 + (lane & 15);
-            const bool ok = key < kvlen && (!causal || key <= qrow);
+            const bool ok = key < kvlen && (!causal || key <= q_off + qrow);
             sv[half] = ok ? s_acc[t][half][r] * scale : -INFINITY;
             rmax_l = fmaxf(rmax_l, sv[half]);
           }
@@ -575,10 +580,10 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
   dim3 grid((S + 4 * wave_rows - 1) / (4 * wave_rows), H, B);
 #define FA_CASE(DV)                                                          \
   case DV:                                                                   \
-    hipLaunchKernelGGL((flash_attn_kernel<DV>), grid, dim3(256), lds,        \
+    hipLaunchKernelGGL((flash_attn_kernel<DV, false>), grid, dim3(256), lds, \
                        stream, (const bf16*)q, (const bf16*)k,               \
                        (const bf16*)v, (bf16*)out, kv_lens, B, S, H, Hkv, D, \
-                       scale, causal);                                       \
+                       scale, causal, 0, 0);                                 \
     break;
   switch (D) {
     FA_CASE(32) FA_CASE(64) FA_CASE(96) FA_CASE(128)
@@ -586,6 +591,31 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
       return hipErrorInvalidValue;
   }
 #undef FA_CASE
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+hipError_t sentio_flash_attn_cache(const void* q, const void* kc,
+                                   const void* vc, void* out,
+                                   const int* kv_lens, int B, int S, int H,
+                                   int Hkv, int Smax, int D, float scale,
+                                   int q_off, hipStream_t stream) {
+  size_t lds = (size_t)KVBLK * D * 2 * 2 + 4 * FA_QT * QBLK * P_STRIDE;
+  const int wave_rows = FA_QT * QBLK;
+  dim3 grid((S + 4 * wave_rows - 1) / (4 * wave_rows), H, B);
+#define FAC_CASE(DV)                                                         \
+  case DV:                                                                   \
+    hipLaunchKernelGGL((flash_attn_kernel<DV, true>), grid, dim3(256), lds,  \
+                       stream, (const bf16*)q, (const bf16*)kc,              \
+                       (const bf16*)vc, (bf16*)out, kv_lens, B, S, H, Hkv,   \
+                       D, scale, 1, Smax, q_off);                            \
+    break;
+  switch (D) {
+    FAC_CASE(32) FAC_CASE(64) FAC_CASE(96) FAC_CASE(128)
+    default:
+      return hipErrorInvalidValue;
+  }
+#undef FAC_CASE
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

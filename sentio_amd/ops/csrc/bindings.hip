@@ -36,6 +36,9 @@ hipError_t sentio_flash_attn(const void*, const void*, const void*, void*,
 hipError_t sentio_decode_attn(const void*, const void*, const void*, void*,
                               const int*, float*, float*, int, int, int, int,
                               int, int, float, hipStream_t);
+hipError_t sentio_flash_attn_cache(const void*, const void*, const void*,
+                                   void*, const int*, int, int, int, int,
+                                   int, int, float, int, hipStream_t);
 hipError_t sentio_gemm_bf16(const void*, const void*, void*, int, int, int,
                             hipStream_t);
 hipError_t sentio_skinny_gemm(const void*, const void*, void*, int, int, int,
@@ -334,6 +337,24 @@ torch::Tensor lt_gemm_tn(torch::Tensor x, torch::Tensor w) {
   return out;
 }
 
+torch::Tensor flash_attn_cache(torch::Tensor q, torch::Tensor kc,
+                               torch::Tensor vc, torch::Tensor kv_lens,
+                               double scale, int64_t q_off) {
+  check_bf16_cuda(q, "q");
+  TORCH_CHECK(q.dim() == 4, "q must be [B,S,H,D] (suffix)");
+  TORCH_CHECK(kc.dim() == 4 && vc.dim() == 4, "caches must be [B,Hkv,Smax,D]");
+  TORCH_CHECK(kv_lens.scalar_type() == torch::kInt, "kv_lens must be i32");
+  const int B = q.size(0), S = q.size(1), H = q.size(2), D = q.size(3);
+  const int Hkv = kc.size(1), Smax = kc.size(2);
+  auto out = torch::empty_like(q);
+  check_hip(sentio_flash_attn_cache(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                                    out.data_ptr(),
+                                    kv_lens.contiguous().data_ptr<int>(), B, S,
+                                    H, Hkv, Smax, D, (float)scale, (int)q_off,
+                                    stream()), "flash_attn_cache");
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("rmsnorm_residual", &rmsnorm_residual);
@@ -347,6 +368,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cosine_scores", &cosine_scores);
   m.def("bm25_score", &bm25_score);
   m.def("flash_attn", &flash_attn);
+  m.def("flash_attn_cache", &flash_attn_cache);
   m.def("decode_attn", &decode_attn);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("skinny_gemm", &skinny_gemm);

@@ -217,3 +217,29 @@ def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
     """gu [..., 2F] rows packed [gate | up] → silu(gate) * up, [..., F]."""
     f = gu.shape[-1] // 2
     return swiglu(gu[..., :f], gu[..., f:])
+
+
+def attention_cache(
+    q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+    kv_lens: torch.Tensor, q_off: int, scale: float | None = None,
+) -> torch.Tensor:
+    """Suffix-against-cache causal attention (prefix-KV-cached prefill).
+    q: [B, S_suf, H, D] at absolute positions q_off + i;
+    k_cache/v_cache: [B, Hkv, Smax, D]; kv_lens: [B] absolute lengths."""
+    B, S, H, D = q.shape
+    Hkv = k_cache.shape[1]
+    rep = H // Hkv
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    out = torch.empty_like(q, dtype=torch.float32)
+    for b in range(B):
+        L = int(kv_lens[b])
+        kf = k_cache[b, :, :L].float().repeat_interleave(rep, dim=0)  # [H,L,D]
+        vf = v_cache[b, :, :L].float().repeat_interleave(rep, dim=0)
+        qf = q[b].float().permute(1, 0, 2)                            # [H,S,D]
+        sc = (qf @ kf.transpose(-1, -2)) * scale                      # [H,S,L]
+        pos = q_off + torch.arange(S).view(1, S, 1)
+        key = torch.arange(L).view(1, 1, L)
+        sc = sc.masked_fill(key > pos, float("-inf"))
+        p = torch.softmax(sc, dim=-1)
+        out[b] = (p @ vf).permute(1, 0, 2)
+    return out.to(q.dtype)

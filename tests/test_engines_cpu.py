@@ -82,3 +82,37 @@ def test_encoder_embedding_cache():
     import torch
     assert torch.allclose(v1[0], v2[0], atol=1e-6)
     assert torch.allclose(v1[1], v2[2], atol=1e-6)
+
+
+def test_prefix_suffix_prefill_matches_full():
+    """prefill_suffix over a prefix-primed cache must reproduce the plain
+    full-prompt prefill logits (CPU, fp32)."""
+    import torch
+
+    from sentio_amd.engines.configs import MODEL_CONFIGS
+    from sentio_amd.engines.transformer import KVCache, Transformer
+
+    cfg = MODEL_CONFIGS["tiny-decoder64"]
+    m = Transformer(cfg, device="cpu", seed=21)
+    B, P, S = 2, 10, 7
+    torch.manual_seed(3)
+    full = torch.randint(0, cfg.vocab_size, (B, P + S))
+    full[:, :P] = full[0, :P]          # shared prefix
+
+    c_full = KVCache(cfg, B, 64, "cpu", m.dtype)
+    want = m.prefill(full, c_full)
+
+    # prefix KV computed batch-1, broadcast into a fresh cache
+    c_pre = KVCache(cfg, 1, P, "cpu", m.dtype)
+    m.forward_hidden(full[:1, :P], cache=c_pre)
+    c_suf = KVCache(cfg, B, 64, "cpu", m.dtype)
+    for i in range(cfg.n_layers):
+        c_suf.k[i][:, :, :P] = c_pre.k[i]
+        c_suf.v[i][:, :, :P] = c_pre.v[i]
+    got = m.prefill_suffix(full[:, P:], c_suf, P)
+    assert torch.allclose(got, want, rtol=1e-4, atol=1e-4)
+    # decode continues identically from both caches
+    nxt = torch.randint(0, cfg.vocab_size, (B, 1))
+    d1 = m.decode_step(nxt, c_full)
+    d2 = m.decode_step(nxt, c_suf)
+    assert torch.allclose(d1, d2, rtol=1e-4, atol=1e-4)
