@@ -149,7 +149,9 @@ __global__ void flash_attn_kernel(
       const int key = kv0 + i / D;
       const int d = i % D;
       if (i < ST_ELEMS && key < kvlen) {
-        const long base = (((long)b * S + key) * Hkv + hkv) * D + d;
+        const long base = CACHE_SRC
+            ? (((long)b * Hkv + hkv) * Smax + key) * D + d
+            : (((long)b * S + key) * Hkv + hkv) * D + d;
         kvals[u] = *reinterpret_cast<const bf16x8_t*>(k + base);
         vvals[u] = *reinterpret_cast<const bf16x8_t*>(v + base);
       } else {
