@@ -70,7 +70,10 @@ class _DecodeSession:
         torch.cuda.current_stream().wait_stream(s)
         self.cache.seq_lens.copy_(saved)
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        # thread_local capture: concurrent engine work on OTHER threads
+        # (pipelined retrieval, encoder graph replays) must not invalidate
+        # this capture — global mode aborts on any foreign hipMalloc
+        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
             self.static_logits = model.decode_step(self.static_tok, self.cache)
         self.cache.seq_lens.copy_(saved)
 
