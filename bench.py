@@ -422,6 +422,13 @@ def main():
             if record:
                 step_times.append(time.perf_counter() - ts)
 
+    # pre-capture the decode hipGraph with NO concurrent retrieval thread:
+    # graph capture and foreign allocator traffic race otherwise
+    generator.generate(["capture warm-up prompt"] * args.batch,
+                       max_new_tokens=2, temperature=0.0, stop_on_eos=False)
+    if on_gpu:
+        torch.cuda.synchronize()
+
     # ---- warmup ----
     run_pipelined(0, args.warmup)
     D.barrier()

@@ -70,10 +70,12 @@ class _DecodeSession:
         torch.cuda.current_stream().wait_stream(s)
         self.cache.seq_lens.copy_(saved)
         self.graph = torch.cuda.CUDAGraph()
-        # thread_local capture: concurrent engine work on OTHER threads
-        # (pipelined retrieval, encoder graph replays) must not invalidate
-        # this capture — global mode aborts on any foreign hipMalloc
-        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
+        # relaxed capture: concurrent engine work on OTHER threads
+        # (pipelined retrieval, encoder graph replays, allocator traffic)
+        # must not invalidate this capture — global/thread_local modes abort
+        # on foreign allocator calls; the captured region itself only
+        # contains this stream's decode ops
+        with torch.cuda.graph(self.graph, capture_error_mode="relaxed"):
             self.static_logits = model.decode_step(self.static_tok, self.cache)
         self.cache.seq_lens.copy_(saved)
 

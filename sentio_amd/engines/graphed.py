@@ -55,9 +55,9 @@ class GraphedForward:
                 self.fn(self.tokens, self.kv_lens)
         torch.cuda.current_stream().wait_stream(s)
         self.graph = torch.cuda.CUDAGraph()
-        # thread_local: see generator._capture — cross-thread engine work
+        # relaxed: see generator._capture — cross-thread engine work
         # (decode on the main thread) must not invalidate this capture
-        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
+        with torch.cuda.graph(self.graph, capture_error_mode="relaxed"):
             self.out = self.fn(self.tokens, self.kv_lens)
 
     def run(self, tokens: torch.Tensor, kv_lens: torch.Tensor) -> torch.Tensor:
