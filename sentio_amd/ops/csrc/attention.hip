@@ -392,6 +392,23 @@ __global__ void decode_attn_split_kernel(
     // One predicated unrolled path for EVERY chunk (a separate dynamic tail
     // loop serialized one load per s_waitcnt and dominated short spans):
     // 8 loads in flight before any ds_write; out-of-range rows stage zeros.
+#ifdef SENTIO_DECODE_GLDS
+    {
+      const int lim = chunk * D;
+#pragma unroll
+      for (int u = 0; u < D / 8; ++u) {
+        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
+        if (i < lim) {
+          const int row = i / D, d = i % D;
+          auto gp = (const __attribute__((address_space(1))) unsigned int*)(
+              kb + (long)(s0 + row) * D + d);
+          auto lp = (__attribute__((address_space(3))) unsigned int*)(
+              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
+          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
+        }
+      }
+    }
+#else
     {
       constexpr int IT = D / 8;          // loads per thread for a full chunk
       constexpr int BATCH = 8;           // loads in flight
@@ -420,6 +437,7 @@ __global__ void decode_attn_split_kernel(
         }
       }
     }
+#endif
     __syncthreads();
     // phase A2: thread-per-key dot vs all G query heads, K read from LDS;
     // scores stay in registers (the softmax below is the same thread).

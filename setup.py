@@ -42,6 +42,9 @@ def build(verbose: bool = True) -> Path:
     # SENTIO_SANITIZE=1: host-side ASan+UBSan test build (SURVEY §5 race/
     # sanitizer row).  Run with LD_PRELOAD of the ASan runtime, e.g.
     #   LD_PRELOAD=$(hipcc -print-file-name=libclang_rt.asan-x86_64.so) pytest
+    if os.environ.get("SENTIO_DECODE_GLDS") == "1":
+        # experimental LDS-DMA K staging in decode attention
+        common += ["-DSENTIO_DECODE_GLDS"]
     if os.environ.get("SENTIO_SANITIZE") == "1":
         common += ["-fsanitize=address,undefined",
                    "-fno-omit-frame-pointer", "-shared-libasan"]
