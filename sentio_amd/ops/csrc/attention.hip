@@ -392,7 +392,10 @@ __global__ void decode_attn_split_kernel(
     // One predicated unrolled path for EVERY chunk (a separate dynamic tail
     // loop serialized one load per s_waitcnt and dominated short spans):
     // 8 loads in flight before any ds_write; out-of-range rows stage zeros.
-#ifdef SENTIO_DECODE_GLDS
+#ifdef SENTIO_DECODE_GLDS   // EXPERIMENTAL, known-broken numerics (kept for
+                             // study): raw global_load_lds staging miscompiles
+                             // or misdrains here — 3 GPU tests fail, no perf
+                             // gain measured; default register path below
     {
       const int lim = chunk * D;
 #pragma unroll
