@@ -39,7 +39,7 @@ LtState& state() {
   static std::once_flag once;
   std::call_once(once, [] {
     hipblasLtCreate(&s.handle);
-    hipMalloc(&s.workspace, kWorkspaceBytes);
+    (void)hipMalloc(&s.workspace, kWorkspaceBytes);
   });
   return s;
 }
@@ -96,8 +96,8 @@ bool autotune(LtState& s, Descs& d, int M, int N, int K, const void* x,
 
   const float alpha = 1.f, beta = 0.f;
   hipEvent_t ev0, ev1;
-  hipEventCreate(&ev0);
-  hipEventCreate(&ev1);
+  (void)hipEventCreate(&ev0);
+  (void)hipEventCreate(&ev1);
   float best_ms = 1e30f;
   int best_i = -1;
   for (int i = 0; i < n_results; ++i) {
@@ -107,22 +107,22 @@ bool autotune(LtState& s, Descs& d, int M, int N, int K, const void* x,
                         d.c, out, d.c, &results[i].algo, s.workspace,
                         kWorkspaceBytes, stream) != HIPBLAS_STATUS_SUCCESS)
       continue;
-    hipEventRecord(ev0, stream);
+    (void)hipEventRecord(ev0, stream);
     for (int it = 0; it < kTimingIters; ++it)
       hipblasLtMatmul(s.handle, d.op, &alpha, w, d.a, x, d.b, &beta, out, d.c,
                       out, d.c, &results[i].algo, s.workspace, kWorkspaceBytes,
                       stream);
-    hipEventRecord(ev1, stream);
-    hipEventSynchronize(ev1);
+    (void)hipEventRecord(ev1, stream);
+    (void)hipEventSynchronize(ev1);
     float ms = 0.f;
-    hipEventElapsedTime(&ms, ev0, ev1);
+    (void)hipEventElapsedTime(&ms, ev0, ev1);
     if (ms < best_ms) {
       best_ms = ms;
       best_i = i;
     }
   }
-  hipEventDestroy(ev0);
-  hipEventDestroy(ev1);
+  (void)hipEventDestroy(ev0);
+  (void)hipEventDestroy(ev1);
   if (best_i < 0) return false;
   *best_out = results[best_i].algo;
   return true;
