@@ -293,8 +293,9 @@ def lt_linear(x, w):
             and os.environ.get("SENTIO_LT_GEMM", "1") != "0"):
         try:
             return _require_hip().lt_gemm_tn(x.contiguous(), w.contiguous())
-        except RuntimeError:
-            _lt_ok = False
+        except RuntimeError as exc:
+            if "rc=5" not in str(exc):   # rc=5: unseen shape mid-capture —
+                _lt_ok = False           # fall back this call only
     return torch.nn.functional.linear(x, w)
 
 

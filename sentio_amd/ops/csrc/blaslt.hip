@@ -152,6 +152,12 @@ int sentio_lt_gemm_tn(const void* x, const void* w, void* out, int M, int N,
     }
   }
   if (!have) {
+    // autotuning syncs the stream — illegal during hipGraph capture.  The
+    // engine warms every decode shape eagerly before capturing; an unseen
+    // shape mid-capture falls back to torch (rc=5) instead of crashing.
+    hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+    (void)hipStreamIsCapturing(stream, &cap);
+    if (cap != hipStreamCaptureStatusNone) return 5;
     if (!autotune(s, d, M, N, K, x, w, out, stream, &algo)) return 3;
     std::lock_guard<std::mutex> lk(s.mu);
     s.algo_cache[{M, N, K}] = algo;
