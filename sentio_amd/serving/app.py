@@ -122,10 +122,53 @@ def create_app(settings: Settings | None = None,
                   lifespan=lifespan)
     app.state.container = container
 
+    # scope requirements when auth is enabled (reference auth.py:444-470
+    # require_scopes dependency guards; DISABLE_AUTH=true skips, as the
+    # reference's CORS-open dev mode did)
+    from sentio_amd.utils.auth import AuthError, AuthScope
+
+    _SCOPE_MAP = {
+        ("POST", "/chat"): AuthScope.CHAT,
+        ("POST", "/chat/stream"): AuthScope.CHAT,
+        ("POST", "/embed"): AuthScope.EMBED,
+        ("POST", "/clear"): AuthScope.DELETE,
+        ("GET", "/metrics"): AuthScope.METRICS,
+        ("GET", "/metrics/performance"): AuthScope.METRICS,
+    }
+
+    def _check_auth(request: Request) -> JSONResponse | None:
+        scope = _SCOPE_MAP.get((request.method, request.url.path))
+        if scope is None:
+            return None                      # health/info/ui stay open
+        header = request.headers.get("Authorization", "")
+        mgr = container.auth_manager()
+        try:
+            if header.startswith("Bearer "):
+                mgr.require_scopes(header[7:].strip(), scope)
+                return None
+            api_key = request.headers.get("X-API-Key", "")
+            if api_key:
+                from sentio_amd.utils.auth import ROLE_SCOPES
+
+                role = mgr.verify_api_key(api_key)
+                if scope in ROLE_SCOPES[role]:
+                    return None
+                raise AuthError(f"api key lacks scope {scope.value}")
+            raise AuthError("missing credentials")
+        except AuthError as exc:
+            return JSONResponse(status_code=401,
+                                content={"error": "AUTH_ERROR",
+                                         "message": str(exc)})
+
     @app.middleware("http")
     async def security_and_rate_limit(request: Request, call_next):
         client = request.client.host if request.client else "unknown"
         path = request.url.path
+        if s.auth_enabled:
+            denied = _check_auth(request)
+            if denied is not None:
+                SecurityHeaders.apply(denied)
+                return denied
         if request.method == "POST" and not limiter.allow(client, path):
             return JSONResponse(
                 status_code=429,

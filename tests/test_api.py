@@ -201,3 +201,46 @@ def test_tracing_spans_recorded_on_chat(client):
     names = [s["name"] for s in tracing.recent_spans()]
     assert "pipeline.invoke" in names
     assert any(n.startswith("stage.") for n in names)
+
+
+def test_auth_enforced_when_enabled():
+    """DISABLE_AUTH=false: scoped endpoints demand Bearer tokens / API keys
+    (reference auth.py:444-470 require_scopes guards)."""
+    from fastapi.testclient import TestClient
+
+    from sentio_amd.config import Settings
+    from sentio_amd.serving.app import create_app
+    from sentio_amd.serving.container import ServiceContainer
+    from sentio_amd.utils.auth import UserRole
+
+    s = Settings()
+    s.mock_compute = True
+    s.device = "cpu"
+    s.disable_auth = False
+    container = ServiceContainer(s)
+    app = create_app(settings=s, container=container)
+    with TestClient(app) as c:
+        # no credentials → 401
+        r = c.post("/chat", json={"question": "who are you?"})
+        assert r.status_code == 401
+        assert c.get("/metrics").status_code == 401
+        # health stays open
+        assert c.get("/health").status_code == 200
+
+        mgr = container.auth_manager()
+        reader = mgr.issue_token("alice", role=UserRole.READER)
+        writer = mgr.issue_token("bob", role=UserRole.WRITER)
+        # reader can chat but not embed
+        r = c.post("/chat", json={"question": "who are you?"},
+                   headers={"Authorization": f"Bearer {reader}"})
+        assert r.status_code == 200
+        r = c.post("/embed", json={"content": "doc body " * 10},
+                   headers={"Authorization": f"Bearer {reader}"})
+        assert r.status_code == 401
+        r = c.post("/embed", json={"content": "doc body " * 10},
+                   headers={"Authorization": f"Bearer {writer}"})
+        assert r.status_code == 200
+        # API key path
+        key = mgr.create_api_key(role=UserRole.ADMIN)
+        assert c.post("/clear", headers={"X-API-Key": key}).status_code == 200
+        assert c.post("/clear", headers={"X-API-Key": "sk-bogus"}).status_code == 401
