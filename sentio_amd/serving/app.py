@@ -122,6 +122,13 @@ def create_app(settings: Settings | None = None,
                   lifespan=lifespan)
     app.state.container = container
 
+    # CORS open only in no-auth dev mode (reference app.py:370-377)
+    if not s.auth_enabled:
+        from fastapi.middleware.cors import CORSMiddleware
+
+        app.add_middleware(CORSMiddleware, allow_origins=["*"],
+                           allow_methods=["*"], allow_headers=["*"])
+
     # scope requirements when auth is enabled (reference auth.py:444-470
     # require_scopes dependency guards; DISABLE_AUTH=true skips, as the
     # reference's CORS-open dev mode did)

@@ -244,3 +244,11 @@ def test_auth_enforced_when_enabled():
         key = mgr.create_api_key(role=UserRole.ADMIN)
         assert c.post("/clear", headers={"X-API-Key": key}).status_code == 200
         assert c.post("/clear", headers={"X-API-Key": "sk-bogus"}).status_code == 401
+
+
+def test_cors_open_when_auth_disabled(client):
+    r = client.options("/chat", headers={
+        "Origin": "http://example.com",
+        "Access-Control-Request-Method": "POST",
+    })
+    assert r.headers.get("access-control-allow-origin") == "*"
