@@ -253,6 +253,15 @@ class GeneratorEngine:
             torch.cuda.synchronize()
         self.last_decode_s = _time.perf_counter() - _t0
         self.last_decode_steps = n_steps
+        try:
+            from sentio_amd.observability.metrics import metrics_collector
+
+            metrics_collector.inc("rag_llm_tokens_total",
+                                  float(B * S), kind="prompt")
+            metrics_collector.inc("rag_llm_tokens_total",
+                                  float(B * n_steps), kind="completion")
+        except Exception:
+            pass
         rows = toks_buf[:, :n_steps].cpu().tolist()
         out = []
         for row in rows:

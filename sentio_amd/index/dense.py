@@ -84,14 +84,19 @@ class DenseIndex:
 
     # ----- search -----
     def search(
-        self, query: torch.Tensor, top_k: int
+        self, query: torch.Tensor, top_k: int,
+        metadata_filter: dict | None = None,
     ) -> list[list[tuple[str, float]]]:
-        """Batched cosine top-k.  query: [B, dim] or [dim]."""
+        """Batched cosine top-k.  query: [B, dim] or [dim].
+        metadata_filter: equality conditions on document metadata (the
+        reference's Qdrant FieldCondition filters, qdrant_store.py:456-471)
+        applied as a post-filter over an over-fetched candidate set."""
         if query.ndim == 1:
             query = query.unsqueeze(0)
         if self._size == 0:
             return [[] for _ in range(query.shape[0])]
-        k = min(top_k, self._size)
+        fetch_k = min(top_k * 4, self._size) if metadata_filter else top_k
+        k = min(fetch_k, self._size)
         q = query.to(self.device, torch.float32)
         q = q / q.norm(dim=1, keepdim=True).clamp_min(1e-12)
 
@@ -106,9 +111,15 @@ class DenseIndex:
         idx_l = idx.cpu().tolist()
         out = []
         for bi in range(len(vals_l)):
-            out.append([
-                (self.doc_ids[i], float(v)) for v, i in zip(vals_l[bi], idx_l[bi])
-            ])
+            hits = [(self.doc_ids[i], float(v))
+                    for v, i in zip(vals_l[bi], idx_l[bi])]
+            if metadata_filter:
+                hits = [
+                    (doc_id, v) for doc_id, v in hits
+                    if all(self._docs[doc_id].metadata.get(key) == val
+                           for key, val in metadata_filter.items())
+                ][:top_k]
+            out.append(hits)
         return out
 
     def get_document(self, doc_id: str) -> Document | None:

@@ -104,3 +104,20 @@ def test_dense_index_save_load(tmp_path):
     q = enc.embed([docs[2].text])
     got = loaded.search(q, 1)[0][0][0]
     assert got == "d2"
+
+
+def test_dense_search_metadata_filter():
+    import torch
+
+    from sentio_amd.index.dense import DenseIndex
+    from sentio_amd.models.document import Document
+
+    idx = DenseIndex(dim=8, device="cpu")
+    docs = [Document(text=f"d{i}", id=f"d{i}",
+                     metadata={"lang": "en" if i % 2 == 0 else "de"})
+            for i in range(10)]
+    vecs = torch.randn(10, 8)
+    idx.add(docs, vecs)
+    hits = idx.search(vecs[1], top_k=5, metadata_filter={"lang": "de"})[0]
+    assert hits and all(h[0] in {"d1", "d3", "d5", "d7", "d9"} for h in hits)
+    assert hits[0][0] == "d1"   # self-match still ranks first among de docs
