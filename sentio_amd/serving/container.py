@@ -212,6 +212,13 @@ class ServiceContainer:
     def initialize_all(self) -> None:
         """Eager startup init in dependency order."""
         self.encoder()
+        try:
+            # warm-up embed (reference warm_up_embeddings startup hook,
+            # embeddings/factory.py:122-137): triggers weight init + the
+            # first hipGraph capture before user traffic
+            self.encoder().embed(["warm up"])
+        except Exception as exc:
+            logger.warning("encoder warm-up failed: %s", exc)
         self.dense_index()
         self.bm25_index()
         self.retriever()
