@@ -281,7 +281,14 @@ class GeneratorEngine:
     def stream(self, prompt: str, max_new_tokens: int = 128,
                temperature: float = 0.3) -> Iterator[str]:
         """Yield text deltas token-by-token (single prompt) — the on-device
-        equivalent of the reference's SSE streaming (openai.py:149-157)."""
+        equivalent of the reference's SSE streaming (openai.py:149-157).
+        Holds the generation lock for the stream's duration (decode sessions
+        and the model's decode-attention selection are engine-level state)."""
+        with self._gen_lock:
+            yield from self._stream_locked(prompt, max_new_tokens, temperature)
+
+    def _stream_locked(self, prompt: str, max_new_tokens: int,
+                       temperature: float) -> Iterator[str]:
         prompt_budget = self.max_seq - max_new_tokens - 1
         ids = self.tokenizer.encode(prompt[-4 * prompt_budget:], prompt_budget)
         tokens = torch.tensor([ids], dtype=torch.int64, device=self.device)
