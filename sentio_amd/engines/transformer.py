@@ -155,7 +155,7 @@ class Transformer:
         B, S, d = x.shape
         hd = cfg.head_dim
         H, Hkv = self.h_local, self.hkv_local
-        qkv = ops.lt_linear(x.view(B * S, d), layer["wqkv"])
+        qkv = torch.nn.functional.linear(x.view(B * S, d), layer["wqkv"])
         qkv = qkv.view(B, S, -1)
         q_end = H * hd
         k_end = q_end + Hkv * hd
@@ -192,7 +192,7 @@ class Transformer:
         else:
             out = ops.attention(q, k, v, causal=cfg.causal, scale=self.scale,
                                 kv_lens=kv_lens)
-        out = ops.lt_linear(out.reshape(B * S, H * hd), layer["wo"])
+        out = torch.nn.functional.linear(out.reshape(B * S, H * hd), layer["wo"])
         out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
         return out.view(B, S, d)
 
@@ -203,9 +203,14 @@ class Transformer:
         # rows additionally go through the autotuned-algorithm binding
         rows = B * S
         xx = x.view(rows, d)
-        gu = ops.lt_linear(xx, layer["w_gate_up"])
-        y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
-        out = ops.lt_linear(y, layer["w_down"])
+        if rows <= 64 and self.device != "cpu":
+            gu = ops.lt_linear(xx, layer["w_gate_up"])
+            y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
+            out = ops.lt_linear(y, layer["w_down"])
+        else:
+            gu = torch.nn.functional.linear(xx, layer["w_gate_up"])
+            y = ops.swiglu_packed(gu)
+            out = torch.nn.functional.linear(y, layer["w_down"])
         out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
         return out.view(B, S, d)
 
@@ -298,7 +303,9 @@ class Transformer:
     def logits(self, hidden: torch.Tensor) -> torch.Tensor:
         B, S, d = hidden.shape
         last = hidden[:, -1, :]
-        return ops.lt_linear(last.contiguous(), self.w.lm_head).float()  # [B, V]
+        if B <= 64 and self.device != "cpu":
+            return ops.lt_linear(last.contiguous(), self.w.lm_head).float()
+        return torch.nn.functional.linear(last, self.w.lm_head).float()  # [B, V]
 
     def prefill(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
         """Prefill the cache; returns last-position logits [B, V].  Also
