@@ -283,9 +283,12 @@ _lt_ok = True
 
 def lt_linear(x, w):
     """F.linear through the autotuned hipBLASLt binding (per-shape algorithm
-    search on first use; ~35% faster than torch's heuristic pick on skinny
-    decode shapes).  Falls back to torch permanently on any failure.
-    Env SENTIO_LT_GEMM=0 disables."""
+    search on first use).  ONLY for shape-stable call sites (decode
+    projections — fixed batch): autotuning costs ~hundreds of ms per new
+    (M,N,K), so variable-M prefill GEMMs must stay on torch's heuristic
+    (routing them here measured 19→5.4 QPS from perpetual re-tuning).
+    Falls back to torch permanently on any failure; SENTIO_LT_GEMM=0
+    disables."""
     global _lt_ok
     import os
 
