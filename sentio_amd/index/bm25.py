@@ -102,15 +102,28 @@ class BM25Index:
     # ----- incremental add (used by /embed) -----
     def add(self, doc_ids: list[str], texts: list[str]) -> None:
         """Rebuild including new docs.  Postings arrays are immutable-by-design
-        so the on-device copy stays flat; ingest batches amortize the rebuild."""
+        so the on-device copy stays flat; ingest batches amortize the rebuild.
+        After a load() the token cache is reconstructed by inverting the CSR
+        postings (token order is lost — irrelevant to BM25)."""
         all_ids = self.doc_ids + list(doc_ids)
         old_tok = getattr(self, "_tokenized_cache", None)
         new_tok = [tokenize(t) for t in texts]
         if old_tok is None:
-            old_tok = [[] for _ in self.doc_ids]
+            old_tok = self._retokenize_from_postings()
         tokenized = old_tok + new_tok
         self._tokenized_cache = tokenized
         self.build_tokenized(all_ids, tokenized)
+
+    def _retokenize_from_postings(self) -> list[list[str]]:
+        out: list[list[str]] = [[] for _ in self.doc_ids]
+        if not self.doc_ids or len(self.post_doc) == 0:
+            return out
+        inv = {tid: tok for tok, tid in self.vocab.items()}
+        for t in range(len(self.vocab)):
+            for p in range(int(self.indptr[t]), int(self.indptr[t + 1])):
+                out[int(self.post_doc[p])].extend(
+                    [inv[t]] * int(self.post_tf[p]))
+        return out
 
     @property
     def n_docs(self) -> int:

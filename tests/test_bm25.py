@@ -92,3 +92,22 @@ def test_empty_query_and_empty_index():
     assert idx.search("anything", top_k=5) == []
     idx.build(["a"], ["some text"])
     assert idx.search("zzz unknown terms", top_k=5) == []
+
+
+def test_add_after_load_keeps_old_postings(tmp_path):
+    """Regression: adding documents after load() must not drop the loaded
+    corpus's postings (token cache reconstructed from CSR)."""
+    from sentio_amd.index.bm25 import BM25Index
+
+    idx = BM25Index()
+    idx.build(["a", "b"], ["gpu kernels stream data", "retrieval ranks docs"])
+    p = str(tmp_path / "bm.pkl")
+    idx.save(p)
+
+    loaded = BM25Index.load(p)
+    loaded.add(["c"], ["new doc about gpu retrieval"])
+    assert loaded.n_docs == 3
+    hits = dict(loaded.search("kernels", top_k=3))
+    assert "a" in hits                     # old posting survived
+    hits2 = [d for d, _ in loaded.search("retrieval", top_k=3)]
+    assert set(hits2) >= {"b", "c"}
