@@ -203,7 +203,7 @@ class GeneratorEngine:
         # keep at least one suffix token per prompt
         P = min(len(prefix_txt), prompt_budget - 8,
                 min((len(c) for c in clipped), default=0) - 1)
-        use_prefix = (self.device != "cpu" and P >= self._PREFIX_MIN_TOKENS
+        use_prefix = (P >= self._PREFIX_MIN_TOKENS
                       and _os.environ.get("SENTIO_PREFIX_KV", "1") != "0")
         sess = self._decode_session(B, self.max_seq)
         _t0 = _time.perf_counter()

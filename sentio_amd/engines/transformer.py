@@ -334,6 +334,8 @@ class Transformer:
         if self.device != "cpu":
             self.decode_attn_fn = self._decode_attn_impl(
                 (prefix_len + S) / max(cache.max_seq, 1))
+        else:
+            self.decode_attn_fn = None
         return self.logits(hidden)
 
     def decode_step(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:

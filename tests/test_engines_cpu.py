@@ -116,3 +116,21 @@ def test_prefix_suffix_prefill_matches_full():
     d1 = m.decode_step(nxt, c_full)
     d2 = m.decode_step(nxt, c_suf)
     assert torch.allclose(d1, d2, rtol=1e-4, atol=1e-4)
+
+
+def test_generate_with_prefix_kv_matches_disabled(monkeypatch):
+    """Full generate() path: identical greedy tokens with prefix caching on
+    and off (CPU fp32 — exact)."""
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
+    prefix = "system preamble shared across requests. " * 3
+    prompts = [prefix + f"question number {i}?" for i in range(3)]
+    monkeypatch.setenv("SENTIO_PREFIX_KV", "0")
+    base = eng.generate(prompts, max_new_tokens=8, temperature=0.0,
+                        stop_on_eos=False)
+    monkeypatch.setenv("SENTIO_PREFIX_KV", "1")
+    cached = eng.generate(prompts, max_new_tokens=8, temperature=0.0,
+                          stop_on_eos=False)
+    assert base == cached
+    assert eng._prefix_store  # the prefix KV was actually computed/cached
