@@ -112,6 +112,26 @@ class GeneratorEngine:
     # ----- prefix-KV cache (shared prompt prefixes prefilled once) -----
     _PREFIX_MIN_TOKENS = 64
 
+    def _split_shared_prefix(self, clipped: list[str], prompt_budget: int
+                             ) -> tuple[list[int], list[str]]:
+        """Split the batch's longest common prompt prefix from the per-prompt
+        suffixes such that prefix_ids + encode(suffix, add_bos=False) is
+        EXACTLY encode(full prompt) for every prompt (byte tokenizer:
+        chars align 1:1 with ids after the single BOS)."""
+        import os as _os
+
+        if len(clipped) < 2:
+            return [], clipped
+        prefix_txt = _os.path.commonprefix(clipped)
+        # leave >= 1 suffix char per prompt; keep room for generation
+        P_chars = min(len(prefix_txt), prompt_budget - 9,
+                      min(len(c) for c in clipped) - 1)
+        if P_chars <= 0:
+            return [], clipped
+        prefix_ids = self.tokenizer.encode(prefix_txt[:P_chars], None)
+        suffixes = [c[P_chars:] for c in clipped]
+        return prefix_ids, suffixes
+
     def _prefix_kv(self, prefix_ids: tuple):
         """Per-prefix KV, computed once with a batch-1 forward and kept in
         HBM (a few tens of MB per distinct prefix; LRU of 4)."""
@@ -199,20 +219,16 @@ class GeneratorEngine:
         import os as _os
 
         clipped = [p[-4 * prompt_budget:] for p in prompts]
-        prefix_txt = _os.path.commonprefix(clipped) if B > 1 else ""
-        # keep at least one suffix token per prompt
-        P = min(len(prefix_txt), prompt_budget - 8,
-                min((len(c) for c in clipped), default=0) - 1)
-        use_prefix = (P >= self._PREFIX_MIN_TOKENS
+        prefix_ids, suffixes = self._split_shared_prefix(clipped, prompt_budget)
+        use_prefix = (len(prefix_ids) >= self._PREFIX_MIN_TOKENS
                       and _os.environ.get("SENTIO_PREFIX_KV", "1") != "0")
         sess = self._decode_session(B, self.max_seq)
         _t0 = _time.perf_counter()
         if use_prefix:
-            prefix_ids = self.tokenizer.encode(prefix_txt[:P], P)
             P = len(prefix_ids)
-            suffixes = [p[len(prefix_txt[:P]):] for p in clipped]
+            # suffixes continue the stream: NO second BOS
             padded, lens = self.tokenizer.encode_batch(
-                suffixes, prompt_budget - P)
+                suffixes, prompt_budget - P, add_bos=False)
             tokens = torch.tensor(padded, dtype=torch.int64,
                                   device=self.device)
             sess.cache.seq_lens.zero_()

@@ -54,10 +54,10 @@ class ByteTokenizer:
         return "".join(parts)
 
     def encode_batch(
-        self, texts: list[str], max_len: int
+        self, texts: list[str], max_len: int, add_bos: bool = True
     ) -> tuple[list[list[int]], list[int]]:
         """Returns (padded id lists, true lengths)."""
-        seqs = [self.encode(t, max_len) for t in texts]
+        seqs = [self.encode(t, max_len, add_bos=add_bos) for t in texts]
         lens = [len(s) for s in seqs]
         width = max(lens) if lens else 1
         padded = [s + [PAD_ID] * (width - len(s)) for s in seqs]

@@ -134,3 +134,19 @@ def test_generate_with_prefix_kv_matches_disabled(monkeypatch):
                           stop_on_eos=False)
     assert base == cached
     assert eng._prefix_store  # the prefix KV was actually computed/cached
+
+
+def test_prefix_split_token_exactness():
+    """prefix_ids + suffix ids (no BOS) must reproduce the full prompt's
+    token stream exactly — regression for an off-by-one + stray-BOS split."""
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
+    prefix = "shared preamble text. " * 5
+    prompts = [prefix + f"tail {i}" for i in range(3)]
+    pre_ids, suffixes = eng._split_shared_prefix(prompts, 200)
+    assert len(pre_ids) >= 64
+    for p, suf in zip(prompts, suffixes):
+        full = eng.tokenizer.encode(p, None)
+        suf_ids = eng.tokenizer.encode(suf, None, add_bos=False)
+        assert pre_ids + suf_ids == full
