@@ -90,3 +90,47 @@ def test_decode_attention_bmm_cpu_path():
     want = ops.torch_ref.decode_attention(torch.randn(0), kc, vc, lens) \
         if False else None
     assert got.shape == (2, 4, 16)
+
+
+def test_batcher_groups_by_params():
+    """Requests with different sampling params never share a batch."""
+    import threading
+
+    from sentio_amd.serving.batcher import DynamicBatcher
+
+    class Rec:
+        def __init__(self):
+            self.batches = []
+
+        def generate(self, prompts, max_new_tokens=0, temperature=0.0, **kw):
+            import time
+            time.sleep(0.02)
+            self.batches.append((len(prompts), max_new_tokens, temperature))
+            return ["r"] * len(prompts)
+
+    eng = Rec()
+    b = DynamicBatcher(eng, max_batch=8, max_wait_ms=50)
+    threads = [
+        threading.Thread(target=b.generate, args=(f"p{i}",),
+                         kwargs={"max_new_tokens": 16 if i % 2 == 0 else 32,
+                                 "temperature": 0.3})
+        for i in range(6)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=10)
+    b.stop()
+    # every executed batch is param-homogeneous
+    assert all(m in (16, 32) for _, m, _ in eng.batches)
+    assert sum(n for n, _, _ in eng.batches) == 6
+    assert len(eng.batches) >= 2          # split across param groups
+
+
+def test_graphed_bucket_batch():
+    from sentio_amd.engines.graphed import bucket_batch
+
+    assert bucket_batch(1) == 1
+    assert bucket_batch(3) == 4
+    assert bucket_batch(33) == 64
+    assert bucket_batch(100) == 100       # beyond largest bucket: unbucketed
