@@ -150,3 +150,20 @@ def test_prefix_split_token_exactness():
         full = eng.tokenizer.encode(p, None)
         suf_ids = eng.tokenizer.encode(suf, None, add_bos=False)
         assert pre_ids + suf_ids == full
+
+
+def test_model_config_param_counts():
+    """Config shapes actually correspond to the named model classes."""
+    from sentio_amd.engines.configs import MODEL_CONFIGS
+
+    def b(name):
+        return MODEL_CONFIGS[name].n_params / 1e9
+
+    assert 7.0 <= b("llama3-8b") <= 9.0
+    assert 65 <= b("llama3-70b") <= 75
+    assert 6.5 <= b("qwen2-7b") <= 8.5
+    assert 6.5 <= b("mistral-7b") <= 8.0
+    enc = MODEL_CONFIGS["sentio-encoder-base"]
+    assert enc.dim == 1024 and not enc.causal    # jina-v3 class output dim
+    rr = MODEL_CONFIGS["sentio-reranker-base"]
+    assert rr.pooled_head == 1                   # scalar relevance head
