@@ -11,6 +11,12 @@ bucket sizes stay whole-tensor (decode activations are tiny).
 The Transformer engine consumes this through `TPContext`, which tells the
 weight container to materialize only this rank's shard (sliced from the
 same seeded generator so TP=N matches TP=1 bit-for-bit in fp32).
+
+Status: the per-layer all-reduces run in-stream (synchronous with compute).
+Overlapping them with the next projection on a dedicated HIP stream
+(SURVEY §7 hard part 2) is a planned round-2 optimization — at decode the
+activations are tiny ([B, dim] bf16 ≈ 256 KB at B=32), so the all-reduce is
+latency- not bandwidth-bound and the overlap window is the GEMM launch gap.
 """
 
 from __future__ import annotations
