@@ -443,8 +443,8 @@ def main():
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks
-    e = torch.tensor([elapsed])
+    # MAX over ranks (NCCL reduces GPU tensors; gloo reduces CPU tensors)
+    e = torch.tensor([elapsed], device=device if on_gpu else "cpu")
     if D.is_distributed():
         import torch.distributed as tdist
 

@@ -127,7 +127,9 @@ class ShardedIndex:
         return local
 
     def total_docs(self) -> int:
-        n = torch.tensor([len(self.dense)], dtype=torch.int64)
+        # NCCL reduces GPU tensors; gloo reduces CPU tensors
+        dev = self.device if self.device != "cpu" else "cpu"
+        n = torch.tensor([len(self.dense)], dtype=torch.int64, device=dev)
         if D.is_distributed():
             D.all_reduce_sum(n)
         return int(n.item())
