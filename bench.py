@@ -366,8 +366,12 @@ def main():
 
     def retrieval_phase_streamed(step: int):
         # side stream: retrieval kernels fill gaps under decode; the phase
-        # returns host data only (every GPU result is .cpu()'d inside)
+        # returns host data only (every GPU result is .cpu()'d inside).
+        # CUDA device selection is THREAD-LOCAL: this runs on the executor
+        # thread, which would otherwise default to device 0 on every rank
+        # and break NCCL object collectives at N >= 2.
         if on_gpu:
+            torch.cuda.set_device(device)
             with torch.cuda.stream(ret_stream):
                 return retrieval_phase(step)
         return retrieval_phase(step)
