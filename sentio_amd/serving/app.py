@@ -292,9 +292,12 @@ def create_app(settings: Settings | None = None,
 
     @app.get("/metrics/performance")
     async def metrics_performance():
+        from sentio_amd.observability.kernel_timer import timer_snapshot
+
         return {
             "metrics": metrics_collector.snapshot(),
             "monitors": performance_monitor.all_summaries(),
+            "gpu_regions": timer_snapshot(),
         }
 
     @app.get("/ui")

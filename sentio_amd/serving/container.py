@@ -38,6 +38,8 @@ class ServiceContainer:
             "reranker": CircuitBreaker("reranker"),
             "generator": CircuitBreaker("generator"),
         }
+        from sentio_amd.resilience.gpu_health import RankHeartbeat
+        self.heartbeat = RankHeartbeat()
 
     def _get(self, name: str, factory) -> Any:
         with self._lock:
@@ -176,6 +178,9 @@ class ServiceContainer:
 
             hc.register("encoder", check_encoder)
             hc.register("device", check_device)
+            from sentio_amd.resilience.gpu_health import gpu_health_check
+            hc.register("gpu_roundtrip",
+                        lambda: gpu_health_check(self.device))
             hc.register("dense_index", lambda: len(self.dense_index()) >= 0)
             for name, br in self.breakers.items():
                 hc.register(f"breaker:{name}",

@@ -57,6 +57,7 @@ class ChatHandler:
             }
             self.cache.l1.set_query(question, result)
             fallback_manager.cache_response(question, answer)
+            self.container.heartbeat.beat()
             return result
         except Exception as exc:
             logger.error("chat pipeline failed: %s", exc)
@@ -132,6 +133,7 @@ class HealthHandler:
                            "bm25": self.container.bm25_index().n_docs}
         checks["breakers"] = {k: b.health()["state"]
                               for k, b in self.container.breakers.items()}
+        checks["heartbeat"] = self.container.heartbeat.snapshot()
         status = "healthy" if checks.get("encoder") == "ok" else "degraded"
         result = {
             "status": status,

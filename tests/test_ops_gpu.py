@@ -542,3 +542,21 @@ def test_prefix_kv_prefill_logits_gpu(dev):
     torch.cuda.synchronize()
     torch.testing.assert_close(got, want, rtol=5e-2, atol=5e-1)
     assert (got.argmax(-1) == want.argmax(-1)).float().mean() >= 0.5
+
+
+def test_kernel_timer_event_path_and_gpu_health():
+    """HIP-event timer resolves real device time; device probe is healthy."""
+    from sentio_amd.observability.kernel_timer import KernelTimer
+    from sentio_amd.resilience.gpu_health import gpu_health_check
+
+    assert gpu_health_check("cuda:0") is True
+
+    t = KernelTimer("gpu_unit")
+    a = torch.randn(2048, 2048, device="cuda", dtype=torch.bfloat16)
+    with t.measure():
+        for _ in range(4):
+            a = a @ a
+    torch.cuda.synchronize()
+    t.flush()
+    assert t.count == 1
+    assert t.last_s > 0.0
