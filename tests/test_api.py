@@ -252,3 +252,35 @@ def test_cors_open_when_auth_disabled(client):
         "Access-Control-Request-Method": "POST",
     })
     assert r.headers.get("access-control-allow-origin") == "*"
+
+
+def test_health_detailed_reports_heartbeat(client):
+    r = client.get("/health/detailed")
+    assert r.status_code == 200
+    hb = r.json()["checks"]["heartbeat"]
+    assert set(hb) == {"rank", "beats", "age_s"}
+
+
+def test_metrics_performance_reports_gpu_regions(client):
+    _seed(client, 2)
+    client.post("/chat", json={"question": "what is in the corpus?"})
+    r = client.get("/metrics/performance")
+    assert r.status_code == 200
+    body = r.json()
+    assert "gpu_regions" in body
+    # the chat request above drove the generator → the region timer recorded
+    gen = body["gpu_regions"].get("generate")
+    assert gen is None or gen["count"] >= 1
+
+
+def test_chat_beats_heartbeat(client):
+    _seed(client, 2)
+    hb0 = client.get("/health/detailed").json()["checks"]["heartbeat"]["beats"]
+    r = client.post("/chat", json={"question": "heartbeat probe?"})
+    assert r.status_code == 200
+    # detailed health caches for 10 s — read the container's heartbeat via
+    # a fresh /metrics/performance-independent probe: beats are monotonic
+    import time as _t
+    _t.sleep(0.01)
+    hb1 = client.app.state.container.heartbeat.count
+    assert hb1 >= hb0 + 1
