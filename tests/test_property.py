@@ -1,0 +1,157 @@
+"""Property-based tests (hypothesis) for the invariant-heavy host paths:
+tokenizer reversibility, BM25 scoring vs a naive oracle, fusion math, and
+the JSON reply extractor's repair rules.
+
+The reference's suite is example-based mocks only; these
+generate adversarial inputs instead, which is where byte-level
+tokenization and JSON repair actually break.
+"""
+
+from __future__ import annotations
+
+import json
+import math
+
+import numpy as np
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from sentio_amd.engines.tokenizer import BOS_ID, BYTE_OFFSET, ByteTokenizer
+from sentio_amd.index.bm25 import BM25Index, tokenize
+from sentio_amd.index.fusion import fuse
+from sentio_amd.pipeline.verifier import extract_json_dict
+
+tok = ByteTokenizer()
+
+
+# ---- tokenizer ----
+
+@given(st.text(max_size=200))
+@settings(max_examples=200, deadline=None)
+def test_tokenizer_roundtrip(text):
+    ids = tok.encode(text, add_bos=False)
+    assert tok.decode(ids) == text
+
+
+@given(st.text(min_size=0, max_size=100))
+@settings(max_examples=100, deadline=None)
+def test_tokenizer_bos_and_count(text):
+    ids = tok.encode(text)
+    assert ids[:1] == [BOS_ID] or text == "" and ids == [BOS_ID]
+    assert tok.count_tokens(text) == len(ids)
+    assert all(i < tok.vocab_size for i in ids)
+
+
+@given(st.text(max_size=100), st.integers(min_value=1, max_value=16))
+@settings(max_examples=100, deadline=None)
+def test_tokenizer_truncation_is_prefix(text, max_len):
+    full = tok.encode(text)
+    cut = tok.encode(text, max_len=max_len)
+    assert cut == full[:max_len]
+
+
+@given(st.lists(st.text(max_size=40), min_size=1, max_size=8))
+@settings(max_examples=50, deadline=None)
+def test_tokenizer_batch_padding(texts):
+    padded, lens = tok.encode_batch(texts, max_len=64)
+    width = max(lens)
+    for i, (row, n) in enumerate(zip(padded, lens)):
+        assert len(row) == width
+        assert all(t == 0 for t in row[n:])          # PAD after the live part
+        assert row[:n] == tok.encode(texts[i], max_len=64)
+
+
+# ---- BM25 vs naive oracle ----
+
+_word = st.text(alphabet=st.characters(whitelist_categories=("Ll",),
+                                       max_codepoint=0x7A), min_size=1, max_size=6)
+_doc = st.lists(_word, min_size=1, max_size=20).map(" ".join)
+
+
+@given(st.lists(_doc, min_size=1, max_size=12), _doc)
+@settings(max_examples=60, deadline=None)
+def test_bm25_matches_naive_oracle(docs, query):
+    idx = BM25Index()
+    idx.build([f"d{i}" for i in range(len(docs))], docs)
+    got = idx.get_scores(query)
+
+    # naive per-doc Okapi computation straight from the formula
+    n = len(docs)
+    toks = [tokenize(d) for d in docs]
+    avgdl = max(sum(len(t) for t in toks) / n, 1e-9)
+    # NOTE: repeated query terms contribute once PER OCCURRENCE — that is
+    # rank_bm25's convention (reference sparse.py uses rank_bm25.get_scores)
+    # and the index matches it.
+    want = np.zeros(n, np.float32)
+    q_terms = [t for t in tokenize(query) if t in idx.vocab]
+    for t in q_terms:
+        df = sum(1 for d in toks if t in d)
+        idf = math.log((n - df + 0.5) / (df + 0.5) + 1.0)
+        for di, d in enumerate(toks):
+            tf = d.count(t)
+            if tf:
+                want[di] += idf * tf * (idx.k1 + 1) / (
+                    tf + idx.k1 * (1 - idx.b + idx.b * len(d) / avgdl))
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+@given(st.lists(_doc, min_size=1, max_size=8), _doc,
+       st.integers(min_value=1, max_value=10))
+@settings(max_examples=40, deadline=None)
+def test_bm25_search_sorted_and_positive(docs, query, k):
+    idx = BM25Index()
+    idx.build([f"d{i}" for i in range(len(docs))], docs)
+    hits = idx.search(query, k)
+    scores = [s for _, s in hits]
+    assert scores == sorted(scores, reverse=True)
+    assert all(s > 0 for s in scores)
+    assert len(hits) <= k
+
+
+# ---- fusion ----
+
+_hitlist = st.lists(
+    st.tuples(st.integers(min_value=0, max_value=30),
+              st.floats(min_value=0.01, max_value=10, allow_nan=False)),
+    min_size=0, max_size=10,
+).map(lambda l: [(f"doc{i}", s) for i, s in
+                 {i: s for i, s in l}.items()])  # dedup ids
+
+
+@given(_hitlist, _hitlist)
+@settings(max_examples=60, deadline=None)
+def test_rrf_fusion_properties(dense, sparse):
+    fused = fuse(dense, sparse, method="rrf", rrf_k=60, top_k=20)
+    ids = [d for d, _ in fused]
+    scores = [s for _, s in fused]
+    assert len(ids) == len(set(ids))                       # dedup
+    assert scores == sorted(scores, reverse=True)          # sorted
+    assert set(ids) <= {d for d, _ in dense} | {d for d, _ in sparse}
+    # a doc in both lists outranks one at identical ranks in only one list
+    if dense and sparse and dense[0][0] == sparse[0][0]:
+        both = dense[0][0]
+        only = next((d for d, _ in dense + sparse if d != both), None)
+        if only is not None:
+            assert ids.index(both) < ids.index(only)
+
+
+# ---- reply extractor ----
+
+@given(st.dictionaries(
+    st.sampled_from(["verdict", "citations_ok", "notes", "x"]),
+    st.one_of(st.text(max_size=20), st.booleans(),
+              st.integers(min_value=-100, max_value=100)),
+    min_size=1, max_size=4))
+@settings(max_examples=80, deadline=None)
+def test_extract_json_roundtrip_with_noise(d):
+    blob = json.dumps(d)
+    for wrapper in (blob,
+                    f"Sure! Here is the JSON:\n```json\n{blob}\n```\nDone.",
+                    f"prefix text {blob} suffix text"):
+        got = extract_json_dict(wrapper)
+        assert got == d, (wrapper, got)
+
+
+def test_extract_json_repairs_trailing_comma_and_constants():
+    got = extract_json_dict('{"verdict": "pass", "ok": True, "n": 1,}')
+    assert got == {"verdict": "pass", "ok": True, "n": 1}
