@@ -1,0 +1,119 @@
+"""Host-side concurrency smoke tests (SURVEY §5 race detection: the
+reference's only story is scattered locks; here the serving stack is
+hammered from many threads to exercise the locking for real — container
+lazy-init, cache tiers, dynamic batcher, breaker counters, metrics).
+
+The C++/HIP side has its own sanitizer build (`SENTIO_SANITIZE=1` in
+setup.py); these cover the Python host paths on CPU.
+"""
+
+from __future__ import annotations
+
+import threading
+from concurrent.futures import ThreadPoolExecutor
+
+import pytest
+from fastapi.testclient import TestClient
+
+from sentio_amd.config import Settings
+from sentio_amd.serving.app import create_app
+from sentio_amd.serving.container import ServiceContainer
+
+
+@pytest.fixture()
+def client():
+    s = Settings()
+    s.mock_compute = True
+    s.device = "cpu"
+    s.use_reranker = True
+    s.use_verifier = False
+    s.rate_limit_chat_per_min = 100000  # not testing the limiter here
+    s.rate_limit_embed_per_min = 100000
+    container = ServiceContainer(s)
+    app = create_app(s, container)
+    with TestClient(app) as c:
+        c.app_container = container
+        yield c
+
+
+def _embed(client, i):
+    return client.post("/embed", json={
+        "content": f"document number {i} about topic {i % 4}",
+        "metadata": {"i": i}}).status_code
+
+
+def test_concurrent_lazy_init_single_instance(client):
+    """16 threads race the container's first-touch lazy init; every getter
+    must hand back the same singleton."""
+    c = client.app_container
+    seen = []
+
+    def grab():
+        seen.append((id(c.encoder()), id(c.pipeline()), id(c.cache_manager())))
+
+    threads = [threading.Thread(target=grab) for _ in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert len(set(seen)) == 1
+
+
+def test_concurrent_embed_then_chat(client):
+    with ThreadPoolExecutor(max_workers=8) as ex:
+        codes = list(ex.map(lambda i: _embed(client, i), range(24)))
+    assert codes == [200] * 24
+
+    def ask(i):
+        r = client.post("/chat", json={"question": f"what about topic {i % 4}?"})
+        return r.status_code, bool(r.json().get("answer"))
+
+    with ThreadPoolExecutor(max_workers=8) as ex:
+        results = list(ex.map(ask, range(32)))
+    assert all(code == 200 and has_answer for code, has_answer in results)
+
+    # heartbeat counted every served request; index holds every embed
+    assert client.app_container.heartbeat.count >= 1
+    assert len(client.app_container.dense_index()) == 24
+
+
+def test_concurrent_mixed_traffic_consistency(client):
+    """Interleave embeds, chats, health checks, metrics scrapes, and cache
+    clears — nothing may 500 and final state must be coherent."""
+    for i in range(6):
+        _embed(client, i)
+
+    def worker(i):
+        kind = i % 5
+        if kind == 0:
+            return client.post("/chat", json={"question": f"q{i}?"}).status_code
+        if kind == 1:
+            return _embed(client, 100 + i)
+        if kind == 2:
+            return client.get("/health/detailed").status_code
+        if kind == 3:
+            return client.get("/metrics/performance").status_code
+        return client.post("/clear").status_code
+
+    with ThreadPoolExecutor(max_workers=10) as ex:
+        codes = list(ex.map(worker, range(40)))
+    assert all(c == 200 for c in codes), codes
+
+    # the service still works after the storm
+    r = client.post("/chat", json={"question": "still alive?"})
+    assert r.status_code == 200 and r.json()["answer"]
+
+
+def test_concurrent_cache_manager_thread_safety(client):
+    cm = client.app_container.cache_manager()
+
+    def churn(tid):
+        for i in range(200):
+            cm.l1.set(f"k{tid}:{i % 20}", {"v": i})
+            cm.l1.get(f"k{(tid + 1) % 8}:{i % 20}")
+        return True
+
+    with ThreadPoolExecutor(max_workers=8) as ex:
+        assert all(ex.map(churn, range(8)))
+    stats = cm.l1.stats()
+    assert stats["hits"] + stats["misses"] >= 8 * 200
