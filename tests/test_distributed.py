@@ -224,3 +224,24 @@ def _tp_generate_worker(rank, world):
 
 def test_tp_generation_consistent_across_ranks():
     _run_workers(_tp_generate_worker, world=2)
+
+
+# ---- per-rank heartbeat gather (SURVEY §5 failure detection) ----
+
+def _heartbeat_worker(rank, world):
+    from sentio_amd.resilience.gpu_health import RankHeartbeat
+
+    hb = RankHeartbeat()
+    for _ in range(rank + 1):            # rank r beats r+1 times
+        hb.beat()
+    snaps = hb.gather_heartbeats()
+    assert len(snaps) == world
+    by_rank = {s["rank"]: s for s in snaps}
+    assert set(by_rank) == set(range(world))
+    for r in range(world):
+        assert by_rank[r]["beats"] == r + 1
+        assert by_rank[r]["age_s"] >= 0.0
+
+
+def test_heartbeat_gather_world2():
+    _run_workers(_heartbeat_worker, world=2)
