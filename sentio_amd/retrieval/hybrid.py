@@ -19,6 +19,8 @@ class HybridRetriever(BaseRetriever):
         dense_weight: float = 0.7,
         sparse_weight: float = 0.3,
         scorer_plugins: list[ScorerPlugin] | None = None,
+        cache_retriever=None,
+        cache_score_threshold: float = 0.9,
     ):
         self.dense = dense
         self.sparse = sparse
@@ -27,9 +29,35 @@ class HybridRetriever(BaseRetriever):
         self.dense_weight = dense_weight
         self.sparse_weight = sparse_weight
         self.scorer_plugins = scorer_plugins or []
+        # optional hot second collection probed before the main corpus
+        # (reference hybrid.py:96-107,146-182 "web_cache" collection):
+        # strong cache hits short-circuit the full hybrid search; weaker
+        # ones join the dense candidate pool for fusion.
+        self.cache_retriever = cache_retriever
+        self.cache_score_threshold = cache_score_threshold
 
     def retrieve(self, query: str, top_k: int = 10) -> list[Document]:
+        cache_docs: list[Document] = []
+        if self.cache_retriever is not None:
+            try:
+                cache_docs = self.cache_retriever.retrieve(query, top_k)
+            except Exception:
+                cache_docs = []
+            for d in cache_docs:
+                d.metadata["from_cache_collection"] = True
+            strong = [d for d in cache_docs
+                      if float(d.metadata.get("score", 0.0))
+                      >= self.cache_score_threshold]
+            if len(strong) >= top_k:
+                return strong[:top_k]
+
         dense_docs = self.dense.retrieve(query, top_k) if self.dense else []
+        if cache_docs:
+            seen = {d.id for d in dense_docs}
+            dense_docs = dense_docs + [d for d in cache_docs
+                                       if d.id not in seen]
+            dense_docs.sort(key=lambda d: float(d.metadata.get("score", 0.0)),
+                            reverse=True)
         sparse_docs = self.sparse.retrieve(query, top_k) if self.sparse else []
 
         dense_hits = [(d.id, float(d.metadata.get("score", 0.0))) for d in dense_docs]

@@ -121,3 +121,62 @@ def test_dense_search_metadata_filter():
     hits = idx.search(vecs[1], top_k=5, metadata_filter={"lang": "de"})[0]
     assert hits and all(h[0] in {"d1", "d3", "d5", "d7", "d9"} for h in hits)
     assert hits[0][0] == "d1"   # self-match still ranks first among de docs
+
+
+# ---- web-cache second collection (reference hybrid.py:96-107,146-182) ----
+
+class _CannedRetriever:
+    def __init__(self, docs):
+        self.docs = docs
+        self.calls = 0
+
+    def retrieve(self, query, top_k=10):
+        self.calls += 1
+        return self.docs[:top_k]
+
+
+def _mk(i, score, cache=False):
+    from sentio_amd.models.document import Document
+    d = Document(text=f"doc {i}", metadata={"score": score}, id=f"{'c' if cache else 'm'}{i}")
+    return d
+
+
+def test_web_cache_short_circuits_on_strong_hits():
+    from sentio_amd.retrieval.hybrid import HybridRetriever
+
+    cache = _CannedRetriever([_mk(i, 0.95, cache=True) for i in range(3)])
+    main = _CannedRetriever([_mk(i, 0.5) for i in range(3)])
+    h = HybridRetriever(dense=main, cache_retriever=cache,
+                        cache_score_threshold=0.9)
+    out = h.retrieve("q", top_k=2)
+    assert [d.id for d in out] == ["c0", "c1"]
+    assert all(d.metadata["from_cache_collection"] for d in out)
+    assert main.calls == 0          # main corpus never touched
+
+
+def test_web_cache_weak_hits_join_fusion():
+    from sentio_amd.retrieval.hybrid import HybridRetriever
+
+    cache = _CannedRetriever([_mk(0, 0.4, cache=True)])
+    main = _CannedRetriever([_mk(i, 0.8 - 0.1 * i) for i in range(3)])
+    h = HybridRetriever(dense=main, cache_retriever=cache,
+                        cache_score_threshold=0.9)
+    out = h.retrieve("q", top_k=4)
+    assert main.calls == 1
+    ids = [d.id for d in out]
+    assert "c0" in ids and "m0" in ids
+    # cache doc carries its marker through fusion
+    assert next(d for d in out if d.id == "c0").metadata["from_cache_collection"]
+
+
+def test_web_cache_errors_are_soft():
+    from sentio_amd.retrieval.hybrid import HybridRetriever
+
+    class _Boom:
+        def retrieve(self, q, top_k=10):
+            raise RuntimeError("cache down")
+
+    main = _CannedRetriever([_mk(i, 0.8) for i in range(2)])
+    h = HybridRetriever(dense=main, cache_retriever=_Boom())
+    out = h.retrieve("q", top_k=2)
+    assert len(out) == 2 and main.calls == 1
