@@ -29,7 +29,10 @@ def create_retriever(
                            device=device)
     if strategy == "dense":
         return dense
-    if strategy in ("bm25", "sparse"):
+    if strategy in ("bm25", "sparse", "pyserini"):
+        # "pyserini" (reference factory.py Lucene backend) maps to the same
+        # CSR BM25 — the GPU index IS the Lucene-class engine here, so a
+        # reference deployment's RETRIEVAL_STRATEGY keeps working.
         return sparse
     if strategy == "hybrid":
         return HybridRetriever(
