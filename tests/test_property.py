@@ -155,3 +155,89 @@ def test_extract_json_roundtrip_with_noise(d):
 def test_extract_json_repairs_trailing_comma_and_constants():
     got = extract_json_dict('{"verdict": "pass", "ok": True, "n": 1,}')
     assert got == {"verdict": "pass", "ok": True, "n": 1}
+
+
+# ---- op reference math invariants ----
+
+import torch
+
+from sentio_amd.ops import torch_ref as R
+
+
+@given(st.integers(min_value=1, max_value=4), st.integers(min_value=2, max_value=32),
+       st.integers(min_value=1, max_value=4))
+@settings(max_examples=30, deadline=None)
+def test_rope_preserves_pairwise_norm(b, s, h):
+    """RoPE is a rotation — it must preserve the norm of every (even,odd)
+    feature pair, hence the whole vector norm."""
+    d = 16
+    torch.manual_seed(0)
+    x = torch.randn(b, s, h, d)
+    cos, sin = R.rope_tables(s, d)
+    pos = torch.arange(s, dtype=torch.int32).unsqueeze(0).expand(b, s)
+    y = R.rope_apply(x, cos, sin, pos)
+    assert torch.allclose(x.norm(dim=-1), y.norm(dim=-1), atol=1e-4)
+    # position 0 is the identity rotation
+    assert torch.allclose(y[:, 0], x[:, 0], atol=1e-6)
+
+
+@given(st.integers(min_value=1, max_value=8), st.floats(min_value=0.5, max_value=4.0))
+@settings(max_examples=30, deadline=None)
+def test_rmsnorm_scale_invariant(rows, scale):
+    """rmsnorm(c·x) == rmsnorm(x) for any positive scalar c (up to eps)."""
+    torch.manual_seed(1)
+    x = torch.randn(rows, 64) + 0.1
+    w = torch.rand(64) + 0.5
+    a = R.rmsnorm(x, w)
+    b = R.rmsnorm(x * scale, w)
+    assert torch.allclose(a, b, atol=1e-3)
+    # unit-RMS output property with w=1
+    ones = R.rmsnorm(x, torch.ones(64))
+    rms = ones.pow(2).mean(-1).sqrt()
+    assert torch.allclose(rms, torch.ones(rows), atol=1e-2)
+
+
+def test_softmax_rows_sum_to_one_and_shift_invariant():
+    torch.manual_seed(2)
+    x = torch.randn(5, 33) * 10
+    p = R.softmax(x)
+    assert torch.allclose(p.sum(-1), torch.ones(5), atol=1e-5)
+    assert torch.allclose(R.softmax(x + 100.0), p, atol=1e-5)
+
+
+def test_swiglu_packed_matches_split():
+    torch.manual_seed(3)
+    g = torch.randn(4, 32)
+    u = torch.randn(4, 32)
+    packed = torch.cat([g, u], dim=-1)
+    assert torch.allclose(R.swiglu_packed(packed), R.swiglu(g, u), atol=1e-6)
+
+
+def test_mean_pool_ignores_masked_positions():
+    torch.manual_seed(4)
+    h = torch.randn(2, 6, 16)
+    mask = torch.tensor([[1, 1, 1, 0, 0, 0], [1, 1, 1, 1, 1, 1]], dtype=torch.float32)
+    out = R.mean_pool_l2norm(h, mask)
+    # poisoning masked positions must not change the output
+    h2 = h.clone()
+    h2[0, 3:] = 1e6
+    out2 = R.mean_pool_l2norm(h2, mask)
+    assert torch.allclose(out[0], out2[0], atol=1e-5)
+    assert torch.allclose(out.norm(dim=-1), torch.ones(2), atol=1e-5)
+
+
+def test_attention_cache_matches_full_attention():
+    """Suffix attention against a prefix KV cache == full-sequence causal
+    attention restricted to the suffix rows."""
+    torch.manual_seed(5)
+    B, H, S, D = 2, 4, 10, 16
+    P = 6  # prefix length
+    q = torch.randn(B, S, H, D)       # [B,S,H,D] layout
+    k = torch.randn(B, S, H, D)
+    v = torch.randn(B, S, H, D)
+    full = R.attention(q, k, v, causal=True)          # [B,S,H,D]
+    kc = k.permute(0, 2, 1, 3).contiguous()           # [B,Hkv,Smax,D]
+    vc = v.permute(0, 2, 1, 3).contiguous()
+    suff = R.attention_cache(q[:, P:], kc, vc,
+                             torch.full((B,), S, dtype=torch.int32), q_off=P)
+    assert torch.allclose(full[:, P:], suff, atol=1e-5)
