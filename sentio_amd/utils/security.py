@@ -111,3 +111,99 @@ class SanitizingFilter(logging.Filter):
 
 def setup_log_sanitization() -> None:
     logging.getLogger().addFilter(SanitizingFilter())
+
+
+# ---- CSRF tokens (reference security.py:331-390 region capability) ----
+
+class CSRFProtection:
+    """HMAC-signed, session-bound, expiring CSRF tokens.
+
+    Stateless: token = "<session>.<expiry>.<hmac(secret, session|expiry)>";
+    verification recomputes the MAC, so no server-side token store."""
+
+    def __init__(self, secret: str, ttl_s: int = 3600):
+        self._secret = secret.encode()
+        self.ttl_s = ttl_s
+
+    def _mac(self, session_id: str, expiry: int) -> str:
+        import hashlib
+        import hmac as _hmac
+
+        msg = f"{session_id}|{expiry}".encode()
+        return _hmac.new(self._secret, msg, hashlib.sha256).hexdigest()[:32]
+
+    def generate(self, session_id: str) -> str:
+        import time
+
+        expiry = int(time.time()) + self.ttl_s
+        return f"{session_id}.{expiry}.{self._mac(session_id, expiry)}"
+
+    def verify(self, token: str, session_id: str) -> bool:
+        import hmac as _hmac
+        import time
+
+        try:
+            sess, expiry_s, mac = token.rsplit(".", 2)
+            expiry = int(expiry_s)
+        except (ValueError, AttributeError):
+            return False
+        if sess != session_id or expiry < time.time():
+            return False
+        return _hmac.compare_digest(mac, self._mac(sess, expiry))
+
+
+# ---- client IP validation (reference security.py IP-validation capability) ----
+
+def validate_client_ip(ip: str,
+                       allow: list[str] | None = None,
+                       block: list[str] | None = None) -> bool:
+    """True iff `ip` parses, is not in any blocked CIDR, and (when an
+    allowlist is given) is inside at least one allowed CIDR."""
+    import ipaddress
+
+    try:
+        addr = ipaddress.ip_address(ip)
+    except ValueError:
+        return False
+    for cidr in block or []:
+        if addr in ipaddress.ip_network(cidr, strict=False):
+            return False
+    if allow:
+        return any(addr in ipaddress.ip_network(c, strict=False)
+                   for c in allow)
+    return True
+
+
+# ---- adaptive rate limits (reference security.py:331-560 capability) ----
+
+class AdaptiveRateLimit:
+    """Per-endpoint request budget that tightens under error pressure.
+
+    effective = base · f(error_rate): full budget while healthy, linearly
+    down to `floor_fraction` of it as the recent error rate climbs to
+    `max_error_rate`.  The serving limiter polls `current_limit()`; errors
+    and successes are reported by the caller (advisory, like the
+    reference's — the static limiter stays the default)."""
+
+    def __init__(self, base_per_min: int, floor_fraction: float = 0.2,
+                 max_error_rate: float = 0.5, window: int = 100):
+        from collections import deque
+
+        self.base_per_min = base_per_min
+        self.floor_fraction = floor_fraction
+        self.max_error_rate = max_error_rate
+        self._events = deque(maxlen=window)
+
+    def record(self, ok: bool) -> None:
+        self._events.append(bool(ok))
+
+    @property
+    def error_rate(self) -> float:
+        if not self._events:
+            return 0.0
+        return 1.0 - (sum(self._events) / len(self._events))
+
+    def current_limit(self) -> int:
+        pressure = min(self.error_rate / self.max_error_rate, 1.0)
+        frac = 1.0 - (1.0 - self.floor_fraction) * pressure
+        return max(1, round(self.base_per_min * frac))

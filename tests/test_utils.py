@@ -107,3 +107,50 @@ def test_resource_monitor_snapshot_keys():
 
     snap = resource_monitor.snapshot()
     assert "cpu_percent" in snap or "memory" in snap or snap  # psutil-backed
+
+
+# ---- CSRF / IP validation / adaptive rate limit (reference security.py:331-560) ----
+
+def test_csrf_token_roundtrip_and_rejections():
+    from sentio_amd.utils.security import CSRFProtection
+
+    c = CSRFProtection("secret-key", ttl_s=60)
+    t = c.generate("sess-1")
+    assert c.verify(t, "sess-1")
+    assert not c.verify(t, "sess-2")            # bound to session
+    assert not c.verify(t + "x", "sess-1")      # tampered MAC
+    assert not c.verify("garbage", "sess-1")
+    expired = CSRFProtection("secret-key", ttl_s=-10)
+    assert not expired.verify(expired.generate("sess-1"), "sess-1")
+    other = CSRFProtection("different-secret", ttl_s=60)
+    assert not other.verify(t, "sess-1")        # wrong secret
+
+
+def test_validate_client_ip():
+    from sentio_amd.utils.security import validate_client_ip
+
+    assert validate_client_ip("10.0.0.5")
+    assert not validate_client_ip("not-an-ip")
+    assert not validate_client_ip("10.0.0.5", block=["10.0.0.0/8"])
+    assert validate_client_ip("10.0.0.5", allow=["10.0.0.0/24"])
+    assert not validate_client_ip("10.0.1.5", allow=["10.0.0.0/24"])
+    assert validate_client_ip("2001:db8::1", allow=["2001:db8::/32"])
+
+
+def test_adaptive_rate_limit_tightens_under_errors():
+    from sentio_amd.utils.security import AdaptiveRateLimit
+
+    rl = AdaptiveRateLimit(base_per_min=100, floor_fraction=0.2,
+                           max_error_rate=0.5, window=10)
+    assert rl.current_limit() == 100            # healthy
+    for _ in range(10):
+        rl.record(True)
+    assert rl.current_limit() == 100
+    for _ in range(10):
+        rl.record(False)                        # 100% errors
+    assert rl.error_rate == 1.0
+    assert rl.current_limit() == 20             # floor = 20% of base
+    for _ in range(5):
+        rl.record(True)                         # recovering: 50% errors
+    lim = rl.current_limit()
+    assert 20 < lim <= 100
