@@ -150,7 +150,8 @@ def test_adaptive_rate_limit_tightens_under_errors():
         rl.record(False)                        # 100% errors
     assert rl.error_rate == 1.0
     assert rl.current_limit() == 20             # floor = 20% of base
-    for _ in range(5):
-        rl.record(True)                         # recovering: 50% errors
+    for _ in range(8):
+        rl.record(True)    # window=10 now holds 2 errors -> rate 0.2
+    assert abs(rl.error_rate - 0.2) < 1e-9
     lim = rl.current_limit()
-    assert 20 < lim <= 100
+    assert 20 < lim < 100  # partial pressure: between floor and base
