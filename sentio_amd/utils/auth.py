@@ -58,6 +58,7 @@ class AuthManager:
     refresh_ttl_s: int = 86400
     api_keys: dict[str, UserRole] = field(default_factory=dict)
     audit_log: list[dict[str, Any]] = field(default_factory=list)
+    sessions: dict[str, dict[str, Any]] = field(default_factory=dict)
 
     def _sign(self, payload: bytes) -> str:
         return base64.urlsafe_b64encode(
@@ -114,6 +115,33 @@ class AuthManager:
         if role is None:
             raise AuthError("unknown api key")
         return role
+
+    # sessions (reference auth.py session-store capability): server-side
+    # revocable handles, unlike the stateless HMAC tokens above
+    def create_session(self, subject: str, role: UserRole = UserRole.READER,
+                       ttl_s: float | None = None) -> str:
+        sid = "sess-" + secrets.token_urlsafe(18)
+        self.sessions[sid] = {
+            "subject": subject, "role": role,
+            "expires": time.time() + (ttl_s or self.token_ttl_s),
+        }
+        self._audit("session.create", subject, sid)
+        return sid
+
+    def validate_session(self, session_id: str) -> dict:
+        s = self.sessions.get(session_id)
+        if s is None:
+            raise AuthError("unknown session")
+        if s["expires"] < time.time():
+            self.sessions.pop(session_id, None)
+            raise AuthError("session expired")
+        return s
+
+    def revoke_session(self, session_id: str) -> bool:
+        s = self.sessions.pop(session_id, None)
+        if s is not None:
+            self._audit("session.revoke", s["subject"], session_id)
+        return s is not None
 
     def require_scopes(self, token: str, *scopes: AuthScope) -> TokenData:
         data = self.verify_token(token)

@@ -155,3 +155,22 @@ def test_adaptive_rate_limit_tightens_under_errors():
     assert abs(rl.error_rate - 0.2) < 1e-9
     lim = rl.current_limit()
     assert 20 < lim < 100  # partial pressure: between floor and base
+
+
+def test_auth_sessions_lifecycle():
+    from sentio_amd.utils.auth import AuthError, AuthManager, UserRole
+
+    am = AuthManager()
+    sid = am.create_session("alice", UserRole.ADMIN, ttl_s=60)
+    s = am.validate_session(sid)
+    assert s["subject"] == "alice" and s["role"] == UserRole.ADMIN
+    assert am.revoke_session(sid)
+    import pytest as _pytest
+    with _pytest.raises(AuthError):
+        am.validate_session(sid)
+    assert not am.revoke_session(sid)           # already gone
+    expired = am.create_session("bob", ttl_s=-1)
+    with _pytest.raises(AuthError):
+        am.validate_session(expired)
+    actions = [e["action"] for e in am.audit_log]
+    assert "session.create" in actions and "session.revoke" in actions
