@@ -29,6 +29,10 @@ class EncoderEngine:
         self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
         self.dim = self.cfg.dim
         self.calls = 0
+        self.cache_hits = 0
+        self.texts_embedded = 0
+        self.errors = 0
+        self.total_time_s = 0.0
         self.cache = None
         if cache_size > 0:
             from sentio_amd.caching.memory import MemoryCache
@@ -58,6 +62,7 @@ class EncoderEngine:
             hit = self.cache.get_embedding(t)
             if hit is not None:
                 out[i] = hit.to(self.device)
+                self.cache_hits += 1
             else:
                 miss_idx.append(i)
         for i0 in range(0, len(miss_idx), batch_size):
@@ -84,7 +89,32 @@ class EncoderEngine:
                                                     self.device)
         return pool
 
+    def stats(self) -> dict:
+        """Usage counters (reference embeddings/base.py:245-284 stats role:
+        calls, cache hits, errors, average embed time)."""
+        n = max(self.texts_embedded, 1)
+        return {
+            "calls": self.calls,
+            "texts_embedded": self.texts_embedded,
+            "cache_hits": self.cache_hits,
+            "errors": self.errors,
+            "avg_time_ms_per_text": round(1e3 * self.total_time_s / n, 3),
+        }
+
     def _embed_batch(self, texts: list[str]) -> torch.Tensor:
+        import time as _time
+
+        t0 = _time.perf_counter()
+        try:
+            v = self._embed_batch_inner(texts)
+        except Exception:
+            self.errors += 1
+            raise
+        self.texts_embedded += len(texts)
+        self.total_time_s += _time.perf_counter() - t0
+        return v
+
+    def _embed_batch_inner(self, texts: list[str]) -> torch.Tensor:
         padded, lens = self.tokenizer.encode_batch(texts, self.max_seq)
         tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
         B, S = tokens.shape

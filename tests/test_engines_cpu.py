@@ -167,3 +167,18 @@ def test_model_config_param_counts():
     assert enc.dim == 1024 and not enc.causal    # jina-v3 class output dim
     rr = MODEL_CONFIGS["sentio-reranker-base"]
     assert rr.pooled_head == 1                   # scalar relevance head
+
+
+def test_encoder_stats_counters():
+    from sentio_amd.engines.encoder import EncoderEngine
+
+    enc = EncoderEngine("sentio-encoder-small", device="cpu", dtype="fp32",
+                        max_seq=32, cache_size=16)
+    enc.embed(["alpha", "beta"])
+    enc.embed(["alpha", "gamma"])      # alpha hits the cache
+    s = enc.stats()
+    assert s["calls"] == 2
+    assert s["texts_embedded"] == 3    # alpha embedded once
+    assert s["cache_hits"] == 1
+    assert s["errors"] == 0
+    assert s["avg_time_ms_per_text"] > 0
