@@ -284,3 +284,25 @@ def test_chat_beats_heartbeat(client):
     _t.sleep(0.01)
     hb1 = client.app.state.container.heartbeat.count
     assert hb1 >= hb0 + 1
+
+
+def test_chat_with_conversation_history(client):
+    _seed(client, 2)
+    r = client.post("/chat", json={
+        "question": "and what about follow-ups?",
+        "history": [
+            {"role": "user", "content": "tell me about gpus"},
+            {"role": "assistant", "content": "gpus are parallel processors"},
+        ],
+    })
+    assert r.status_code == 200
+    assert r.json()["answer"]
+
+
+def test_chat_history_malformed_entries_tolerated(client):
+    _seed(client, 1)
+    r = client.post("/chat", json={
+        "question": "robust?",
+        "history": [{"not_role": "x"}, {}],
+    })
+    assert r.status_code == 200
