@@ -153,6 +153,8 @@ class Settings:
         "enable_metrics": "ENABLE_METRICS",
         "enable_tracing": "ENABLE_TRACING",
         "auth_secret": "AUTH_SECRET",
+        "rate_limit_chat_per_min": "RATE_LIMIT_CHAT_PER_MIN",
+        "rate_limit_embed_per_min": "RATE_LIMIT_EMBED_PER_MIN",
     }
 
     @classmethod

@@ -210,3 +210,13 @@ def test_verifier_parses_fail_with_revision():
     out = v.verify("q", "ctx", "ans")
     assert out["verdict"] == "fail"
     assert out["revised_answer"] == "fixed [1]"
+
+
+def test_rate_limit_settings_from_env(monkeypatch):
+    from sentio_amd.config import Settings
+
+    monkeypatch.setenv("RATE_LIMIT_CHAT_PER_MIN", "555")
+    monkeypatch.setenv("RATE_LIMIT_EMBED_PER_MIN", "44")
+    s = Settings.from_env()
+    assert s.rate_limit_chat_per_min == 555
+    assert s.rate_limit_embed_per_min == 44
