@@ -241,3 +241,64 @@ def test_attention_cache_matches_full_attention():
     suff = R.attention_cache(q[:, P:], kc, vc,
                              torch.full((B,), S, dtype=torch.int32), q_off=P)
     assert torch.allclose(full[:, P:], suff, atol=1e-5)
+
+
+# ---- selector node invariants (reference nodes.py:249-372 semantics) ----
+
+from sentio_amd.models.document import Document
+from sentio_amd.pipeline.nodes import create_selector_node
+from sentio_amd.pipeline.state import create_initial_state
+
+
+@given(st.lists(
+    st.tuples(st.integers(min_value=0, max_value=6),      # id collisions
+              st.floats(min_value=-5, max_value=5, allow_nan=False),
+              st.integers(min_value=0, max_value=400)),   # text length
+    min_size=0, max_size=15),
+    st.integers(min_value=1, max_value=6),
+    st.integers(min_value=10, max_value=200))
+@settings(max_examples=80, deadline=None)
+def test_selector_budget_dedup_order(items, k, budget):
+    docs = [Document(text="x" * n, metadata={"score": s}, id=f"d{i}")
+            for i, s, n in items]
+    state = create_initial_state("q", {})
+    state["retrieved_documents"] = docs
+    node = create_selector_node(top_k=k, max_tokens=budget)
+    out = node(state)
+    sel = out.get("selected_documents") or []
+
+    ids = [d.id for d in sel]
+    assert len(ids) == len(set(ids))                       # dedup by id
+    assert len(sel) <= k
+    total = sum(len(d.text) // 4 for d in sel)
+    assert total <= budget                                 # 4 chars ≈ 1 token
+    scores = [float(d.metadata["score"]) for d in sel]
+    assert scores == sorted(scores, reverse=True)          # score order kept
+    assert all(d.text.strip() for d in sel)                # no empty docs
+
+
+# ---- chunker invariants ----
+
+from sentio_amd.ingest.chunker import TextChunker
+
+
+@given(st.text(alphabet=st.characters(whitelist_categories=("Ll", "Zs"),
+                                      max_codepoint=0x7A), max_size=800),
+       st.integers(min_value=16, max_value=128))
+@settings(max_examples=60, deadline=None)
+def test_chunker_size_bounds_and_coverage(text, size):
+    ch = TextChunker(chunk_size=size, chunk_overlap=size // 4)
+    doc = Document(text=text, metadata={}, id="src")
+    chunks = ch.split([doc])
+    joined = "".join(c.text for c in chunks)
+    # every chunk respects the size bound and carries lineage
+    for c in chunks:
+        assert len(c.text) <= size
+        assert c.metadata.get("parent_id") == "src"
+    # no content is lost: every non-space char of the input appears in
+    # the concatenation at least as often as in the source (chunk overlap
+    # may duplicate; only whitespace may be trimmed at boundaries)
+    for c0 in set(text.replace(" ", "")):
+        assert joined.count(c0) >= text.count(c0)
+    if text.strip():
+        assert chunks
