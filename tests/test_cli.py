@@ -57,3 +57,25 @@ def test_cli_help_lists_subcommands():
     assert r.exit_code == 0
     for sub in ("ingest", "api", "bench", "index", "chat", "run"):
         assert sub in r.output
+
+
+def test_cli_api_start_and_run_invoke_uvicorn(monkeypatch):
+    """`api start` and `run` build the app and hand it to uvicorn with the
+    resolved host/port (uvicorn itself is stubbed)."""
+    import uvicorn
+    from typer.testing import CliRunner
+
+    from sentio_amd.cli import app as cli_app
+
+    calls = []
+    monkeypatch.setenv("MOCK_COMPUTE", "true")
+    monkeypatch.setenv("SENTIO_DEVICE", "cpu")
+    monkeypatch.setattr(uvicorn, "run",
+                        lambda a, host, port: calls.append((host, port)))
+    runner = CliRunner()
+    r1 = runner.invoke(cli_app, ["api", "start", "--port", "9911"])
+    assert r1.exit_code == 0, r1.output
+    r2 = runner.invoke(cli_app, ["run", "--host", "127.0.0.1", "--port", "9912"])
+    assert r2.exit_code == 0, r2.output
+    assert ("UI at /ui") in r2.output
+    assert calls == [("0.0.0.0", 9911), ("127.0.0.1", 9912)]
