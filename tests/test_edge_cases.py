@@ -134,3 +134,30 @@ def test_graphed_bucket_batch():
     assert bucket_batch(3) == 4
     assert bucket_batch(33) == 64
     assert bucket_batch(100) == 100       # beyond largest bucket: unbucketed
+
+
+def test_generator_edge_lengths():
+    """max_new_tokens=1, empty prompt, and prompt at the budget edge all
+    return without error on the CPU reference path."""
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    g = GeneratorEngine("tiny-decoder64", device="cpu", dtype="fp32",
+                        max_seq=64)
+    out = g.generate(["hi"], max_new_tokens=1, temperature=0.0)
+    assert len(out) == 1
+    out = g.generate([""], max_new_tokens=4, temperature=0.0)
+    assert len(out) == 1
+    long_prompt = "x" * 4096      # far beyond max_seq — must truncate
+    out = g.generate([long_prompt], max_new_tokens=4, temperature=0.0)
+    assert len(out) == 1
+    assert g.generate([], max_new_tokens=4) == []
+
+
+def test_generator_temperature_zero_deterministic():
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    g = GeneratorEngine("tiny-decoder64", device="cpu", dtype="fp32",
+                        max_seq=64)
+    a = g.generate(["same prompt"], max_new_tokens=8, temperature=0.0)[0]
+    b = g.generate(["same prompt"], max_new_tokens=8, temperature=0.0)[0]
+    assert a == b
