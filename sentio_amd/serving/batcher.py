@@ -77,6 +77,15 @@ class DynamicBatcher:
         if self._thread is not None:
             self._thread.join(timeout=2.0)
             self._thread = None
+        # fail anything still queued so callers unblock immediately
+        # instead of waiting out their own timeout
+        while True:
+            try:
+                item = self._q.get_nowait()
+            except queue.Empty:
+                break
+            if not item.future.done():
+                item.future.set_exception(RuntimeError("batcher stopped"))
 
     def _loop(self) -> None:
         while not self._stop.is_set():

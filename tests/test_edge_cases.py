@@ -3,6 +3,8 @@ empty-input paths."""
 
 from __future__ import annotations
 
+import time
+
 import torch
 
 from sentio_amd.index import fusion
@@ -161,3 +163,40 @@ def test_generator_temperature_zero_deterministic():
     a = g.generate(["same prompt"], max_new_tokens=8, temperature=0.0)[0]
     b = g.generate(["same prompt"], max_new_tokens=8, temperature=0.0)[0]
     assert a == b
+
+
+def test_batcher_stop_fails_queued_requests():
+    """stop() must unblock waiting callers with an error, not leave them
+    hanging until their own timeout."""
+    import threading
+
+    from sentio_amd.serving.batcher import DynamicBatcher
+
+    started = threading.Event()
+
+    class _SlowGen:
+        def generate(self, prompts, **kw):
+            started.set()
+            time.sleep(0.3)
+            return ["ok"] * len(prompts)
+
+    b = DynamicBatcher(_SlowGen(), max_batch=1, max_wait_ms=1.0)
+    results = {}
+
+    def call(i):
+        try:
+            results[i] = b.generate(f"p{i}", max_new_tokens=4, timeout_s=5.0)
+        except Exception as e:
+            results[i] = e
+
+    threads = [threading.Thread(target=call, args=(i,)) for i in range(3)]
+    for t in threads:
+        t.start()
+    started.wait(2.0)          # first batch is in the slow generate
+    b.stop()                   # queued (not-yet-started) items must fail fast
+    for t in threads:
+        t.join(timeout=6.0)
+    assert len(results) == 3
+    vals = list(results.values())
+    assert any(v == "ok" for v in vals)                      # in-flight one finished
+    assert any(isinstance(v, RuntimeError) for v in vals)    # queued ones failed
