@@ -71,9 +71,11 @@ class TextChunker:
                 if cur.strip():
                     chunks.append(cur.strip())
                 tail = cur[-self.chunk_overlap:] if self.chunk_overlap else ""
-                cur = (tail + p)[-max(len(tail + p), 0):]
+                cur = tail + p
                 if len(cur) > self.chunk_size:
-                    cur = cur[: self.chunk_size]
+                    # overlap is best-effort: trim carried-over tail chars
+                    # from the FRONT, never the new piece's content
+                    cur = cur[-self.chunk_size:]
         if cur.strip():
             chunks.append(cur.strip())
         return chunks
