@@ -302,3 +302,40 @@ def test_chunker_size_bounds_and_coverage(text, size):
         assert joined.count(c0) >= text.count(c0)
     if text.strip():
         assert chunks
+
+
+# ---- auth token tamper resistance ----
+
+from sentio_amd.utils.auth import AuthError, AuthManager, UserRole
+
+
+@given(st.integers(min_value=0, max_value=10_000), st.integers(min_value=0, max_value=61))
+@settings(max_examples=100, deadline=None)
+def test_auth_token_any_single_char_tamper_rejected(seed, pos):
+    am = AuthManager()
+    token = am.issue_token(f"user{seed % 7}", UserRole.READER)
+    p = pos % len(token)
+    # flip one character to a different base64url character
+    repl = "A" if token[p] != "A" else "B"
+    tampered = token[:p] + repl + token[p + 1:]
+    if tampered == token:
+        return
+    try:
+        am.verify_token(tampered)
+        assert False, f"tampered token accepted (pos {p})"
+    except AuthError:
+        pass
+
+
+@given(st.text(max_size=64))
+@settings(max_examples=60, deadline=None)
+def test_auth_garbage_tokens_rejected(garbage):
+    am = AuthManager()
+    real = am.issue_token("u", UserRole.READER)
+    if garbage == real:
+        return
+    try:
+        am.verify_token(garbage)
+        assert False, "garbage accepted"
+    except AuthError:
+        pass
