@@ -85,10 +85,19 @@ class AuthManager:
             body, sig = token.rsplit(".", 1)
         except ValueError:
             raise AuthError("malformed token")
-        if not hmac.compare_digest(self._sign(body.encode()), sig):
+        try:
+            ok = hmac.compare_digest(self._sign(body.encode()), sig)
+        except (TypeError, UnicodeError):
+            # non-ASCII signature chars: invalid by construction — must be
+            # an auth failure (401), never an unhandled 500
+            raise AuthError("malformed token")
+        if not ok:
             raise AuthError("bad signature")
         pad = "=" * (-len(body) % 4)
-        claims = json.loads(base64.urlsafe_b64decode(body + pad))
+        try:
+            claims = json.loads(base64.urlsafe_b64decode(body + pad))
+        except (ValueError, UnicodeError):
+            raise AuthError("malformed token body")
         if time.time() > float(claims.get("exp", 0)):
             raise AuthError("token expired")
         return TokenData(
