@@ -389,3 +389,43 @@ def test_dense_metadata_filter_only_matching(n_docs, top_k, want_topic):
     assert len(hits) <= top_k
     scores = [s for _, s in hits]
     assert scores == sorted(scores, reverse=True)
+
+
+# ---- input validator: never crashes, always neutralizes markup ----
+
+from sentio_amd.utils.security import InputValidator, ValidationError
+
+
+@given(st.text(max_size=3000))
+@settings(max_examples=120, deadline=None)
+def test_validate_query_never_crashes_and_escapes(text):
+    try:
+        out = InputValidator.validate_query(text)
+    except ValidationError:
+        return                      # rejection is a valid outcome
+    # accepted queries carry no active-content patterns (screen-and-reject
+    # design: the API is JSON, so escaping would corrupt legitimate text)
+    low = out.lower()
+    assert "<script" not in low.replace(" ", "")
+    assert "javascript:" not in low.replace(" ", "")
+    assert len(out) <= 2000
+    assert out == out.strip()
+
+
+# ---- disk cache: arbitrary keys can never escape the cache directory ----
+
+@given(st.text(min_size=1, max_size=80))
+@settings(max_examples=60, deadline=None)
+def test_disk_cache_keys_stay_inside_dir(tmp_path_factory, key):
+    import os
+
+    from sentio_amd.caching.disk import DiskCache
+
+    base = tmp_path_factory.mktemp("dc")
+    dc = DiskCache(str(base))
+    dc.set(key, {"v": 1})
+    assert dc.get(key) == {"v": 1}
+    for root, _dirs, files in os.walk(str(base)):
+        for f in files:
+            full = os.path.realpath(os.path.join(root, f))
+            assert full.startswith(os.path.realpath(str(base)))
