@@ -429,3 +429,32 @@ def test_disk_cache_keys_stay_inside_dir(tmp_path_factory, key):
         for f in files:
             full = os.path.realpath(os.path.join(root, f))
             assert full.startswith(os.path.realpath(str(base)))
+
+
+# ---- Document dict roundtrip (the cross-rank payload-fetch wire format) ----
+
+@given(st.text(max_size=200),
+       st.dictionaries(st.text(min_size=1, max_size=10),
+                       st.one_of(st.text(max_size=20), st.integers(),
+                                 st.floats(allow_nan=False), st.booleans()),
+                       max_size=5))
+@settings(max_examples=60, deadline=None)
+def test_document_dict_roundtrip(text, meta):
+    d = Document(text=text, metadata=meta, id="fixed-id")
+    d2 = Document.from_dict(d.to_dict())
+    assert d2.text == d.text
+    assert d2.metadata == d.metadata
+    assert d2.id == d.id
+
+
+# ---- comb_sum fusion bounds ----
+
+@given(_hitlist, _hitlist)
+@settings(max_examples=60, deadline=None)
+def test_comb_sum_scores_bounded(dense, sparse):
+    fused = fuse(dense, sparse, method="comb_sum", top_k=50,
+                 dense_weight=0.7, sparse_weight=0.3)
+    for _id, s in fused:
+        assert -1e-6 <= s <= 1.0 + 1e-6     # minmax-normalized weighted sum
+    ids = [i for i, _ in fused]
+    assert len(ids) == len(set(ids))
