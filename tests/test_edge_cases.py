@@ -173,14 +173,15 @@ def test_batcher_stop_fails_queued_requests():
     from sentio_amd.serving.batcher import DynamicBatcher
 
     started = threading.Event()
+    gate = threading.Event()   # holds the in-flight batch until we release it
 
-    class _SlowGen:
+    class _GatedGen:
         def generate(self, prompts, **kw):
             started.set()
-            time.sleep(0.3)
+            gate.wait(5.0)
             return ["ok"] * len(prompts)
 
-    b = DynamicBatcher(_SlowGen(), max_batch=1, max_wait_ms=1.0)
+    b = DynamicBatcher(_GatedGen(), max_batch=1, max_wait_ms=1.0)
     results = {}
 
     def call(i):
@@ -192,11 +193,19 @@ def test_batcher_stop_fails_queued_requests():
     threads = [threading.Thread(target=call, args=(i,)) for i in range(3)]
     for t in threads:
         t.start()
-    started.wait(2.0)          # first batch is in the slow generate
-    b.stop()                   # queued (not-yet-started) items must fail fast
+    assert started.wait(2.0)   # batch 1 is inside generate, gated
+    deadline = time.monotonic() + 2.0
+    while b._q.qsize() < 2 and time.monotonic() < deadline:
+        time.sleep(0.005)      # the other two requests are now queued
+    assert b._q.qsize() == 2
+    stopper = threading.Thread(target=b.stop)
+    stopper.start()            # sets _stop, then blocks joining the worker
+    time.sleep(0.05)
+    gate.set()                 # in-flight batch completes; worker exits;
+    stopper.join(timeout=6.0)  # stop() drains + fails the 2 queued items
     for t in threads:
         t.join(timeout=6.0)
     assert len(results) == 3
     vals = list(results.values())
-    assert any(v == "ok" for v in vals)                      # in-flight one finished
-    assert any(isinstance(v, RuntimeError) for v in vals)    # queued ones failed
+    assert sum(v == "ok" for v in vals) == 1                   # in-flight served
+    assert sum(isinstance(v, RuntimeError) for v in vals) == 2  # queued failed
