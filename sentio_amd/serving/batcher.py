@@ -52,6 +52,7 @@ class DynamicBatcher:
         self._q: queue.Queue[_Item] = queue.Queue()
         self._stop = threading.Event()
         self._thread: threading.Thread | None = None
+        self._start_lock = threading.Lock()
         self.stats = {"requests": 0, "batches": 0, "max_batch_seen": 0}
 
     # ----- caller side -----
@@ -65,12 +66,15 @@ class DynamicBatcher:
 
     # ----- worker side -----
     def start(self) -> None:
-        if self._thread is not None and self._thread.is_alive():
-            return
-        self._stop.clear()
-        self._thread = threading.Thread(target=self._loop, daemon=True,
-                                        name="sentio-batcher")
-        self._thread.start()
+        # locked check-then-act: concurrent FIRST requests must not each
+        # spawn a worker — multiple workers would fragment batching
+        with self._start_lock:
+            if self._thread is not None and self._thread.is_alive():
+                return
+            self._stop.clear()
+            self._thread = threading.Thread(target=self._loop, daemon=True,
+                                            name="sentio-batcher")
+            self._thread.start()
 
     def stop(self) -> None:
         self._stop.set()

@@ -178,7 +178,7 @@ def test_batcher_stop_fails_queued_requests():
     class _GatedGen:
         def generate(self, prompts, **kw):
             started.set()
-            gate.wait(5.0)
+            gate.wait(60.0)
             return ["ok"] * len(prompts)
 
     b = DynamicBatcher(_GatedGen(), max_batch=1, max_wait_ms=1.0)
@@ -186,23 +186,25 @@ def test_batcher_stop_fails_queued_requests():
 
     def call(i):
         try:
-            results[i] = b.generate(f"p{i}", max_new_tokens=4, timeout_s=5.0)
+            results[i] = b.generate(f"p{i}", max_new_tokens=4, timeout_s=30.0)
         except Exception as e:
             results[i] = e
 
     threads = [threading.Thread(target=call, args=(i,)) for i in range(3)]
     for t in threads:
         t.start()
-    assert started.wait(2.0)   # batch 1 is inside generate, gated
-    deadline = time.monotonic() + 2.0
+    assert started.wait(30.0)  # batch 1 is inside generate, gated
+    deadline = time.monotonic() + 30.0
     while b._q.qsize() < 2 and time.monotonic() < deadline:
         time.sleep(0.005)      # the other two requests are now queued
     assert b._q.qsize() == 2
-    stopper = threading.Thread(target=b.stop)
-    stopper.start()            # sets _stop, then blocks joining the worker
-    time.sleep(0.05)
-    gate.set()                 # in-flight batch completes; worker exits;
-    stopper.join(timeout=6.0)  # stop() drains + fails the 2 queued items
+    # strict ordering, no scheduler-dependent sleeps: flag the stop BEFORE
+    # releasing the gate, so the worker exits right after batch 1
+    b._stop.set()
+    gate.set()
+    b._thread.join(timeout=6.0)
+    assert not b._thread.is_alive()
+    b.stop()                   # drains + fails the 2 still-queued items
     for t in threads:
         t.join(timeout=6.0)
     assert len(results) == 3
