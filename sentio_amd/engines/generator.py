@@ -37,12 +37,15 @@ class _DecodeSession:
         self.use_graphs = use_graphs
         self.batch = batch
 
-    def prefill(self, tokens: torch.Tensor) -> torch.Tensor:
+    def prefill(self, tokens: torch.Tensor,
+                lens: torch.Tensor | None = None) -> torch.Tensor:
         self.cache.seq_lens.zero_()
-        return self.engine.model.prefill(tokens, self.cache)
+        return self.engine.model.prefill(tokens, self.cache, lens=lens)
 
     def prefill_with_prefix(self, prefix_ids: list[int],
-                            suffix_tokens: torch.Tensor) -> torch.Tensor:
+                            suffix_tokens: torch.Tensor,
+                            suffix_lens: torch.Tensor | None = None
+                            ) -> torch.Tensor:
         """Copy the (cached) prefix KV into every batch slot, then prefill
         only the suffix (prefix-KV caching: the shared system-prompt +
         instruction header is prefilled ONCE per prefix, not per request)."""
@@ -52,7 +55,8 @@ class _DecodeSession:
         for i in range(engine.cfg.n_layers):
             self.cache.k[i][:, :, :P].copy_(pk[i])   # [1,...] broadcasts
             self.cache.v[i][:, :, :P].copy_(pv[i])
-        return engine.model.prefill_suffix(suffix_tokens, self.cache, P)
+        return engine.model.prefill_suffix(suffix_tokens, self.cache, P,
+                                           suffix_lens=suffix_lens)
 
     def _capture(self):
         model = self.engine.model
@@ -116,21 +120,24 @@ class GeneratorEngine:
                              ) -> tuple[list[int], list[str]]:
         """Split the batch's longest common prompt prefix from the per-prompt
         suffixes such that prefix_ids + encode(suffix, add_bos=False) is
-        EXACTLY encode(full prompt) for every prompt (byte tokenizer:
-        chars align 1:1 with ids after the single BOS)."""
+        EXACTLY encode(full prompt) for every prompt.  The cap is computed in
+        TOKENS via tokenizer.prefix_split — a char-based cap overshoots the
+        budget for multi-byte UTF-8 prefixes (len(prefix_ids) > chars) and
+        crashed prefill with negative suffix budgets."""
         import os as _os
 
         if len(clipped) < 2:
             return [], clipped
         prefix_txt = _os.path.commonprefix(clipped)
-        # leave >= 1 suffix char per prompt; keep room for generation
-        P_chars = min(len(prefix_txt), prompt_budget - 9,
-                      min(len(c) for c in clipped) - 1)
-        if P_chars <= 0:
+        # leave >= 1 suffix char per prompt and >= 8 suffix tokens of budget
+        limit_chars = min(len(prefix_txt), min(len(c) for c in clipped) - 1)
+        if limit_chars <= 0 or prompt_budget <= 9:
             return [], clipped
-        prefix_ids = self.tokenizer.encode(prefix_txt[:P_chars], None)
-        suffixes = [c[P_chars:] for c in clipped]
-        return prefix_ids, suffixes
+        n, prefix_ids = self.tokenizer.prefix_split(
+            prefix_txt[:limit_chars], prompt_budget - 8)
+        if n <= 0 or not prefix_ids:
+            return [], clipped
+        return prefix_ids, [c[n:] for c in clipped]
 
     def _prefix_kv(self, prefix_ids: tuple):
         """Per-prefix KV, computed once with a batch-1 forward and kept in
@@ -235,19 +242,21 @@ class GeneratorEngine:
                 suffixes, prompt_budget - P, add_bos=False)
             tokens = torch.tensor(padded, dtype=torch.int64,
                                   device=self.device)
+            lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
             sess.cache.seq_lens.zero_()
-            logits = sess.prefill_with_prefix(prefix_ids, tokens)
-            S = P + tokens.shape[1]
+            logits = sess.prefill_with_prefix(prefix_ids, tokens, lens_t)
+            n_prompt = B * P + sum(lens)
         else:
             padded, lens = self.tokenizer.encode_batch(clipped, prompt_budget)
             tokens = torch.tensor(padded, dtype=torch.int64,
                                   device=self.device)
-            logits = sess.prefill(tokens)
-            S = tokens.shape[1]
+            lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
+            logits = sess.prefill(tokens, lens_t)
+            n_prompt = sum(lens)
         if self.device != "cpu":
             torch.cuda.synchronize()
         self.last_prefill_s = _time.perf_counter() - _t0
-        self.last_prompt_tokens = B * S
+        self.last_prompt_tokens = n_prompt
         _t0 = _time.perf_counter()
 
         # tokens accumulate on-device; the host syncs only for EOS checks
@@ -277,7 +286,7 @@ class GeneratorEngine:
             from sentio_amd.observability.metrics import metrics_collector
 
             metrics_collector.inc("rag_llm_tokens_total",
-                                  float(B * S), kind="prompt")
+                                  float(n_prompt), kind="prompt")
             metrics_collector.inc("rag_llm_tokens_total",
                                   float(B * n_steps), kind="completion")
         except Exception:

@@ -65,3 +65,21 @@ class ByteTokenizer:
 
     def count_tokens(self, text: str) -> int:
         return 1 + len(text.encode("utf-8"))
+
+    def prefix_split(self, text: str, max_tokens: int) -> tuple[int, list[int]]:
+        """Largest char count n such that encode(text[:n]) fits in max_tokens
+        ids AND encode(text[:n]) + encode(text[n:], add_bos=False) ==
+        encode(text) (split-exactness).  The budget is counted in TOKENS —
+        multi-byte UTF-8 chars produce several ids each, so a char-based cap
+        can overshoot a token budget.  Returns (n_chars, encode(text[:n]))."""
+        budget = max_tokens - 1  # BOS
+        if budget <= 0:
+            return 0, []
+        n = used = 0
+        for ch in text:
+            b = len(ch.encode("utf-8"))
+            if used + b > budget:
+                break
+            used += b
+            n += 1
+        return n, self.encode(text[:n], None)

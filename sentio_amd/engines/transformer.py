@@ -183,9 +183,12 @@ class Transformer:
             ).view(B, 1, H, hd)
         elif cache is not None and q_off > 0:
             # prefix-KV-cached suffix prefill: attend to cache rows
-            # [0, q_off + S) — the prefix KV was computed once and copied in
-            lens_abs = torch.full((B,), q_off + S, dtype=torch.int32,
-                                  device=x.device)
+            # [0, kv_lens[b]) — the prefix KV was computed once and copied
+            # in; kv_lens carries per-row ABSOLUTE lengths (prefix + true
+            # suffix length), masking right-pad keys out of shorter rows
+            lens_abs = (kv_lens if kv_lens is not None
+                        else torch.full((B,), q_off + S, dtype=torch.int32,
+                                        device=x.device))
             out = ops.attention_cache(q, cache.k[layer_idx],
                                       cache.v[layer_idx], lens_abs,
                                       q_off, self.scale)
@@ -300,43 +303,70 @@ class Transformer:
         return normed
 
     # ----- decoder-specific -----
-    def logits(self, hidden: torch.Tensor) -> torch.Tensor:
+    def logits(self, hidden: torch.Tensor,
+               last_idx: torch.Tensor | None = None) -> torch.Tensor:
+        """lm_head at each row's LAST REAL position.  last_idx [B] selects
+        per-row positions for right-padded batches — without it, shorter
+        prompts in a batch would be conditioned on their PAD tail."""
         B, S, d = hidden.shape
-        last = hidden[:, -1, :]
+        if last_idx is None:
+            last = hidden[:, -1, :]
+        else:
+            last = hidden[torch.arange(B, device=hidden.device),
+                          last_idx.to(hidden.device).long()]
         if B <= 64 and self.device != "cpu":
             return ops.lt_linear(last.contiguous(), self.w.lm_head).float()
         return torch.nn.functional.linear(last, self.w.lm_head).float()  # [B, V]
 
-    def prefill(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
-        """Prefill the cache; returns last-position logits [B, V].  Also
-        picks the decode-attention implementation for the upcoming decode
-        steps from the cache fill ratio (prompt length is host-side here —
-        no sync)."""
+    def prefill(self, tokens: torch.Tensor, cache: KVCache,
+                lens: torch.Tensor | None = None) -> torch.Tensor:
+        """Prefill the cache; returns last-real-position logits [B, V].
+        lens: per-row true prompt lengths for right-padded batches — masks
+        pad keys out of attention, starts decode at each row's own length,
+        and gathers logits at lens-1 (batch composition no longer changes a
+        request's output).  Also picks the decode-attention implementation
+        for the upcoming decode steps from the cache fill ratio (prompt
+        length is host-side here — no sync)."""
         B, S = tokens.shape
-        hidden = self.forward_hidden(tokens, cache=cache)
-        cache.seq_lens[:] = S
+        lens_i = None if lens is None else lens.to(torch.int32)
+        hidden = self.forward_hidden(tokens, cache=cache, kv_lens=lens_i)
+        if lens_i is None:
+            cache.seq_lens[:] = S
+            last_idx = None
+        else:
+            cache.seq_lens.copy_(lens_i)
+            last_idx = lens_i.long() - 1
         if self.device != "cpu":
             self.decode_attn_fn = self._decode_attn_impl(
                 S / max(cache.max_seq, 1))
-        return self.logits(hidden)
+        return self.logits(hidden, last_idx)
 
     def prefill_suffix(self, suffix_tokens: torch.Tensor, cache: KVCache,
-                       prefix_len: int) -> torch.Tensor:
+                       prefix_len: int,
+                       suffix_lens: torch.Tensor | None = None) -> torch.Tensor:
         """Prefill only the suffix against a cache whose rows [0, prefix_len)
         already hold the shared prefix's KV (prefix-KV caching).  Returns
-        last-position logits."""
+        last-real-position logits; suffix_lens: per-row true suffix lengths
+        for right-padded suffix batches."""
         B, S = suffix_tokens.shape
         pos = (prefix_len + torch.arange(S, device=suffix_tokens.device)
                ).unsqueeze(0).expand(B, S)
+        lens_abs = (None if suffix_lens is None
+                    else (suffix_lens.to(torch.int32) + prefix_len))
         hidden = self.forward_hidden(suffix_tokens, pos=pos, cache=cache,
-                                     q_off=prefix_len)
-        cache.seq_lens[:] = prefix_len + S
+                                     q_off=prefix_len, kv_lens=lens_abs)
+        if lens_abs is None:
+            cache.seq_lens[:] = prefix_len + S
+            last_idx = None
+        else:
+            cache.seq_lens.copy_(lens_abs)
+            last_idx = suffix_lens.long() - 1
         if self.device != "cpu":
             self.decode_attn_fn = self._decode_attn_impl(
                 (prefix_len + S) / max(cache.max_seq, 1))
         else:
             self.decode_attn_fn = None
-        return self.logits(hidden)
+        return self.logits(hidden, last_idx)
 
     def decode_step(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:
         """tokens: [B, 1] the latest sampled token; returns logits [B, V]."""

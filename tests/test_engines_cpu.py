@@ -152,6 +152,45 @@ def test_prefix_split_token_exactness():
         assert pre_ids + suf_ids == full
 
 
+def test_prefix_split_nonascii_token_budget():
+    """The prefix cap is a TOKEN budget: multi-byte UTF-8 prefixes must not
+    overshoot it (a char-based cap yielded 548 prefix tokens from a 200-token
+    budget and crashed prefill — ADVICE r1 high)."""
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
+    prefix = "共有システムプリアンブル。質問に答えてください。" * 8   # 3 B/char
+    prompts = [prefix + f"質問{i}?" for i in range(3)]
+    budget = 200
+    pre_ids, suffixes = eng._split_shared_prefix(prompts, budget)
+    assert len(pre_ids) <= budget - 8
+    for p, suf in zip(prompts, suffixes):
+        full = eng.tokenizer.encode(p, None)
+        suf_ids = eng.tokenizer.encode(suf, None, add_bos=False)
+        assert pre_ids + suf_ids == full
+        assert len(suf_ids) >= 1
+    # and the full generate() path survives a CJK shared prefix
+    out = eng.generate(prompts, max_new_tokens=4, temperature=0.0,
+                       stop_on_eos=False)
+    assert len(out) == 3
+
+
+def test_generate_batch_composition_invariance():
+    """A request's greedy output must not depend on its co-batched partners'
+    prompt lengths: logits are gathered at each row's true last position and
+    pad keys are masked."""
+    g = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
+    prompts = ["tell me about graphs",
+               "a considerably longer prompt about distributed retrieval "
+               "engines and how their sharded indexes merge candidates",
+               "hi"]
+    batched = g.generate(prompts, max_new_tokens=6, temperature=0.0,
+                         stop_on_eos=False)
+    singles = [g.generate([p], max_new_tokens=6, temperature=0.0,
+                          stop_on_eos=False)[0] for p in prompts]
+    assert batched == singles
+
+
 def test_model_config_param_counts():
     """Config shapes actually correspond to the named model classes."""
     from sentio_amd.engines.configs import MODEL_CONFIGS
