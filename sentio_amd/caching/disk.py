@@ -7,9 +7,10 @@ same interface, same compression threshold, survives process restarts
 
 from __future__ import annotations
 
+import base64
 import hashlib
+import json
 import os
-import pickle
 import threading
 import time
 import zlib
@@ -20,12 +21,60 @@ _MAGIC_Z = b"SZ1"
 _MAGIC_P = b"SP1"
 
 
+def _encode(v: Any) -> Any:
+    """Tagged-JSON encoding of the typed cache payloads (response dicts,
+    embedding vectors).  Deliberately NOT pickle: cache files must never be
+    able to execute code in the serving process (a writable cache dir would
+    otherwise be an RCE vector — ADVICE r1)."""
+    import numpy as np
+    import torch
+
+    if isinstance(v, torch.Tensor):
+        v = v.detach().cpu().numpy()
+    if isinstance(v, np.ndarray):
+        return {"__nd__": base64.b64encode(v.tobytes()).decode(),
+                "dtype": str(v.dtype), "shape": list(v.shape)}
+    if isinstance(v, (bytes, bytearray)):
+        return {"__b__": base64.b64encode(bytes(v)).decode()}
+    if isinstance(v, dict):
+        return {k: _encode(x) for k, x in v.items()}
+    if isinstance(v, (list, tuple)):
+        return [_encode(x) for x in v]
+    if isinstance(v, (str, int, float, bool)) or v is None:
+        return v
+    if isinstance(v, (np.integer,)):
+        return int(v)
+    if isinstance(v, (np.floating,)):
+        return float(v)
+    raise TypeError(f"DiskCache cannot serialize {type(v).__name__}")
+
+
+def _decode(v: Any) -> Any:
+    import numpy as np
+
+    if isinstance(v, dict):
+        if "__nd__" in v and "dtype" in v:
+            return np.frombuffer(
+                base64.b64decode(v["__nd__"]), dtype=np.dtype(v["dtype"])
+            ).reshape(v["shape"]).copy()
+        if "__b__" in v and len(v) == 1:
+            return base64.b64decode(v["__b__"])
+        return {k: _decode(x) for k, x in v.items()}
+    if isinstance(v, list):
+        return [_decode(x) for x in v]
+    return v
+
+
 class DiskCache:
     def __init__(self, directory: str | None = None, default_ttl: float = 3600.0,
                  max_entries: int = 100_000):
         self.dir = directory or os.path.join(
             os.path.expanduser("~"), ".cache", "sentio_amd", "l2")
-        os.makedirs(self.dir, exist_ok=True)
+        os.makedirs(self.dir, mode=0o700, exist_ok=True)
+        try:
+            os.chmod(self.dir, 0o700)   # pre-existing dir: tighten anyway
+        except OSError:
+            pass
         self.default_ttl = default_ttl
         self.max_entries = max_entries
         self._lock = threading.Lock()
@@ -54,12 +103,17 @@ class DiskCache:
             blob = zlib.decompress(blob[3:])
         elif blob[:3] == _MAGIC_P:
             blob = blob[3:]
-        return pickle.loads(blob)
+        else:
+            return None   # unknown/legacy (pickle-era) format: treat as miss
+        try:
+            return _decode(json.loads(blob.decode()))
+        except (ValueError, UnicodeDecodeError):
+            return None
 
     def set(self, key: str, value: Any, ttl: float | None = None) -> None:
         p = self._path(key)
-        os.makedirs(os.path.dirname(p), exist_ok=True)
-        blob = pickle.dumps(value)
+        os.makedirs(os.path.dirname(p), mode=0o700, exist_ok=True)
+        blob = json.dumps(_encode(value)).encode()
         if len(blob) > _COMPRESS_MIN:
             blob = _MAGIC_Z + zlib.compress(blob)
         else:

@@ -38,9 +38,10 @@ class _Item:
 
     @property
     def group_key(self) -> tuple:
-        # temperature bucketed to 0.1 — sampling is per-batch in the engine
-        return (round(self.temperature, 1), self.max_new_tokens,
-                self.stop_on_eos)
+        # EXACT temperature: sampling is per-batch in the engine, so a
+        # bucketed key would silently run a request at batch[0]'s
+        # temperature (up to ~0.05 off, arrival-order dependent)
+        return (self.temperature, self.max_new_tokens, self.stop_on_eos)
 
 
 class DynamicBatcher:
