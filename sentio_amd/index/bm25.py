@@ -201,6 +201,44 @@ class BM25Index:
         idx = idx.cpu().tolist()
         return [(self.doc_ids[i], float(v)) for i, v in zip(idx, vals) if v > 0.0]
 
+    def search_rows(self, query: str, top_k: int, device: str = "cpu"):
+        """Tensor-only top-k: (scores [k] f32, rows [k] i64) on `device`,
+        padded with (-inf, -1); zero/negative scores are padded out
+        (reference sparse.py:183 keeps score>0 only).  No host sync on the
+        GPU path — the sharded merge all-gathers these tensors directly."""
+        import torch
+
+        if self.n_docs == 0:
+            return (torch.full((top_k,), float("-inf"), device=device),
+                    torch.full((top_k,), -1, dtype=torch.int64, device=device))
+        if device != "cpu":
+            from sentio_amd import ops
+
+            if not self._device_arrays:
+                self.to_device(device)
+            tids = torch.from_numpy(self.query_term_ids(query)).to(device)
+            a = self._device_arrays
+            scores = ops.bm25_score(
+                tids, a["indptr"], a["post_doc"], a["post_tf"], a["idf"],
+                a["doc_len"], n_docs=self.n_docs, k1=self.k1, b=self.b,
+                avgdl=max(self.avgdl, 1e-9),
+                plus_delta=self.delta if self.variant == "plus" else 0.0)
+        else:
+            scores = torch.from_numpy(self.get_scores(query))
+        k = min(top_k, self.n_docs)
+        vals, idx = torch.topk(scores, k)
+        idx = idx.long()
+        dead = vals <= 0.0
+        vals = vals.float().masked_fill(dead, float("-inf"))
+        idx = idx.masked_fill(dead, -1)
+        if k < top_k:
+            vals = torch.cat([vals, torch.full((top_k - k,), float("-inf"),
+                                               device=vals.device)])
+            idx = torch.cat([idx, torch.full((top_k - k,), -1,
+                                             dtype=torch.int64,
+                                             device=idx.device)])
+        return vals, idx
+
     # ----- persistence (reference sparse.py:102-157) -----
     def save(self, path: str) -> None:
         state = {k: getattr(self, k) for k in (

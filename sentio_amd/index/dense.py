@@ -122,8 +122,47 @@ class DenseIndex:
             out.append(hits)
         return out
 
+    def search_rows(self, query: torch.Tensor, top_k: int
+                    ) -> tuple[torch.Tensor, torch.Tensor]:
+        """Tensor-only batched cosine top-k: (scores [B,k] f32, rows [B,k]
+        i64), padded with (-inf, -1) when the shard holds fewer than top_k
+        rows.  NO host sync — the multi-GPU merge all-gathers these tensors
+        directly (no pickled objects on the query hot path)."""
+        if query.ndim == 1:
+            query = query.unsqueeze(0)
+        B = query.shape[0]
+        if self._size == 0:
+            return (torch.full((B, top_k), float("-inf"), device=self.device),
+                    torch.full((B, top_k), -1, dtype=torch.int64,
+                               device=self.device))
+        k = min(top_k, self._size)
+        q = query.to(self.device, torch.float32)
+        q = q / q.norm(dim=1, keepdim=True).clamp_min(1e-12)
+        if self.device != "cpu":
+            from sentio_amd import ops
+
+            vals, idx = ops.cosine_topk(q.to(self.dtype),
+                                        self._vecs[: self._size], k)
+        else:
+            scores = q @ self._vecs[: self._size].T.float()
+            vals, idx = torch.topk(scores, k, dim=1)
+        vals = vals.float()
+        idx = idx.long()
+        if k < top_k:
+            vals = torch.cat([vals, torch.full((B, top_k - k), float("-inf"),
+                                               device=vals.device)], 1)
+            idx = torch.cat([idx, torch.full((B, top_k - k), -1,
+                                             dtype=torch.int64,
+                                             device=idx.device)], 1)
+        return vals, idx
+
     def get_document(self, doc_id: str) -> Document | None:
         return self._docs.get(doc_id)
+
+    def get_document_by_row(self, row: int) -> Document | None:
+        if 0 <= row < self._size and self.doc_ids:
+            return self._docs.get(self.doc_ids[row])
+        return None
 
     def clear(self) -> None:
         with self._lock:

@@ -90,8 +90,60 @@ def broadcast_object(obj, src: int = 0):
 
 
 def all_gather_objects(obj) -> list:
+    """Pickled object all-gather — ONLY for cold paths (init, tests).  The
+    query hot path uses tensor collectives (all_gather_tensor) and
+    exchange_bytes; see VERDICT r1 item 3."""
     if not is_distributed():
         return [obj]
     out = [None] * get_world_size()
     dist.all_gather_object(out, obj)
     return out
+
+
+def collective_device() -> str:
+    """Device the active backend can reduce on: NCCL(=RCCL) moves GPU
+    tensors over xGMI; gloo wants CPU tensors."""
+    if dist.is_initialized() and dist.get_backend() == "nccl":
+        return f"cuda:{torch.cuda.current_device()}"
+    return "cpu"
+
+
+def exchange_bytes(to_send: dict[int, bytes]) -> dict[int, bytes]:
+    """Targeted pairwise byte exchange: rank r receives exactly the payloads
+    other ranks addressed to it (point-to-point send/recv — xGMI is p2p, so
+    this is the native shape; replaces the O(W^2) object broadcast).
+    Collective: EVERY rank must call it (with {} if it has nothing to send).
+    Returns {src_rank: payload}."""
+    if not is_distributed():
+        return {}
+    world, rank = get_world_size(), get_rank()
+    dev = collective_device()
+    # size matrix via one tiny all-gather: sizes[r][p] = bytes r sends to p
+    sizes = torch.zeros(world, dtype=torch.int64, device=dev)
+    for p, b in to_send.items():
+        if p != rank:
+            sizes[p] = len(b)
+    all_sizes = [torch.zeros_like(sizes) for _ in range(world)]
+    dist.all_gather(all_sizes, sizes)
+    reqs = []
+    recv_bufs: dict[int, torch.Tensor] = {}
+    for src in range(world):
+        if src == rank:
+            continue
+        n = int(all_sizes[src][rank])
+        if n > 0:
+            buf = torch.empty(n, dtype=torch.uint8, device=dev)
+            recv_bufs[src] = buf
+            reqs.append(dist.irecv(buf, src=src))
+    total = 0
+    for dst, b in to_send.items():
+        if dst == rank or not b:
+            continue
+        t = torch.frombuffer(bytearray(b), dtype=torch.uint8).to(dev)
+        reqs.append(dist.isend(t, dst=dst))
+        total += len(b)
+    for r in reqs:
+        r.wait()
+    _account("p2p_exchange", total)
+    return {src: bytes(buf.cpu().numpy().tobytes())
+            for src, buf in recv_bufs.items()}

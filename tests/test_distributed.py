@@ -83,6 +83,14 @@ def _make_docs():
 def _sharded_dense_worker(rank, world):
     from sentio_amd.index.dense import DenseIndex
     from sentio_amd.parallel.shard import ShardedIndex
+    import sentio_amd.parallel.dist as PD
+
+    # the query hot path must be tensor collectives only — no pickled
+    # object gathers
+    def _no_objects(*a, **k):
+        raise AssertionError("pickled object collective on the query hot path")
+
+    PD.all_gather_objects = _no_objects
 
     docs, vecs = _make_docs()
     mine = [i for i in range(N_DOCS) if i % world == rank]
@@ -101,20 +109,27 @@ def _sharded_dense_worker(rank, world):
     ref.add(docs, torch.from_numpy(vecs))
     want = ref.search(q, 5)
 
+    # refs are "shard:d<row>" handles — resolve payloads through the
+    # targeted p2p fetch (ONE collective call per rank) and compare doc ids
+    all_refs = [r for qi in range(2) for r, _ in got[qi]]
+    resolved = sharded.fetch_documents(all_refs)
+    assert len(resolved) == len(set(all_refs))
     for qi in range(2):
-        got_ids = [r.split(":", 1)[1] for r, _ in got[qi]]
+        got_ids = [resolved[r].id for r, _ in got[qi]]
         want_ids = [i for i, _ in want[qi]]
         assert got_ids == want_ids, (rank, got_ids, want_ids)
         got_scores = [s for _, s in got[qi]]
         want_scores = [s for _, s in want[qi]]
         np.testing.assert_allclose(got_scores, want_scores, rtol=1e-4)
 
-    # payload fetch across shards
-    refs = [r for r, _ in got[0][:3]]
-    resolved = sharded.fetch_documents(refs)
-    assert len(resolved) == len(refs)
-    for ref_key, doc in resolved.items():
-        assert doc.id == ref_key.split(":", 1)[1]
+    # comm byte accounting saw the tensor gathers + p2p payload exchange
+    from sentio_amd.observability.metrics import metrics_collector
+
+    counters = metrics_collector.snapshot().get("counters", {})
+    assert any("rccl_bytes_total" in k and "all_gather" in k
+               for k in counters), counters
+    assert any("rccl_bytes_total" in k and "p2p_exchange" in k
+               for k in counters), counters
 
 
 def test_sharded_dense_search_matches_single_index():

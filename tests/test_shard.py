@@ -33,7 +33,7 @@ def test_sharded_index_single_rank_matches_local():
     sharded = ShardedIndex(dense, bm)
     q = vecs[3].unsqueeze(0)
     hits = sharded.search_dense(q, top_k=3)[0]
-    assert hits[0][0] == "0:doc3"        # shard-prefixed ref, self-match first
+    assert hits[0][0] == "0:d3"          # shard-prefixed row ref, self-match first
     s_hits = sharded.search_sparse("alpha beta", top_k=5)
     assert s_hits and all(ref.startswith("0:") for ref, _ in s_hits)
     fetched = sharded.fetch_documents([hits[0][0]])
