@@ -36,6 +36,59 @@ class HybridRetriever(BaseRetriever):
         self.cache_retriever = cache_retriever
         self.cache_score_threshold = cache_score_threshold
 
+    # "host" | "device" — which fusion path the last retrieve() took
+    # (GPU tests assert the K4 kernel really runs on the /chat path)
+    last_fusion_path: str = "none"
+
+    def _index_device(self) -> str:
+        idx = getattr(self.dense, "index", None)
+        return str(getattr(idx, "device", "cpu"))
+
+    def _fuse(self, dense_hits, sparse_hits, top_k: int):
+        """Candidate fusion.  When the dense index is GPU-resident the K4
+        fuse_topk kernel runs on device (one launch, one sync — the /chat
+        hot path); the host fusion.py oracle is the CPU path and the
+        semantics reference."""
+        import torch
+
+        dev = self._index_device()
+        n_union = len(dense_hits) + len(sparse_hits)
+        if (dev != "cpu" and torch.cuda.is_available()
+                and 0 < n_union <= 128 and top_k <= 128):
+            from sentio_amd import ops
+
+            interned: dict[str, int] = {}
+
+            def ref(doc_id: str) -> int:
+                return interned.setdefault(doc_id, len(interned))
+
+            Kd, Ks = max(len(dense_hits), 1), max(len(sparse_hits), 1)
+            d_i = torch.full((1, Kd), -1, dtype=torch.int64)
+            d_s = torch.zeros(1, Kd)
+            for j, (i, s) in enumerate(dense_hits):
+                d_i[0, j] = ref(i)
+                d_s[0, j] = s
+            s_i = torch.full((1, Ks), -1, dtype=torch.int64)
+            s_s = torch.zeros(1, Ks)
+            for j, (i, s) in enumerate(sparse_hits):
+                s_i[0, j] = ref(i)
+                s_s[0, j] = s
+            out_i, out_s = ops.fuse_topk(
+                d_i.to(dev), d_s.to(dev), s_i.to(dev), s_s.to(dev),
+                method=self.fusion_method, top_k=min(top_k, 128),
+                rrf_k=self.rrf_k, dense_weight=self.dense_weight,
+                sparse_weight=self.sparse_weight)
+            back = {v: k for k, v in interned.items()}
+            self.last_fusion_path = "device"
+            return [(back[int(i)], float(s))
+                    for i, s in zip(out_i[0].cpu().tolist(),
+                                    out_s[0].cpu().tolist()) if int(i) >= 0]
+        self.last_fusion_path = "host"
+        return fusion.fuse(
+            dense_hits, sparse_hits, method=self.fusion_method, top_k=top_k,
+            rrf_k=self.rrf_k, dense_weight=self.dense_weight,
+            sparse_weight=self.sparse_weight)
+
     def retrieve(self, query: str, top_k: int = 10) -> list[Document]:
         cache_docs: list[Document] = []
         if self.cache_retriever is not None:
@@ -63,12 +116,8 @@ class HybridRetriever(BaseRetriever):
         dense_hits = [(d.id, float(d.metadata.get("score", 0.0))) for d in dense_docs]
         sparse_hits = [(d.id, float(d.metadata.get("bm25_score", 0.0))) for d in sparse_docs]
 
-        fused = fusion.fuse(
-            dense_hits, sparse_hits,
-            method=self.fusion_method, top_k=max(top_k, len(dense_hits) + len(sparse_hits)),
-            rrf_k=self.rrf_k, dense_weight=self.dense_weight,
-            sparse_weight=self.sparse_weight,
-        )
+        fused = self._fuse(dense_hits, sparse_hits,
+                           top_k=max(top_k, len(dense_hits) + len(sparse_hits)))
 
         id_to_doc: dict[str, Document] = {}
         for d in dense_docs:

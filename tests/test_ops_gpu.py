@@ -470,6 +470,11 @@ def test_full_pipeline_with_verifier_gpu(dev):
     assert out["answer"]
     v = out["metadata"].get("verification")
     assert v is not None and v.get("verdict") in ("pass", "warn", "fail")
+    # K4 runs ON DEVICE on the /chat hot path:
+    # the hybrid retriever must have fused through the fuse_topk kernel
+    retr = c.retriever()
+    assert getattr(retr, "last_fusion_path", None) == "device", \
+        getattr(retr, "last_fusion_path", None)
 
 
 def test_lt_gemm_tn_matches_linear(dev):
