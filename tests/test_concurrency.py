@@ -117,3 +117,73 @@ def test_concurrent_cache_manager_thread_safety(client):
         assert all(ex.map(churn, range(8)))
     stats = cm.l1.stats()
     assert stats["hits"] + stats["misses"] >= 8 * 200
+
+
+def test_stream_does_not_block_concurrent_chat():
+    """A slow streaming consumer must not starve other generations: streams
+    join batched decode through the DynamicBatcher (tokens buffer in a
+    per-request queue) instead of holding the engine lock for the stream's
+    life."""
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.serving.batcher import BatchedGenerator
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=128)
+    gen = BatchedGenerator(eng, max_batch=4, max_wait_ms=5.0)
+    try:
+        stream_it = gen.stream("a very long streaming prompt about graphs",
+                               max_new_tokens=32, temperature=0.0)
+        first = next(stream_it)          # stream is live, NOT drained
+        assert isinstance(first, str) and first
+
+        # while the stream iterator sits un-drained, a chat request must
+        # still complete promptly through the shared engine
+        import time as _t
+
+        t0 = _t.monotonic()
+        out = gen.generate(["quick concurrent chat"], max_new_tokens=4,
+                           temperature=0.0)
+        assert out[0] is not None
+        assert _t.monotonic() - t0 < 30.0
+
+        rest = "".join(stream_it)        # drain afterwards: stream completed
+        assert first + rest
+        # and the streamed text matches the engine's own output for the
+        # same prompt (greedy)
+        want = eng.generate(["a very long streaming prompt about graphs"],
+                            max_new_tokens=32, temperature=0.0)[0]
+        assert (first + rest) == want
+    finally:
+        gen.batcher.stop()
+
+
+def test_stream_and_chat_coalesce_into_one_batch():
+    """A stream and a same-params chat arriving together share ONE engine
+    batch (continuous-batching-lite)."""
+    import queue as _q
+    import threading as _th
+
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.serving.batcher import BatchedGenerator
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=128)
+    gen = BatchedGenerator(eng, max_batch=4, max_wait_ms=200.0)
+    try:
+        out_q: _q.Queue = _q.Queue()
+
+        def chat():
+            out_q.put(gen.generate(["chat prompt"], max_new_tokens=8,
+                                   temperature=0.0, stop_on_eos=True)[0])
+
+        def stream():
+            out_q.put("".join(gen.stream("stream prompt", max_new_tokens=8,
+                                         temperature=0.0)))
+
+        t1 = _th.Thread(target=chat)
+        t2 = _th.Thread(target=stream)
+        t1.start(); t2.start()
+        t1.join(timeout=60); t2.join(timeout=60)
+        assert out_q.qsize() == 2
+        st = gen.batcher.stats
+        assert st["max_batch_seen"] >= 2, st   # they coalesced
+    finally:
+        gen.batcher.stop()
