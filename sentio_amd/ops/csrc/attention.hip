@@ -5,9 +5,11 @@
 //    (row-major D-stride tiles are an up-to-16-way bank conflict — guide §6
 //    G4), V staged transposed so the PV B-fragment is a contiguous
 //    ds_read_b128, P round-trips through padded LDS for the C→A relayout.
-//  * decode_attn — single-token decode: one block per (batch, head),
-//    chunked online softmax over the contiguous KV cache; score phase is
-//    thread-per-key, PV phase is thread-per-dim (coalesced V reads).
+//  * decode_attn — single-token decode: one block per (batch, kv-head,
+//    split), chunked online softmax over the contiguous KV cache; score
+//    phase is thread-per-key with K rows read DIRECT from HBM into
+//    registers (all D/8 16-byte loads in flight; the G-head reuse happens
+//    on the register copy), PV phase is thread-per-dim (coalesced V reads).
 //
 // Replaces (K6 prefill/decode in SURVEY §2.3) the reference's remote
 // /chat/completions calls (reference src/core/llm/providers/openai.py:117).
@@ -345,7 +347,6 @@ __global__ void decode_attn_split_kernel(
   float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
   float* red = q_sh + G * D;                               // [32] scratch
   float* o_sh = red + 32;                                  // [G][D] final reduce
-  char* k_lds = reinterpret_cast<char*>(o_sh + G * D);     // [DEC_CHUNK][D] bf16 swizzled
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
@@ -385,82 +386,36 @@ __global__ void decode_attn_split_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) o_part[g][e] = 0.f;
 
-  const int Dbytes = D * 2;
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    // phase A1: cooperative coalesced K staging into XOR-swizzled LDS.
-    // One predicated unrolled path for EVERY chunk (a separate dynamic tail
-    // loop serialized one load per s_waitcnt and dominated short spans):
-    // 8 loads in flight before any ds_write; out-of-range rows stage zeros.
-#ifdef SENTIO_DECODE_GLDS   // EXPERIMENTAL, known-broken numerics (kept for
-                             // study): raw global_load_lds staging miscompiles
-                             // or misdrains here — 3 GPU tests fail, no perf
-                             // gain measured; default register path below
-    {
-      const int lim = chunk * D;
-#pragma unroll
-      for (int u = 0; u < D / 8; ++u) {
-        const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
-        if (i < lim) {
-          const int row = i / D, d = i % D;
-          auto gp = (const __attribute__((address_space(1))) unsigned int*)(
-              kb + (long)(s0 + row) * D + d);
-          auto lp = (__attribute__((address_space(3))) unsigned int*)(
-              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
-          __builtin_amdgcn_global_load_lds(gp, lp, 16, 0, 0);
-        }
-      }
-    }
-#else
-    {
-      constexpr int IT = D / 8;          // loads per thread for a full chunk
-      constexpr int BATCH = 8;           // loads in flight
-      const int lim = chunk * D;
-#pragma unroll
-      for (int u0 = 0; u0 < IT; u0 += BATCH) {
-        bf16x8 tmp[BATCH];
-#pragma unroll
-        for (int u = 0; u < BATCH; ++u) {
-          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
-          if (i < lim) {
-            tmp[u] = nt_load8(
-                reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
-                + (i % D));
-          } else {
-            bf16x8 z = {};
-            tmp[u] = z;
-          }
-        }
-#pragma unroll
-        for (int u = 0; u < BATCH; ++u) {
-          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
-          const int row = i / D, d = i % D;
-          *reinterpret_cast<bf16x8*>(
-              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = tmp[u];
-        }
-      }
-    }
-#endif
-    __syncthreads();
-    // phase A2: thread-per-key dot vs all G query heads, K read from LDS;
-    // scores stay in registers (the softmax below is the same thread).
+    // phase A: thread-per-key dot with K read DIRECT from HBM — each K
+    // element is consumed by exactly one thread (the G-head reuse happens
+    // on the register copy), so the old LDS staging round-trip
+    // (global→reg→ds_write→barrier→ds_read) was pure overhead.  A thread
+    // streams its own 256 B row with all D/8 16-byte loads in flight
+    // (consecutive threads read consecutive rows — wave-coalesced).
     float sc[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) sc[g] = -INFINITY;
     {
       const int row = threadIdx.x;
       if (row < chunk) {
+        constexpr int NL = D / 8;        // 16 B loads per key row
+        const short* kp =
+            reinterpret_cast<const short*>(kb + (long)(s0 + row) * D);
+        bf16x8 kreg[NL];
+#pragma unroll
+        for (int u = 0; u < NL; ++u) kreg[u] = nt_load8(kp + u * 8);
 #pragma unroll
         for (int g = 0; g < G; ++g) sc[g] = 0.f;
-        for (int d = 0; d < D; d += 8) {
-          bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
-              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
+#pragma unroll
+        for (int u = 0; u < NL; ++u) {
 #pragma unroll
           for (int g = 0; g < G; ++g) {
             float acc = 0.f;
 #pragma unroll
             for (int e = 0; e < 8; ++e)
-              acc += bits2f(k8[e]) * q_sh[g * D + d + e];
+              acc += bits2f(kreg[u][e]) * q_sh[g * D + u * 8 + e];
             sc[g] += acc;
           }
         }
@@ -651,8 +606,7 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
   if (D > 256 || (D % 8)) return hipErrorInvalidValue;
   const int G = H / Hkv;
   if (G > DEC_MAXG || H % Hkv) return hipErrorInvalidValue;
-  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float)
-               + (size_t)DEC_CHUNK * D * 2;   // swizzled K stage
+  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float);
   dim3 grid(Hkv, B, splits);
   if (D != 64 && D != 128) return hipErrorInvalidValue;
 #define DEC_CASE(GV, DV)                                                      \
