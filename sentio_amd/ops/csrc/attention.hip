@@ -347,6 +347,7 @@ __global__ void decode_attn_split_kernel(
   float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
   float* red = q_sh + G * D;                               // [32] scratch
   float* o_sh = red + 32;                                  // [G][D] final reduce
+  char* k_lds = reinterpret_cast<char*>(o_sh + G * D);     // [DEC_CHUNK][D] bf16 swizzled
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
@@ -386,36 +387,68 @@ __global__ void decode_attn_split_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) o_part[g][e] = 0.f;
 
+  // T14 software-pipelined K staging (guide §6): chunk kt's K rows ride in
+  // registers while chunk kt-1 computes from LDS — without this the HBM
+  // pipe idled through the dot + both softmax reductions every chunk.
+  // Loads are wave-coalesced (thread t's element i = t*8 + u*2048:
+  // consecutive lanes read consecutive 16 B → 4 KB/instruction); the
+  // swizzled LDS image turns them into conflict-free per-key row reads
+  // (a DIRECT thread-per-key load pattern was measured at 1.9 TB/s —
+  // 64 lanes hitting 64 rows 256 B apart).
+  constexpr int KIT = D / 8;             // staging loads per thread
+  const int Dbytes = D * 2;
+  bf16x8 kst[KIT];
+  auto load_k = [&](int s0_, int chunk_) {
+    const int lim = chunk_ * D;
+#pragma unroll
+    for (int u = 0; u < KIT; ++u) {
+      const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
+      if (i < lim) {
+        kst[u] = nt_load8(
+            reinterpret_cast<const short*>(kb + (long)(s0_ + i / D) * D) +
+            (i % D));
+      } else {
+        bf16x8 z = {};
+        kst[u] = z;
+      }
+    }
+  };
+  auto store_k = [&]() {
+#pragma unroll
+    for (int u = 0; u < KIT; ++u) {
+      const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
+      const int row = i / D, d = i % D;
+      *reinterpret_cast<bf16x8*>(
+          k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = kst[u];
+    }
+  };
+
+  if (s_begin < s_end) load_k(s_begin, min(DEC_CHUNK, s_end - s_begin));
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    // phase A: thread-per-key dot with K read DIRECT from HBM — each K
-    // element is consumed by exactly one thread (the G-head reuse happens
-    // on the register copy), so the old LDS staging round-trip
-    // (global→reg→ds_write→barrier→ds_read) was pure overhead.  A thread
-    // streams its own 256 B row with all D/8 16-byte loads in flight
-    // (consecutive threads read consecutive rows — wave-coalesced).
+    store_k();
+    __syncthreads();                     // staging visible to all waves
+    const int s1 = s0 + DEC_CHUNK;
+    if (s1 < s_end) load_k(s1, min(DEC_CHUNK, s_end - s1));  // in flight
+    // phase A: thread-per-key dot vs all G query heads, K read from LDS;
+    // scores stay in registers (the softmax below is the same thread).
     float sc[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) sc[g] = -INFINITY;
     {
       const int row = threadIdx.x;
       if (row < chunk) {
-        constexpr int NL = D / 8;        // 16 B loads per key row
-        const short* kp =
-            reinterpret_cast<const short*>(kb + (long)(s0 + row) * D);
-        bf16x8 kreg[NL];
-#pragma unroll
-        for (int u = 0; u < NL; ++u) kreg[u] = nt_load8(kp + u * 8);
 #pragma unroll
         for (int g = 0; g < G; ++g) sc[g] = 0.f;
-#pragma unroll
-        for (int u = 0; u < NL; ++u) {
+        for (int d = 0; d < D; d += 8) {
+          bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
+              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
 #pragma unroll
           for (int g = 0; g < G; ++g) {
             float acc = 0.f;
 #pragma unroll
             for (int e = 0; e < 8; ++e)
-              acc += bits2f(kreg[u][e]) * q_sh[g * D + u * 8 + e];
+              acc += bits2f(k8[e]) * q_sh[g * D + d + e];
             sc[g] += acc;
           }
         }
@@ -606,7 +639,8 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
   if (D > 256 || (D % 8)) return hipErrorInvalidValue;
   const int G = H / Hkv;
   if (G > DEC_MAXG || H % Hkv) return hipErrorInvalidValue;
-  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float);
+  size_t lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float)
+               + (size_t)DEC_CHUNK * D * 2;   // swizzled K stage
   dim3 grid(Hkv, B, splits);
   if (D != 64 && D != 128) return hipErrorInvalidValue;
 #define DEC_CASE(GV, DV)                                                      \
