@@ -5,11 +5,9 @@
 //    (row-major D-stride tiles are an up-to-16-way bank conflict — guide §6
 //    G4), V staged transposed so the PV B-fragment is a contiguous
 //    ds_read_b128, P round-trips through padded LDS for the C→A relayout.
-//  * decode_attn — single-token decode: one block per (batch, kv-head,
-//    split), chunked online softmax over the contiguous KV cache; score
-//    phase is thread-per-key with K rows read DIRECT from HBM into
-//    registers (all D/8 16-byte loads in flight; the G-head reuse happens
-//    on the register copy), PV phase is thread-per-dim (coalesced V reads).
+//  * decode_attn — single-token decode: one block per (batch, head),
+//    chunked online softmax over the contiguous KV cache; score phase is
+//    thread-per-key, PV phase is thread-per-dim (coalesced V reads).
 //
 // Replaces (K6 prefill/decode in SURVEY §2.3) the reference's remote
 // /chat/completions calls (reference src/core/llm/providers/openai.py:117).
@@ -348,6 +346,11 @@ __global__ void decode_attn_split_kernel(
   float* red = q_sh + G * D;                               // [32] scratch
   float* o_sh = red + 32;                                  // [G][D] final reduce
   char* k_lds = reinterpret_cast<char*>(o_sh + G * D);     // [DEC_CHUNK][D] bf16 swizzled
+  // typed 16-byte-element view: keeps every staging/dot ds op a full
+  // ds_write_b128/ds_read_b128 — char* + XOR addressing hid the 16 B
+  // alignment from the compiler, which fragmented the dot phase into
+  // b96/b64/b32 pieces with serial lgkmcnt waits (seen in ISA)
+  bf16x8* k_vec = reinterpret_cast<bf16x8*>(k_lds);
 
   const int hkv = blockIdx.x;
   const int b = blockIdx.y;
@@ -387,50 +390,42 @@ __global__ void decode_attn_split_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) o_part[g][e] = 0.f;
 
-  // T14 software-pipelined K staging (guide §6): chunk kt's K rows ride in
-  // registers while chunk kt-1 computes from LDS — without this the HBM
-  // pipe idled through the dot + both softmax reductions every chunk.
-  // Loads are wave-coalesced (thread t's element i = t*8 + u*2048:
-  // consecutive lanes read consecutive 16 B → 4 KB/instruction); the
-  // swizzled LDS image turns them into conflict-free per-key row reads
-  // (a DIRECT thread-per-key load pattern was measured at 1.9 TB/s —
-  // 64 lanes hitting 64 rows 256 B apart).
-  constexpr int KIT = D / 8;             // staging loads per thread
   const int Dbytes = D * 2;
-  bf16x8 kst[KIT];
-  auto load_k = [&](int s0_, int chunk_) {
-    const int lim = chunk_ * D;
-#pragma unroll
-    for (int u = 0; u < KIT; ++u) {
-      const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
-      if (i < lim) {
-        kst[u] = nt_load8(
-            reinterpret_cast<const short*>(kb + (long)(s0_ + i / D) * D) +
-            (i % D));
-      } else {
-        bf16x8 z = {};
-        kst[u] = z;
-      }
-    }
-  };
-  auto store_k = [&]() {
-#pragma unroll
-    for (int u = 0; u < KIT; ++u) {
-      const int i = threadIdx.x * 8 + u * DEC_CHUNK * 8;
-      const int row = i / D, d = i % D;
-      *reinterpret_cast<bf16x8*>(
-          k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4))) = kst[u];
-    }
-  };
-
-  if (s_begin < s_end) load_k(s_begin, min(DEC_CHUNK, s_end - s_begin));
   for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
     const int chunk = min(DEC_CHUNK, s_end - s0);
-    store_k();
-    __syncthreads();                     // staging visible to all waves
-    const int s1 = s0 + DEC_CHUNK;
-    if (s1 < s_end) load_k(s1, min(DEC_CHUNK, s_end - s1));  // in flight
-    // phase A: thread-per-key dot vs all G query heads, K read from LDS;
+    // phase A1: cooperative coalesced K staging into XOR-swizzled LDS.
+    // One predicated unrolled path for EVERY chunk (a separate dynamic tail
+    // loop serialized one load per s_waitcnt and dominated short spans):
+    // 8 loads in flight before any ds_write; out-of-range rows stage zeros.
+    {
+      constexpr int IT = D / 8;          // loads per thread for a full chunk
+      constexpr int BATCH = 8;           // loads in flight
+      const int lim = chunk * D;
+#pragma unroll
+      for (int u0 = 0; u0 < IT; u0 += BATCH) {
+        bf16x8 tmp[BATCH];
+#pragma unroll
+        for (int u = 0; u < BATCH; ++u) {
+          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
+          if (i < lim) {
+            tmp[u] = nt_load8(
+                reinterpret_cast<const short*>(kb + (long)(s0 + i / D) * D)
+                + (i % D));
+          } else {
+            bf16x8 z = {};
+            tmp[u] = z;
+          }
+        }
+#pragma unroll
+        for (int u = 0; u < BATCH; ++u) {
+          const int i = threadIdx.x * 8 + (u0 + u) * DEC_CHUNK * 8;
+          const int row = i / D, d = i % D;
+          k_vec[((row * Dbytes + d * 2) ^ ((row & 7) << 4)) >> 4] = tmp[u];
+        }
+      }
+    }
+    __syncthreads();
+    // phase A2: thread-per-key dot vs all G query heads, K read from LDS;
     // scores stay in registers (the softmax below is the same thread).
     float sc[G];
 #pragma unroll
@@ -441,8 +436,7 @@ __global__ void decode_attn_split_kernel(
 #pragma unroll
         for (int g = 0; g < G; ++g) sc[g] = 0.f;
         for (int d = 0; d < D; d += 8) {
-          bf16x8 k8 = *reinterpret_cast<const bf16x8*>(
-              k_lds + ((row * Dbytes + d * 2) ^ ((row & 7) << 4)));
+          bf16x8 k8 = k_vec[((row * Dbytes + d * 2) ^ ((row & 7) << 4)) >> 4];
 #pragma unroll
           for (int g = 0; g < G; ++g) {
             float acc = 0.f;
@@ -458,7 +452,12 @@ __global__ void decode_attn_split_kernel(
     }
     // issue ALL of this thread's V loads for the chunk NOW — they do not
     // depend on the softmax, so the HBM pipe stays busy while the block
-    // reductions run (otherwise it idles through both reduction barriers)
+    // reductions run (otherwise it idles through both reduction barriers).
+    // (Measured dead ends kept out: half-depth V prefetch to save VGPRs
+    // lost 6% — LDS, not VGPRs, caps occupancy at 2 blocks/CU; a
+    // cross-chunk register-pipelined K prefetch lost 20% — the in-order
+    // vmcnt counter makes later V waits drain it, and the doubled VGPRs
+    // halved occupancy.)
     constexpr int JT = DEC_CHUNK / 16;
     bf16x8 v8[JT];
     if (dg_ok) {
