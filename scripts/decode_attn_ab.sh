@@ -1,7 +1,7 @@
 #!/bin/bash
 # A/B decode-attention kernel variants on one box
-for v in r1vec lowreg; do
-  cp variants/hip_$v.so sentio_amd/ops/_sentio_hip.so
+for v in "$@"; do
+  cp variants/hip_$v.so sentio_amd/ops/_sentio_hip.so  # variants/ is scratch: stage .so builds there before calling
   echo "=== variant $v ==="
   python - <<'PY'
 import os, time, torch, sys

@@ -40,7 +40,7 @@ def seed(url: str, n: int) -> None:
             r.raise_for_status()
 
 
-def run(url: str, clients: int, total: int) -> dict:
+def run(url: str, clients: int, total: int, stream_frac: float = 0.0) -> dict:
     import threading
 
     latencies: list[float] = []
@@ -48,11 +48,27 @@ def run(url: str, clients: int, total: int) -> dict:
     throttled = 0
     tls = threading.local()   # persistent connection per worker thread
 
+    n_stream = int(total * stream_frac)
+
     def one(i: int) -> float | str | None:
         c = getattr(tls, "client", None)
         if c is None:
             c = tls.client = _client(url)
         t0 = time.perf_counter()
+        if i < n_stream:
+            # SSE stream: consume every delta; latency = full stream drain
+            body = ""
+            with c.stream("POST", "/chat/stream",
+                          json={"question": f"what about topic {i % 7}?"}) as r:
+                if r.status_code == 429:
+                    return "throttled"
+                if r.status_code != 200:
+                    return None
+                for line in r.iter_lines():
+                    if line.startswith("data: "):
+                        body += line[6:]
+            dt = time.perf_counter() - t0
+            return dt if body else None
         r = c.post("/chat", json={"question": f"what about topic {i % 7}?"})
         dt = time.perf_counter() - t0
         if r.status_code == 429:      # server-side rate limit, not a failure
@@ -61,9 +77,13 @@ def run(url: str, clients: int, total: int) -> dict:
             return None
         return dt
 
+    import random
+
+    order = list(range(total))
+    random.Random(7).shuffle(order)   # interleave stream/non-stream arrivals
     t_start = time.perf_counter()
     with ThreadPoolExecutor(max_workers=clients) as ex:
-        for dt in ex.map(one, range(total)):
+        for dt in ex.map(one, order):
             if dt is None:
                 errors += 1
             elif dt == "throttled":
@@ -79,6 +99,7 @@ def run(url: str, clients: int, total: int) -> dict:
     return {
         "requests": total,
         "clients": clients,
+        "stream_requests": n_stream,
         "served": served,
         "throttled": throttled,
         "errors": errors,
@@ -98,13 +119,15 @@ def main() -> int:
     ap.add_argument("--clients", type=int, default=16)
     ap.add_argument("--requests", type=int, default=200)
     ap.add_argument("--seed-docs", type=int, default=0)
+    ap.add_argument("--stream-frac", type=float, default=0.0,
+                    help="fraction of requests sent to /chat/stream (SSE)")
     args = ap.parse_args()
 
     with _client(args.url) as c:
         c.get("/health").raise_for_status()
     if args.seed_docs:
         seed(args.url, args.seed_docs)
-    result = run(args.url, args.clients, args.requests)
+    result = run(args.url, args.clients, args.requests, args.stream_frac)
     print(json.dumps(result))
     return 0 if result["errors"] == 0 else 1
 
