@@ -246,11 +246,15 @@ class ServiceContainer:
                 self._cache.pop(k, None)
 
     def health(self) -> dict[str, Any]:
+        gen = self._cache.get("generator")
+        batcher = getattr(gen, "batcher", None)
         return {
             "device": self.device,
             "index_size": len(self.dense_index()),
             "bm25_docs": self.bm25_index().n_docs,
             "breakers": {k: b.health() for k, b in self.breakers.items()},
+            # dynamic-batcher coalescing stats (requests/batches/max seen)
+            "batcher": batcher.health() if batcher is not None else None,
             "initialized": sorted(self._cache.keys()),
         }
 
