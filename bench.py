@@ -380,7 +380,7 @@ def main():
             top_n = min(len(docs), 2 * args.rerank_top_k)
             spans.append((len(pair_texts), top_n))
             pair_texts.extend(f"{queries[qi]}\n{d.text}" for d in docs[:top_n])
-        scores = reranker.score_pairs("", pair_texts) if pair_texts else []
+        scores = reranker.score_packed(pair_texts) if pair_texts else []
         # NOTE: pair text already contains the query; score_pairs prefixes
         # query="" so the packed text is the pair.
         reranked = []

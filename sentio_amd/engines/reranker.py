@@ -43,13 +43,20 @@ class RerankerEngine:
 
     @torch.inference_mode()
     def score_pairs(self, query: str, texts: list[str], batch_size: int = 32) -> list[float]:
-        if not texts:
+        return self.score_packed([f"{query}\n{t}" for t in texts], batch_size)
+
+    @torch.inference_mode()
+    def score_packed(self, pair_texts_all: list[str],
+                     batch_size: int = 32) -> list[float]:
+        """Score pre-packed "query\\ntext" pairs — the batched entry point
+        (serving micro-batcher, bench) where each pair carries its own
+        query; identical packing to score_pairs."""
+        if not pair_texts_all:
             return []
         pool = self._graph_pool()
         scores: list[float] = []
-        for i in range(0, len(texts), batch_size):
-            chunk = texts[i : i + batch_size]
-            pair_texts = [f"{query}\n{t}" for t in chunk]
+        for i in range(0, len(pair_texts_all), batch_size):
+            pair_texts = pair_texts_all[i : i + batch_size]
             padded, lens = self.tokenizer.encode_batch(pair_texts, self.max_seq)
             tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
             kv_lens = torch.tensor(lens, dtype=torch.int32, device=self.device)

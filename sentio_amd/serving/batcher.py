@@ -188,6 +188,171 @@ class DynamicBatcher:
                 **self.stats}
 
 
+@dataclass
+class _MicroItem:
+    units: list
+    future: Future = field(default_factory=Future)
+
+
+class MicroBatcher:
+    """Coalesce concurrent small ENGINE calls (single-query embeds,
+    per-request rerank scores) into one batched forward.  The GPU load test
+    showed generation batching alone is not enough: 16 concurrent /chat
+    requests each ran their own tiny encoder/reranker forward between
+    generation batches, serializing ~0.8 s of retrieval per wave.  Each
+    submit() contributes a list of units; the worker concatenates queued
+    units, runs batch_fn(all_units) ONCE, and fans the per-item slices
+    back out."""
+
+    def __init__(self, batch_fn, max_units: int = 64, max_wait_ms: float = 3.0,
+                 name: str = "micro"):
+        self.batch_fn = batch_fn
+        self.max_units = max_units
+        self.max_wait_s = max_wait_ms / 1e3
+        self.name = name
+        self._q: queue.Queue[_MicroItem] = queue.Queue()
+        self._stop = threading.Event()
+        self._thread: threading.Thread | None = None
+        self._start_lock = threading.Lock()
+        self.stats = {"calls": 0, "batches": 0, "max_units_seen": 0}
+
+    def submit(self, units: list, timeout_s: float = 120.0):
+        """Blocking: returns batch_fn(flat)[i0:i1] for this item's units."""
+        self._ensure_worker()
+        item = _MicroItem(units)
+        self._q.put(item)
+        return item.future.result(timeout=timeout_s)
+
+    def _ensure_worker(self) -> None:
+        with self._start_lock:
+            if self._thread is not None and self._thread.is_alive():
+                return
+            self._stop.clear()
+            self._thread = threading.Thread(
+                target=self._loop, daemon=True,
+                name=f"sentio-micro-{self.name}")
+            self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=2.0)
+            self._thread = None
+        while True:
+            try:
+                item = self._q.get_nowait()
+            except queue.Empty:
+                break
+            if not item.future.done():
+                item.future.set_exception(RuntimeError("micro-batcher stopped"))
+
+    def _loop(self) -> None:
+        while not self._stop.is_set():
+            try:
+                first = self._q.get(timeout=0.1)
+            except queue.Empty:
+                continue
+            batch = [first]
+            n_units = len(first.units)
+            deadline = time.monotonic() + self.max_wait_s
+            while n_units < self.max_units:
+                remaining = deadline - time.monotonic()
+                if remaining <= 0:
+                    break
+                try:
+                    nxt = self._q.get(timeout=remaining)
+                except queue.Empty:
+                    break
+                batch.append(nxt)
+                n_units += len(nxt.units)
+            self._run(batch, n_units)
+
+    def _run(self, batch: list[_MicroItem], n_units: int) -> None:
+        self.stats["calls"] += len(batch)
+        self.stats["batches"] += 1
+        self.stats["max_units_seen"] = max(self.stats["max_units_seen"],
+                                           n_units)
+        try:
+            flat: list = []
+            for it in batch:
+                flat.extend(it.units)
+            res = self.batch_fn(flat)
+            off = 0
+            for it in batch:
+                it.future.set_result(res[off: off + len(it.units)])
+                off += len(it.units)
+        except Exception as exc:
+            for it in batch:
+                if not it.future.done():
+                    it.future.set_exception(exc)
+
+    def health(self) -> dict[str, Any]:
+        return {"queued": self._q.qsize(), **self.stats}
+
+
+class BatchedEncoder:
+    """Encoder frontend: small embed calls (a request's single query)
+    coalesce through a MicroBatcher; bulk calls (ingest, bench batches)
+    pass straight through."""
+
+    _BULK = 17   # > this many texts: caller already batches
+
+    def __init__(self, raw, max_units: int = 64, max_wait_ms: float = 3.0):
+        self.raw = raw
+        self.micro = MicroBatcher(lambda texts: raw.embed(texts),
+                                  max_units=max_units,
+                                  max_wait_ms=max_wait_ms, name="encoder")
+
+    def embed(self, texts, batch_size: int = 64):
+        if not texts or len(texts) >= self._BULK:
+            return self.raw.embed(texts, batch_size)
+        return self.micro.submit(list(texts))
+
+    def embed_one(self, text: str):
+        return self.embed([text])[0]
+
+    def __getattr__(self, name):
+        return getattr(self.raw, name)
+
+
+class BatchedReranker:
+    """Reranker frontend: concurrent requests' pair scores coalesce into
+    one cross-encoder forward (pair text carries its own query, so mixed
+    queries batch fine — same packing the bench uses)."""
+
+    _BULK = 33
+
+    def __init__(self, raw, max_units: int = 64, max_wait_ms: float = 3.0):
+        self.raw = raw
+        self.micro = MicroBatcher(
+            raw.score_packed,
+            max_units=max_units, max_wait_ms=max_wait_ms, name="reranker")
+
+    def score_pairs(self, query: str, texts: list[str],
+                    batch_size: int = 32) -> list[float]:
+        if not texts or len(texts) >= self._BULK:
+            return self.raw.score_pairs(query, texts, batch_size)
+        return list(self.micro.submit([f"{query}\n{t}" for t in texts]))
+
+    def rerank(self, query: str, docs, top_k: int):
+        if not docs:
+            return []
+        top_n = min(len(docs), 2 * top_k)
+        cand = docs[:top_n]
+        scores = self.score_pairs(query, [d.text for d in cand])
+        order = sorted(range(len(cand)), key=lambda i: scores[i], reverse=True)
+        out = []
+        for i in order[:top_k]:
+            d = cand[i]
+            d.metadata["rerank_score"] = float(scores[i])
+            d.metadata["score"] = float(scores[i])
+            out.append(d)
+        return out
+
+    def __getattr__(self, name):
+        return getattr(self.raw, name)
+
+
 class BatchedGenerator:
     """Drop-in generator frontend: single-prompt calls route through the
     shared DynamicBatcher; already-batched calls pass straight through."""

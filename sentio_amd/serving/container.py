@@ -57,10 +57,18 @@ class ServiceContainer:
                                          device=self.device)
             from sentio_amd.engines.encoder import EncoderEngine
 
-            return EncoderEngine(self.settings.encoder_model, device=self.device,
-                                 dtype=self.settings.compute_dtype,
-                                 cache_size=self.settings.embedding_cache_size,
-                                 cache_ttl=self.settings.embedding_cache_ttl_s)
+            eng = EncoderEngine(self.settings.encoder_model, device=self.device,
+                                dtype=self.settings.compute_dtype,
+                                cache_size=self.settings.embedding_cache_size,
+                                cache_ttl=self.settings.embedding_cache_ttl_s)
+            if self.settings.dynamic_batching:
+                # concurrent requests' single-query embeds coalesce into one
+                # encoder forward (the GPU load test showed per-request
+                # retrieval serializing between generation batches)
+                from sentio_amd.serving.batcher import BatchedEncoder
+
+                return BatchedEncoder(eng)
+            return eng
 
         return self._get("encoder", make)
 
@@ -72,8 +80,14 @@ class ServiceContainer:
                 return MockRerankerEngine()
             from sentio_amd.engines.reranker import RerankerEngine
 
-            return RerankerEngine(self.settings.reranker_model, device=self.device,
-                                  dtype=self.settings.compute_dtype)
+            eng = RerankerEngine(self.settings.reranker_model,
+                                 device=self.device,
+                                 dtype=self.settings.compute_dtype)
+            if self.settings.dynamic_batching:
+                from sentio_amd.serving.batcher import BatchedReranker
+
+                return BatchedReranker(eng)
+            return eng
 
         return self._get("reranker", make)
 
@@ -255,6 +269,11 @@ class ServiceContainer:
             "breakers": {k: b.health() for k, b in self.breakers.items()},
             # dynamic-batcher coalescing stats (requests/batches/max seen)
             "batcher": batcher.health() if batcher is not None else None,
+            "micro_batchers": {
+                k: getattr(self._cache.get(k), "micro", None).health()
+                for k in ("encoder", "reranker")
+                if getattr(self._cache.get(k), "micro", None) is not None
+            },
             "initialized": sorted(self._cache.keys()),
         }
 
