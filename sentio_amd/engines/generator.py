@@ -220,8 +220,26 @@ class GeneratorEngine:
         finally:
             self._gen_lock.release()
 
+    # Decode-session batch buckets: serving batches arrive at arbitrary
+    # sizes, and every DISTINCT (batch, cache) shape costs a fresh KV-cache
+    # allocation + hipGraph capture (~1 s) — the GPU load test spent more
+    # wall on captures than on decoding.  Padding rows are near-free
+    # (decode streams the same weights either way), so round up and slice.
+    _BATCH_BUCKETS = (1, 2, 4, 8, 16, 24, 32, 48, 64)
+
+    def _bucket_batch(self, b: int) -> int:
+        for cap in self._BATCH_BUCKETS:
+            if b <= cap:
+                return cap
+        return b
+
     def _generate_locked(self, prompts, max_new_tokens, temperature,
                          stop_on_eos, on_token, _time) -> list[str]:
+        B_req = len(prompts)
+        if self.device != "cpu":
+            pad = self._bucket_batch(B_req) - B_req
+            if pad:
+                prompts = list(prompts) + [prompts[-1]] * pad
         B = len(prompts)
         prompt_budget = self.max_seq - max_new_tokens - 1
         # prefix-KV caching: requests share the system-prompt + instruction
@@ -245,14 +263,14 @@ class GeneratorEngine:
             lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
             sess.cache.seq_lens.zero_()
             logits = sess.prefill_with_prefix(prefix_ids, tokens, lens_t)
-            n_prompt = B * P + sum(lens)
+            n_prompt = B_req * P + sum(lens[:B_req])
         else:
             padded, lens = self.tokenizer.encode_batch(clipped, prompt_budget)
             tokens = torch.tensor(padded, dtype=torch.int64,
                                   device=self.device)
             lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
             logits = sess.prefill(tokens, lens_t)
-            n_prompt = sum(lens)
+            n_prompt = sum(lens[:B_req])
         if self.device != "cpu":
             torch.cuda.synchronize()
         self.last_prefill_s = _time.perf_counter() - _t0
@@ -268,7 +286,7 @@ class GeneratorEngine:
         for step in range(max_new_tokens):
             toks_buf[:, step] = cur
             if on_token is not None:
-                on_token(step, cur.cpu().tolist())
+                on_token(step, cur.cpu().tolist()[:B_req])
             if stop_on_eos and (step % 16 == 15 or on_token is not None):
                 done = (toks_buf[:, : step + 1] == EOS_ID).any(dim=1)
                 if bool(done.all()):
@@ -288,10 +306,10 @@ class GeneratorEngine:
             metrics_collector.inc("rag_llm_tokens_total",
                                   float(n_prompt), kind="prompt")
             metrics_collector.inc("rag_llm_tokens_total",
-                                  float(B * n_steps), kind="completion")
+                                  float(B_req * n_steps), kind="completion")
         except Exception:
             pass
-        rows = toks_buf[:, :n_steps].cpu().tolist()
+        rows = toks_buf[:B_req, :n_steps].cpu().tolist()
         out = []
         for row in rows:
             ids = []
