@@ -244,6 +244,18 @@ class ServiceContainer:
         if self.settings.use_reranker:
             self.reranker()
         self.generator()
+        if self.device != "cpu" and not self.settings.mock_compute:
+            # pre-capture the decode hipGraphs for the serving batch
+            # buckets: a first-occurrence capture costs ~1 s, which the GPU
+            # load test otherwise pays mid-traffic (p95 blowout)
+            try:
+                gen = self.generator()
+                for b in (1, 2, 4, 8, 16, 24, 32, 48, 64):
+                    if b <= self.settings.max_batch_size:
+                        gen.generate([f"warm {i}" for i in range(b)],
+                                     max_new_tokens=2, temperature=0.0)
+            except Exception as exc:
+                logger.warning("generator warm-up failed: %s", exc)
         self.pipeline()
         self.ingestor()
         self.auth_manager()
