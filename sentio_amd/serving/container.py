@@ -228,21 +228,38 @@ class ServiceContainer:
                 self._cache.pop(k, None)
         return {"docs": len(self.dense_index())}
 
+    _WARM_BUCKETS = (1, 2, 4, 8, 16, 32, 64)
+
     def initialize_all(self) -> None:
         """Eager startup init in dependency order."""
         self.encoder()
+        on_gpu = self.device != "cpu" and not self.settings.mock_compute
         try:
-            # warm-up embed (reference warm_up_embeddings startup hook,
-            # embeddings/factory.py:122-137): triggers weight init + the
-            # first hipGraph capture before user traffic
-            self.encoder().embed(["warm up"])
+            # warm-up embeds (reference warm_up_embeddings startup hook,
+            # embeddings/factory.py:122-137).  On GPU: capture the hipGraph
+            # for EVERY batch bucket now, serially — a capture taken
+            # mid-traffic runs concurrently with decode-graph replays on
+            # other threads, which can wedge the HIP context (observed as a
+            # hung /chat/stream + every later CUDA call blocking)
+            enc = self.encoder()
+            raw = getattr(enc, "raw", enc)
+            for b in (self._WARM_BUCKETS if on_gpu else (1,)):
+                raw.embed([f"warm up {b}.{i}" for i in range(b)])
         except Exception as exc:
             logger.warning("encoder warm-up failed: %s", exc)
         self.dense_index()
         self.bm25_index()
         self.retriever()
         if self.settings.use_reranker:
-            self.reranker()
+            rr = self.reranker()
+            if on_gpu:
+                try:
+                    raw = getattr(rr, "raw", rr)
+                    for b in self._WARM_BUCKETS:
+                        raw.score_packed([f"warm\ndoc {b}.{i}"
+                                          for i in range(b)])
+                except Exception as exc:
+                    logger.warning("reranker warm-up failed: %s", exc)
         self.generator()
         if self.device != "cpu" and not self.settings.mock_compute:
             # pre-capture the decode hipGraphs for the serving batch

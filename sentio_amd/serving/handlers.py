@@ -134,6 +134,16 @@ class HealthHandler:
         checks["breakers"] = {k: b.health()["state"]
                               for k, b in self.container.breakers.items()}
         checks["heartbeat"] = self.container.heartbeat.snapshot()
+        # request/engine coalescing evidence (dynamic + micro batchers)
+        gen = self.container._cache.get("generator_frontend")
+        if gen is not None and hasattr(gen, "batcher"):
+            checks["batcher"] = gen.batcher.health()
+        checks["micro_batchers"] = {
+            k: m.health() for k, m in (
+                (k, getattr(self.container._cache.get(k), "micro", None))
+                for k in ("encoder", "reranker"))
+            if m is not None
+        }
         status = "healthy" if checks.get("encoder") == "ok" else "degraded"
         result = {
             "status": status,
