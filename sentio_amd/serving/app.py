@@ -243,7 +243,12 @@ def create_app(settings: Settings | None = None,
             builder = PromptBuilder(s.generation_mode)
             prompt = builder.system_prompt() + "\n\n" + builder.build_qa_prompt(
                 req.question, prepare_context(docs))
-            for delta in container.generator().stream(
+            # generator_frontend, NOT the raw engine: the raw stream()
+            # holds the engine's generation lock between SSE yields, so an
+            # abandoned/slow client (e.g. read timeout) leaks the lock and
+            # wedges every later generation.  The batched frontend streams
+            # from a per-request token queue instead.
+            for delta in container.generator_frontend().stream(
                     prompt, max_new_tokens=s.llm_max_tokens,
                     temperature=req.temperature or 0.3):
                 yield f"data: {delta}\n\n"

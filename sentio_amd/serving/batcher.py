@@ -89,8 +89,13 @@ class DynamicBatcher:
         emitted = ""
         deadline = time.monotonic() + timeout_s
         while True:
-            t = item.stream_q.get(timeout=max(0.1,
-                                              deadline - time.monotonic()))
+            try:
+                t = item.stream_q.get(timeout=max(0.1,
+                                                  deadline - time.monotonic()))
+            except queue.Empty:
+                raise TimeoutError(
+                    f"stream starved for {timeout_s:.0f}s (engine stalled "
+                    "or request never scheduled)") from None
             if t is None or t == EOS_ID:
                 break
             generated.append(t)
