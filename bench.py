@@ -541,6 +541,10 @@ def main():
                     sorted(step_times)[min(len(step_times) - 1,
                                            int(len(step_times) * 0.95))] * 1e3,
                     1) if step_times else None,
+                "prompt_tokens_per_req": round(
+                    getattr(generator, "last_prompt_tokens", 0)
+                    / max(args.batch, 1)),
+                "tokenizer": type(generator.tokenizer).__name__,
                 "init_s": round(init_s, 1),
                 "device": "cuda" if on_gpu else "cpu-plumbing",
             },

@@ -66,36 +66,36 @@ MODEL_CONFIGS: dict[str, ModelConfig] = {
     ),
     "tiny-decoder": ModelConfig(
         name="tiny-decoder", dim=64, n_layers=2, n_heads=4, n_kv_heads=2,
-        ffn_dim=128, vocab_size=512, max_seq=1024, rope_base=10000.0,
+        ffn_dim=128, vocab_size=4608, max_seq=1024, rope_base=10000.0,
     ),
     "tiny-decoder64": ModelConfig(  # head_dim 64 — GPU-kernel-compatible tiny
         name="tiny-decoder64", dim=256, n_layers=2, n_heads=4, n_kv_heads=2,
-        ffn_dim=512, vocab_size=512, max_seq=1024, rope_base=10000.0,
+        ffn_dim=512, vocab_size=4608, max_seq=1024, rope_base=10000.0,
     ),
     # --- encoders (jina-v3 class: 1024-dim output, reference jina.py:23-27) ---
     "sentio-encoder-base": ModelConfig(
         name="sentio-encoder-base", dim=1024, n_layers=24, n_heads=16,
-        n_kv_heads=16, ffn_dim=4096, vocab_size=512, max_seq=2048,
+        n_kv_heads=16, ffn_dim=4096, vocab_size=4608, max_seq=2048,
         rope_base=10000.0, causal=False,
     ),
     "sentio-encoder-small": ModelConfig(
         name="sentio-encoder-small", dim=1024, n_layers=6, n_heads=16,
-        n_kv_heads=16, ffn_dim=2048, vocab_size=512, max_seq=2048,
+        n_kv_heads=16, ffn_dim=2048, vocab_size=4608, max_seq=2048,
         rope_base=10000.0, causal=False,
     ),
     "tiny-encoder": ModelConfig(
         name="tiny-encoder", dim=64, n_layers=2, n_heads=4, n_kv_heads=4,
-        ffn_dim=128, vocab_size=512, max_seq=512, rope_base=10000.0, causal=False,
+        ffn_dim=128, vocab_size=4608, max_seq=512, rope_base=10000.0, causal=False,
     ),
     # --- rerankers (bge-reranker-base class) ---
     "sentio-reranker-base": ModelConfig(
         name="sentio-reranker-base", dim=768, n_layers=12, n_heads=12,
-        n_kv_heads=12, ffn_dim=3072, vocab_size=512, max_seq=1024,
+        n_kv_heads=12, ffn_dim=3072, vocab_size=4608, max_seq=1024,
         rope_base=10000.0, causal=False, pooled_head=1,
     ),
     "tiny-reranker": ModelConfig(
         name="tiny-reranker", dim=64, n_layers=2, n_heads=4, n_kv_heads=4,
-        ffn_dim=128, vocab_size=512, max_seq=512, rope_base=10000.0,
+        ffn_dim=128, vocab_size=4608, max_seq=512, rope_base=10000.0,
         causal=False, pooled_head=1,
     ),
 }

@@ -14,7 +14,7 @@ import torch
 
 from sentio_amd import ops
 from sentio_amd.engines.configs import get_model_config
-from sentio_amd.engines.tokenizer import ByteTokenizer
+from sentio_amd.engines.bpe import get_tokenizer
 from sentio_amd.engines.transformer import Transformer
 
 
@@ -25,7 +25,7 @@ class EncoderEngine:
         self.cfg = get_model_config(model)
         self.device = device
         self.max_seq = min(max_seq, self.cfg.max_seq)
-        self.tokenizer = ByteTokenizer()
+        self.tokenizer = get_tokenizer()
         self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
         self.dim = self.cfg.dim
         self.calls = 0

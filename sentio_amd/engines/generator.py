@@ -17,7 +17,8 @@ import torch
 
 from sentio_amd import ops
 from sentio_amd.engines.configs import get_model_config
-from sentio_amd.engines.tokenizer import EOS_ID, ByteTokenizer
+from sentio_amd.engines.bpe import get_tokenizer
+from sentio_amd.engines.tokenizer import EOS_ID
 from sentio_amd.engines.transformer import KVCache, Transformer
 
 MODE_TEMPERATURE = {"fast": 0.0, "balanced": 0.3, "quality": 0.2, "creative": 0.7}
@@ -101,7 +102,7 @@ class GeneratorEngine:
         self.cfg = get_model_config(model)
         self.device = device
         self.max_seq = min(max_seq, self.cfg.max_seq)
-        self.tokenizer = ByteTokenizer()
+        self.tokenizer = get_tokenizer()
         self.model = Transformer(self.cfg, device=device, dtype=dtype,
                                  seed=seed, tp=tp)
         self._step_seed = 0

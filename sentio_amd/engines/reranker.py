@@ -12,7 +12,7 @@ from __future__ import annotations
 import torch
 
 from sentio_amd.engines.configs import get_model_config
-from sentio_amd.engines.tokenizer import ByteTokenizer
+from sentio_amd.engines.bpe import get_tokenizer
 from sentio_amd.engines.transformer import Transformer
 from sentio_amd.models.document import Document
 
@@ -23,7 +23,7 @@ class RerankerEngine:
         self.cfg = get_model_config(model)
         self.device = device
         self.max_seq = min(max_seq, self.cfg.max_seq)
-        self.tokenizer = ByteTokenizer()
+        self.tokenizer = get_tokenizer()
         self.model = Transformer(self.cfg, device=device, dtype=dtype, seed=seed)
 
     def _graph_pool(self):

@@ -124,7 +124,7 @@ def test_generate_with_prefix_kv_matches_disabled(monkeypatch):
     from sentio_amd.engines.generator import GeneratorEngine
 
     eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
-    prefix = "system preamble shared across requests. " * 3
+    prefix = "system preamble shared across requests. " * 10
     prompts = [prefix + f"question number {i}?" for i in range(3)]
     monkeypatch.setenv("SENTIO_PREFIX_KV", "0")
     base = eng.generate(prompts, max_new_tokens=8, temperature=0.0,
@@ -142,7 +142,7 @@ def test_prefix_split_token_exactness():
     from sentio_amd.engines.generator import GeneratorEngine
 
     eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
-    prefix = "shared preamble text. " * 5
+    prefix = "shared preamble text. " * 16
     prompts = [prefix + f"tail {i}" for i in range(3)]
     pre_ids, suffixes = eng._split_shared_prefix(prompts, 200)
     assert len(pre_ids) >= 64
@@ -164,12 +164,35 @@ def test_prefix_split_nonascii_token_budget():
     budget = 200
     pre_ids, suffixes = eng._split_shared_prefix(prompts, budget)
     assert len(pre_ids) <= budget - 8
+    if pre_ids:   # BPE declines whitespace-free prefixes (no safe boundary)
+        for p, suf in zip(prompts, suffixes):
+            full = eng.tokenizer.encode(p, None)
+            suf_ids = eng.tokenizer.encode(suf, None, add_bos=False)
+            assert pre_ids + suf_ids == full
+            assert len(suf_ids) >= 1
+    # and the full generate() path survives a CJK shared prefix
+    out = eng.generate(prompts, max_new_tokens=4, temperature=0.0,
+                       stop_on_eos=False)
+    assert len(out) == 3
+
+
+def test_prefix_split_nonascii_byte_tokenizer(monkeypatch):
+    """Byte tokenizer (fallback): CJK prefixes DO split, exactly, within the
+    token budget — the original ADVICE r1 regression case."""
+    monkeypatch.setenv("SENTIO_TOKENIZER", "byte")
+    from sentio_amd.engines.generator import GeneratorEngine
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=256)
+    prefix = "共有システムプリアンブル。質問に答えてください。" * 8
+    prompts = [prefix + f"質問{i}?" for i in range(3)]
+    budget = 200
+    pre_ids, suffixes = eng._split_shared_prefix(prompts, budget)
+    assert 0 < len(pre_ids) <= budget - 8
     for p, suf in zip(prompts, suffixes):
         full = eng.tokenizer.encode(p, None)
         suf_ids = eng.tokenizer.encode(suf, None, add_bos=False)
         assert pre_ids + suf_ids == full
         assert len(suf_ids) >= 1
-    # and the full generate() path survives a CJK shared prefix
     out = eng.generate(prompts, max_new_tokens=4, temperature=0.0,
                        stop_on_eos=False)
     assert len(out) == 3

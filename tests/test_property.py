@@ -458,3 +458,45 @@ def test_comb_sum_scores_bounded(dense, sparse):
         assert -1e-6 <= s <= 1.0 + 1e-6     # minmax-normalized weighted sum
     ids = [i for i, _ in fused]
     assert len(ids) == len(set(ids))
+
+
+# ---------------- BPE tokenizer (offline-trained, byte-level) ----------------
+
+from sentio_amd.engines.bpe import BPETokenizer
+
+bpe = BPETokenizer()
+
+
+@given(st.text(max_size=200))
+@settings(max_examples=150, deadline=None)
+def test_bpe_roundtrip_exact(text):
+    """Byte-level BPE is fully reversible on arbitrary unicode."""
+    ids = bpe.encode(text, None, add_bos=False)
+    assert bpe.decode(ids) == text
+
+
+@given(st.text(max_size=120), st.integers(min_value=1, max_value=32))
+@settings(max_examples=100, deadline=None)
+def test_bpe_truncation_is_prefix(text, max_len):
+    assert bpe.encode(text, max_len) == bpe.encode(text, None)[:max_len]
+
+
+@given(st.text(max_size=160), st.text(min_size=1, max_size=40),
+       st.integers(min_value=2, max_value=64))
+@settings(max_examples=150, deadline=None)
+def test_bpe_prefix_split_exactness(prefix_txt, tail, budget):
+    """prefix_split must be split-exact: encode(prefix)+encode(rest,no BOS)
+    == encode(full) — the property prefix-KV caching relies on."""
+    full_txt = prefix_txt + tail
+    n, pre = bpe.prefix_split(prefix_txt, budget)
+    assert len(pre) <= budget
+    if n:
+        suf = bpe.encode(full_txt[n:], None, add_bos=False)
+        assert pre + suf == bpe.encode(full_txt, None)
+
+
+def test_bpe_compression_on_english():
+    text = ("retrieval augmented generation systems combine a search index "
+            "with a language model to answer questions about documents") * 3
+    ids = bpe.encode(text, None)
+    assert len(text) / len(ids) > 2.5   # multi-char tokens, not bytes
