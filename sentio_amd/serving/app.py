@@ -115,7 +115,11 @@ def create_app(settings: Settings | None = None,
         hc = container.health_checker()
         hc.run_checks()
         hc.start()
+        from sentio_amd.observability import tracing as _tracing
+
+        _tracing.start_otlp_exporter()   # no-op without OTLP_ENDPOINT
         yield
+        _tracing.stop_otlp_exporter()
         hc.stop()
 
     app = FastAPI(title="sentio-amd", version=__import__("sentio_amd").__version__,

@@ -299,3 +299,60 @@ def test_kernel_timer_feeds_metrics_histogram():
     snap = metrics_collector.snapshot()
     keys = [k for k in snap["durations"] if "gpu_region_seconds" in str(k)]
     assert keys, f"gpu_region_seconds series missing: {list(snap['durations'])[:10]}"
+
+
+# ---- OTLP/HTTP JSON trace export (reference tracing.py:87-179 role) ----
+
+def test_otlp_export_ships_spans_to_collector():
+    """flush_otlp POSTs the ring buffer's new spans as OTLP JSON to a
+    collector endpoint; the watermark prevents re-export."""
+    import http.server
+    import json as _json
+    import threading
+
+    from sentio_amd.observability import tracing
+
+    received: list[dict] = []
+
+    class Collector(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            body = self.rfile.read(int(self.headers["Content-Length"]))
+            received.append(_json.loads(body))
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(b"{}")
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), Collector)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        endpoint = f"http://127.0.0.1:{srv.server_port}/v1/traces"
+        tracing.clear_spans()
+        with tracing.trace_operation("otlp-test-span", stage="retrieve"):
+            pass
+        try:
+            with tracing.trace_operation("otlp-error-span"):
+                raise ValueError("boom")
+        except ValueError:
+            pass
+        n = tracing.flush_otlp(endpoint)
+        assert n == 2
+        assert len(received) == 1
+        spans = received[0]["resourceSpans"][0]["scopeSpans"][0]["spans"]
+        names = {s["name"] for s in spans}
+        assert names == {"otlp-test-span", "otlp-error-span"}
+        ok = next(s for s in spans if s["name"] == "otlp-test-span")
+        assert int(ok["endTimeUnixNano"]) >= int(ok["startTimeUnixNano"])
+        assert {"key": "stage", "value": {"stringValue": "retrieve"}} in ok["attributes"]
+        err = next(s for s in spans if s["name"] == "otlp-error-span")
+        assert err["status"]["code"] == 2
+        # watermark: nothing new -> nothing shipped
+        assert tracing.flush_otlp(endpoint) == 0
+        # background flusher wiring
+        assert tracing.start_otlp_exporter(endpoint, interval_s=30.0)
+        tracing.stop_otlp_exporter()
+    finally:
+        srv.shutdown()
