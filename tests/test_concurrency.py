@@ -187,3 +187,40 @@ def test_stream_and_chat_coalesce_into_one_batch():
         assert st["max_batch_seen"] >= 2, st   # they coalesced
     finally:
         gen.batcher.stop()
+
+
+def test_batcher_collects_while_engine_busy():
+    """Requests arriving while the engine decodes the previous batch must
+    coalesce into ONE next batch (not fragment into 8 ms windows)."""
+    import time as _t
+
+    from sentio_amd.serving.batcher import DynamicBatcher
+
+    class SlowEngine:
+        import threading as _th
+
+        def __init__(self):
+            import threading
+            self._gen_lock = threading.Lock()
+            self.tokenizer = None
+
+        def generate(self, prompts, **kw):
+            with self._gen_lock:
+                _t.sleep(0.25)
+                return [f"out:{p}" for p in prompts]
+
+    eng = SlowEngine()
+    b = DynamicBatcher(eng, max_batch=16, max_wait_ms=8.0)
+    from concurrent.futures import ThreadPoolExecutor
+
+    with ThreadPoolExecutor(max_workers=8) as ex:
+        futs = [ex.submit(b.generate, "p0")]
+        _t.sleep(0.05)          # batch 1 (just p0) is now running
+        for i in range(1, 7):
+            futs.append(ex.submit(b.generate, f"p{i}"))
+            _t.sleep(0.02)      # staggered arrivals during batch 1
+        outs = [f.result(timeout=30) for f in futs]
+    assert len(outs) == 7
+    b.stop()
+    assert b.stats["batches"] == 2, b.stats       # [p0], [p1..p6]
+    assert b.stats["max_batch_seen"] == 6, b.stats
