@@ -174,10 +174,14 @@ class GeneratorEngine:
         use_graphs = (
             self._GRAPHS_ENABLED
             and self.device != "cpu"
-            # RCCL collectives inside hipGraph capture are not worth the
-            # risk for TP decode (the all-reduce graph-capture path varies
-            # by RCCL version); TP decode runs eager.
-            and not self.model.tp.enabled
+            # TP decode: the chunked async all-reduces (parallel/tp.py)
+            # are stream-ordered and RCCL supports graph capture, but this
+            # environment has no multi-GPU lease to validate the captured
+            # collective schedule on real xGMI — graphed TP decode is
+            # therefore opt-in (SENTIO_TP_HIPGRAPH=1); the default stays
+            # eager, correct by construction.
+            and (not self.model.tp.enabled
+                 or os.environ.get("SENTIO_TP_HIPGRAPH", "0") == "1")
             and os.environ.get("SENTIO_DISABLE_HIPGRAPH", "0") != "1"
         )
         key = (batch, cache_len)

@@ -25,7 +25,8 @@ import torch
 
 from sentio_amd import ops
 from sentio_amd.engines.configs import ModelConfig
-from sentio_amd.parallel.tp import TPContext, shard_columns, shard_qkv, shard_rows
+from sentio_amd.parallel.tp import (TPContext, linear_row_parallel,
+                                    shard_columns, shard_qkv, shard_rows)
 
 
 def _dtype(name: str) -> torch.dtype:
@@ -195,8 +196,10 @@ class Transformer:
         else:
             out = ops.attention(q, k, v, causal=cfg.causal, scale=self.scale,
                                 kv_lens=kv_lens)
-        out = torch.nn.functional.linear(out.reshape(B * S, H * hd), layer["wo"])
-        out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
+        # row-parallel output projection: partial-sum all-reduce overlapped
+        # chunk-wise with the GEMM (parallel/tp.py)
+        out = linear_row_parallel(out.reshape(B * S, H * hd), layer["wo"],
+                                  self.tp)
         return out.view(B, S, d)
 
     def _ffn(self, x: torch.Tensor, layer: dict) -> torch.Tensor:
@@ -209,12 +212,12 @@ class Transformer:
         if rows <= 64 and self.device != "cpu":
             gu = ops.lt_linear(xx, layer["w_gate_up"])
             y = ops.swiglu_packed(gu)   # fused [gate|up] split + silu·up
-            out = ops.lt_linear(y, layer["w_down"])
+            out = linear_row_parallel(y, layer["w_down"], self.tp,
+                                      linear=ops.lt_linear)
         else:
             gu = torch.nn.functional.linear(xx, layer["w_gate_up"])
             y = ops.swiglu_packed(gu)
-            out = torch.nn.functional.linear(y, layer["w_down"])
-        out = self.tp.all_reduce(out)   # row-parallel sum over TP ranks
+            out = linear_row_parallel(y, layer["w_down"], self.tp)
         return out.view(B, S, d)
 
     def forward_hidden(
@@ -274,8 +277,8 @@ class Transformer:
         attn = attn_fn or ops.decode_attention
         out = attn(q, cache.k[layer_idx], cache.v[layer_idx],
                    attn_lens, self.scale)
-        out = ops.lt_linear(out.view(B, self.h_local * hd), layer["wo"])
-        out = self.tp.all_reduce(out)
+        out = linear_row_parallel(out.view(B, self.h_local * hd),
+                                  layer["wo"], self.tp, linear=ops.lt_linear)
         return out.view(B, 1, d)
 
     def forward_decode(self, tokens: torch.Tensor, cache: KVCache) -> torch.Tensor:

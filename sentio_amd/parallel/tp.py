@@ -12,11 +12,16 @@ The Transformer engine consumes this through `TPContext`, which tells the
 weight container to materialize only this rank's shard (sliced from the
 same seeded generator so TP=N matches TP=1 bit-for-bit in fp32).
 
-Status: the per-layer all-reduces run in-stream (synchronous with compute).
-Overlapping them with the next projection on a dedicated HIP stream
-(SURVEY §7 hard part 2) is a planned round-2 optimization — at decode the
-activations are tiny ([B, dim] bf16 ≈ 256 KB at B=32), so the all-reduce is
-latency- not bandwidth-bound and the overlap window is the GEMM launch gap.
+Comm/compute overlap (SURVEY §7 hard part 2): every row-parallel output
+projection goes through `linear_row_parallel`, which splits the GEMM along
+its OUTPUT dim into chunks and launches each chunk's partial-sum
+all-reduce with async_op=True as soon as that chunk's GEMM is enqueued —
+RCCL runs the collective on its own stream, so chunk i's reduction rides
+under chunk i+1's GEMM (and under the next kernel for the last chunk until
+the wait).  xGMI is point-to-point (7 links x ~153 GB/s), so the prefill
+reduces are per-link bandwidth-bound and chunking shortens the exposed
+tail; decode reduces are latency-bound and the overlap window is the
+launch gap.
 """
 
 from __future__ import annotations
@@ -42,11 +47,46 @@ class TPContext:
             dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.group)
         return t
 
+    def all_reduce_async(self, t: torch.Tensor):
+        """Launch the partial-sum reduce without blocking the compute
+        stream; returns the Work handle (None when TP is off)."""
+        if not self.enabled:
+            return None
+        return dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.group,
+                               async_op=True)
+
     @classmethod
     def from_env(cls) -> "TPContext":
         if dist.is_initialized() and dist.get_world_size() > 1:
             return cls(rank=dist.get_rank(), world=dist.get_world_size())
         return cls()
+
+
+def linear_row_parallel(x: torch.Tensor, w: torch.Tensor, tp: TPContext,
+                        chunks: int = 2, linear=None) -> torch.Tensor:
+    """y = x @ W_shard^T summed over TP ranks, with the all-reduce
+    OVERLAPPED: W (stored [out, in], TN layout) splits along `out` into
+    `chunks`; chunk i's async all-reduce is in flight while chunk i+1's
+    GEMM runs.  `linear` overrides the GEMM (e.g. ops.lt_linear for
+    shape-stable decode rows).  TP off → one plain GEMM."""
+    lin = linear or torch.nn.functional.linear
+    if not tp.enabled:
+        return lin(x, w)
+    n = w.shape[0]
+    if chunks <= 1 or n < 2 * chunks:
+        out = lin(x, w)
+        tp.all_reduce(out)
+        return out
+    step = (n + chunks - 1) // chunks
+    outs, works = [], []
+    for i in range(chunks):
+        o = lin(x, w[i * step: (i + 1) * step])
+        works.append(tp.all_reduce_async(o))   # rides under the next GEMM
+        outs.append(o)
+    for wk in works:
+        if wk is not None:
+            wk.wait()   # stream-ordered on NCCL/RCCL (no host sync)
+    return torch.cat(outs, dim=-1)
 
 
 def shard_columns(w: torch.Tensor, tp: TPContext) -> torch.Tensor:

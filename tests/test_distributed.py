@@ -167,6 +167,22 @@ def _tp_worker(rank, world):
     from sentio_amd.engines.configs import get_model_config
     from sentio_amd.engines.transformer import KVCache, Transformer
     from sentio_amd.parallel.tp import TPContext
+    import torch.distributed as dist
+
+    # count overlapped (async) partial-sum reduces: the row-parallel
+    # projections must go through the chunked async path, not in-stream
+    # all-reduces
+    async_calls = [0]
+    real_all_reduce = dist.all_reduce
+
+    def counting_all_reduce(*a, **kw):
+        if kw.get("async_op"):
+            async_calls[0] += 1
+        return real_all_reduce(*a, **kw)
+
+    dist.all_reduce = counting_all_reduce
+    import sentio_amd.parallel.tp as tp_mod
+    tp_mod.dist.all_reduce = counting_all_reduce
 
     cfg = get_model_config("tiny-decoder64")
     tp = TPContext.from_env()
@@ -190,6 +206,10 @@ def _tp_worker(rank, world):
     d_tp = model_tp.decode_step(step_tok, cache)
     d_ref = model_ref.decode_step(step_tok, cache_ref)
     torch.testing.assert_close(d_tp, d_ref, rtol=1e-3, atol=1e-3)
+
+    # every row-parallel projection (attn-out + ffn-down per layer, across
+    # forward/prefill/decode) reduced via the overlapped async path
+    assert async_calls[0] >= 2 * 2 * cfg.n_layers, async_calls[0]
 
 
 def test_tp2_matches_tp1():
