@@ -51,6 +51,9 @@ __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
 // 4 waves per block: waves handle 4 consecutive 16-row Q tiles of ONE
 // (batch, head) and SHARE the K/V LDS staging — 4x less HBM/LDS staging
 // traffic than wave-private tiles, cooperative 256-thread staging.
+// (Measured r2: FA_WAVES=8 with correctly sized LDS is 3-4% SLOWER —
+// 153 vs 159 TF/s — and KVBLK=128 much slower, 94 TF/s: bigger shared
+// tiles cost occupancy/barrier latency more than they save staging.)
 #define FA_WAVES 4
 
 // QT q-tiles (16 rows each) per wave: the K fragment loads and the V
@@ -63,7 +66,7 @@ __device__ __forceinline__ int v_tr_off(int key, int d) {  // element offset
 // and the queries are a SUFFIX starting at absolute position q_off —
 // prefix-KV-cached prefill attends to cache rows [0, kv_lens[b]).
 template <int DT, bool CACHE_SRC>
-__launch_bounds__(256)
+__launch_bounds__(FA_WAVES * WAVE)
 __global__ void flash_attn_kernel(
     const bf16* __restrict__ q,    // [B, S, H, D] (S = suffix len if CACHE_SRC)
     const bf16* __restrict__ k,
@@ -585,12 +588,13 @@ hipError_t sentio_flash_attn(const void* q, const void* k, const void* v,
                              int H, int Hkv, int D, float scale, int causal,
                              hipStream_t stream) {
   size_t lds = (size_t)KVBLK * D * 2 * 2   // K (swizzled) + V (tr image)
-               + 4 * FA_QT * QBLK * P_STRIDE;
+               + FA_WAVES * FA_QT * QBLK * P_STRIDE;
   const int wave_rows = FA_QT * QBLK;
-  dim3 grid((S + 4 * wave_rows - 1) / (4 * wave_rows), H, B);
+  dim3 grid((S + FA_WAVES * wave_rows - 1) / (FA_WAVES * wave_rows), H, B);
 #define FA_CASE(DV)                                                          \
   case DV:                                                                   \
-    hipLaunchKernelGGL((flash_attn_kernel<DV, false>), grid, dim3(256), lds, \
+    hipLaunchKernelGGL((flash_attn_kernel<DV, false>), grid,               \
+                       dim3(FA_WAVES * WAVE), lds,                            \
                        stream, (const bf16*)q, (const bf16*)k,               \
                        (const bf16*)v, (bf16*)out, kv_lens, B, S, H, Hkv, D, \
                        scale, causal, 0, 0);                                 \
@@ -610,12 +614,14 @@ hipError_t sentio_flash_attn_cache(const void* q, const void* kc,
                                    const int* kv_lens, int B, int S, int H,
                                    int Hkv, int Smax, int D, float scale,
                                    int q_off, hipStream_t stream) {
-  size_t lds = (size_t)KVBLK * D * 2 * 2 + 4 * FA_QT * QBLK * P_STRIDE;
+  size_t lds = (size_t)KVBLK * D * 2 * 2
+               + FA_WAVES * FA_QT * QBLK * P_STRIDE;
   const int wave_rows = FA_QT * QBLK;
-  dim3 grid((S + 4 * wave_rows - 1) / (4 * wave_rows), H, B);
+  dim3 grid((S + FA_WAVES * wave_rows - 1) / (FA_WAVES * wave_rows), H, B);
 #define FAC_CASE(DV)                                                         \
   case DV:                                                                   \
-    hipLaunchKernelGGL((flash_attn_kernel<DV, true>), grid, dim3(256), lds,  \
+    hipLaunchKernelGGL((flash_attn_kernel<DV, true>), grid,                \
+                       dim3(FA_WAVES * WAVE), lds,                            \
                        stream, (const bf16*)q, (const bf16*)kc,              \
                        (const bf16*)vc, (bf16*)out, kv_lens, B, S, H, Hkv,   \
                        D, scale, 1, Smax, q_off);                            \
