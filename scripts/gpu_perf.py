@@ -95,16 +95,25 @@ def bench_encoder():
     texts = ["sample document text for embedding " * 10] * 64
     t = timeit(lambda: enc.embed(texts), warmup=2, iters=5)
     RESULTS["encoder_texts_s"] = 64 / t
+    # serving/bench query shape: short sequences, and the per-request
+    # single-text latency (encoder stage cost)
+    encq = EncoderEngine("sentio-encoder-base", device=DEV, max_seq=128)
+    queries = ["what does this mean for retrieval?"] * 32
+    t = timeit(lambda: encq.embed(queries), warmup=3, iters=10)
+    RESULTS["encoder_query_batch32_ms"] = t * 1e3
+    t = timeit(lambda: encq.embed(queries[:1]), warmup=3, iters=20)
+    RESULTS["encoder_query_single_ms"] = t * 1e3
 
 
-def bench_generator(name="llama3-1b", B=8, S=512, new=32):
+def bench_generator(name="llama3-1b", B=8, S=512, new=32, tag=None):
     from sentio_amd.engines.generator import GeneratorEngine
     from sentio_amd.engines.transformer import KVCache
 
+    tag = tag or name
     t0 = time.time()
     g = GeneratorEngine(name, device=DEV, max_seq=S + new + 8)
     torch.cuda.synchronize()
-    RESULTS[f"{name}_init_s"] = time.time() - t0
+    RESULTS[f"{tag}_init_s"] = time.time() - t0
 
     tokens = torch.randint(3, 258, (B, S), device=DEV)
     cache = KVCache(g.cfg, B, S + new + 4, DEV, g.model.dtype)
@@ -114,7 +123,7 @@ def bench_generator(name="llama3-1b", B=8, S=512, new=32):
         return g.model.prefill(tokens, cache)
 
     t = timeit(prefill, warmup=1, iters=3)
-    RESULTS[f"{name}_prefill_tok_s"] = B * S / t
+    RESULTS[f"{tag}_prefill_tok_s"] = B * S / t
 
     logits = prefill()
     cur = logits.argmax(-1, keepdim=True)
@@ -123,8 +132,8 @@ def bench_generator(name="llama3-1b", B=8, S=512, new=32):
         return g.model.decode_step(cur, cache)
 
     t = timeit(decode, warmup=3, iters=10)
-    RESULTS[f"{name}_decode_tok_s"] = B / t
-    RESULTS[f"{name}_decode_step_ms"] = t * 1e3
+    RESULTS[f"{tag}_decode_tok_s"] = B / t
+    RESULTS[f"{tag}_decode_step_ms"] = t * 1e3
 
 
 def main():
@@ -141,6 +150,9 @@ def main():
         ("encoder", bench_encoder),
         ("gen1b", lambda: bench_generator("llama3-1b")),
         ("gen8b", lambda: bench_generator("llama3-8b", B=8, S=512)),
+        # flagship decode anchor: batch 32, ~900-token prompts (BPE era)
+        ("gen8b32", lambda: bench_generator("llama3-8b", B=32, S=896,
+                                            new=128, tag="llama3-8b_b32")),
     ]
     for name, fn in steps:
         if which != "all" and which != name:
