@@ -101,6 +101,7 @@ class Settings:
     kv_cache_max_tokens: int = 8192
     max_batch_size: int = 32
     dynamic_batching: bool = True           # batch concurrent /chat generations
+    continuous_batching: bool = True        # requests join decode mid-flight
     batch_wait_ms: float = 8.0
     device: str = "auto"                    # auto | cuda | cpu
 
@@ -148,6 +149,7 @@ class Settings:
         "kv_cache_max_tokens": "KV_CACHE_MAX_TOKENS",
         "max_batch_size": "MAX_BATCH_SIZE",
         "dynamic_batching": "DYNAMIC_BATCHING",
+        "continuous_batching": "CONTINUOUS_BATCHING",
         "batch_wait_ms": "BATCH_WAIT_MS",
         "device": "SENTIO_DEVICE",
         "enable_metrics": "ENABLE_METRICS",

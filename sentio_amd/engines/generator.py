@@ -196,6 +196,72 @@ class GeneratorEngine:
         pool[key] = sess
         return sess
 
+    # ----- continuous-batching support (serving.ContinuousBatcher) -----
+    def make_slot_session(self, slots: int) -> "_DecodeSession":
+        """A PRIVATE decode session for the continuous batcher: its KV
+        slots persist across requests, so it must never come from (or be
+        reset by) the shared _decode_session pool."""
+        import os
+
+        use_graphs = (
+            self._GRAPHS_ENABLED
+            and self.device != "cpu"
+            and (not self.model.tp.enabled
+                 or os.environ.get("SENTIO_TP_HIPGRAPH", "0") == "1")
+            and os.environ.get("SENTIO_DISABLE_HIPGRAPH", "0") != "1"
+        )
+        return _DecodeSession(self, slots, self.max_seq, use_graphs)
+
+    @torch.inference_mode()
+    def prefill_into_slots(self, sess: "_DecodeSession", rows: list[int],
+                           prompts: list[str],
+                           max_new_list: list[int]) -> torch.Tensor:
+        """Admission prefill (continuous batching): encode + prefill the
+        prompts in a temporary cache, copy their KV into sess.cache at
+        `rows`, set those rows' seq_lens, and return last-real-position
+        logits [len(rows), V].  Other rows keep decoding untouched —
+        a joining request never stalls resident ones beyond this call."""
+        with self._gen_lock:
+            ids_list = []
+            for p, mn in zip(prompts, max_new_list):
+                budget = max(self.max_seq - mn - 1, 8)
+                ids_list.append(self.tokenizer.encode(p[-4 * budget:], budget))
+            lens = [len(i) for i in ids_list]
+            S = max(lens)
+            padded = [i + [0] * (S - len(i)) for i in ids_list]
+            tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
+            lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
+            tmp = KVCache(self.cfg, len(rows), S, self.device,
+                          self.model.dtype, n_kv_heads=self.model.hkv_local)
+            logits = self.model.prefill(tokens, tmp, lens=lens_t)
+            rows_t = torch.tensor(rows, dtype=torch.int64, device=self.device)
+            for li in range(self.cfg.n_layers):
+                sess.cache.k[li][rows_t, :, :S] = tmp.k[li]
+                sess.cache.v[li][rows_t, :, :S] = tmp.v[li]
+            sess.cache.seq_lens[rows_t] = lens_t
+            return logits
+
+    @torch.inference_mode()
+    def decode_step_session(self, sess: "_DecodeSession",
+                            cur: torch.Tensor) -> torch.Tensor:
+        """One decode step over ALL slots (hipGraph replay when captured)."""
+        with self._gen_lock:
+            return sess.decode_step(cur)
+
+    @torch.inference_mode()
+    def sample_rows(self, logits: torch.Tensor,
+                    temps: torch.Tensor) -> torch.Tensor:
+        """Per-ROW-temperature sampling — continuous batches mix requests
+        with different temperatures: greedy rows take argmax, the rest
+        Gumbel-argmax (torch eager; the split sampling kernel takes one
+        scalar temperature and stays on the uniform-batch path)."""
+        self._step_seed += 1
+        t = temps.to(logits.device).view(-1, 1)
+        g = -torch.log(-torch.log(
+            torch.rand_like(logits).clamp_min(1e-20)).clamp_min(1e-20))
+        scores = torch.where(t > 0, logits / t.clamp_min(1e-6) + g, logits)
+        return scores.argmax(-1)
+
     @torch.inference_mode()
     def generate(
         self,

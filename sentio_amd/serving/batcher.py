@@ -208,6 +208,238 @@ class DynamicBatcher:
                 **self.stats}
 
 
+class ContinuousBatcher:
+    """TRUE continuous batching: requests JOIN the decode loop at step
+    boundaries (admission prefill into free KV-cache slots), stream their
+    tokens out per step, and leave when done.  Unlike the wave-batched
+    DynamicBatcher, a request's latency is its own prefill + its own
+    tokens — it never waits for co-tenants' remaining tokens, and
+    max_new_tokens / temperature / stop_on_eos may differ per request
+    (per-row sampling).  This is what the decode engine's per-row
+    seq_lens design exists for.
+
+    Slot safety: free rows keep riding the (graph-replayed) decode step
+    with garbage inputs — their seq_lens are re-zeroed every step so the
+    KV write position can never run off the cache end, and their outputs
+    are never read."""
+
+    def __init__(self, generator, slots: int = 32, admit_max: int = 8):
+        self.generator = generator
+        self.n_slots = slots
+        self.admit_max = admit_max
+        self._q: queue.Queue[_Item] = queue.Queue()
+        self._stop = threading.Event()
+        self._thread: threading.Thread | None = None
+        self._start_lock = threading.Lock()
+        self.stats = {"requests": 0, "completed": 0, "steps": 0,
+                      "max_concurrent": 0, "admissions": 0}
+
+    # ----- caller side -----
+    def generate(self, prompt: str, max_new_tokens: int = 128,
+                 temperature: float = 0.3, stop_on_eos: bool = True,
+                 timeout_s: float = 300.0) -> str:
+        self.start()
+        item = _Item(prompt, max_new_tokens, float(temperature), stop_on_eos)
+        self._q.put(item)
+        return item.future.result(timeout=timeout_s)
+
+    def generate_stream(self, prompt: str, max_new_tokens: int = 128,
+                        temperature: float = 0.3, timeout_s: float = 300.0):
+        self.start()
+        item = _Item(prompt, max_new_tokens, float(temperature), True,
+                     stream_q=queue.Queue())
+        self._q.put(item)
+        from sentio_amd.engines.tokenizer import EOS_ID
+
+        tok = self.generator.tokenizer
+        generated: list[int] = []
+        emitted = ""
+        deadline = time.monotonic() + timeout_s
+        while True:
+            try:
+                t = item.stream_q.get(timeout=max(0.1,
+                                                  deadline - time.monotonic()))
+            except queue.Empty:
+                raise TimeoutError("stream starved (engine stalled)") from None
+            if t is None or t == EOS_ID:
+                break
+            generated.append(t)
+            text = tok.decode(generated)
+            if len(text) > len(emitted):
+                yield text[len(emitted):]
+                emitted = text
+
+    # ----- worker side -----
+    def start(self) -> None:
+        with self._start_lock:
+            if self._thread is not None and self._thread.is_alive():
+                return
+            self._stop.clear()
+            self._thread = threading.Thread(target=self._loop, daemon=True,
+                                            name="sentio-continuous")
+            self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=5.0)
+            self._thread = None
+        while True:
+            try:
+                item = self._q.get_nowait()
+            except queue.Empty:
+                break
+            if item.stream_q is not None:
+                item.stream_q.put(None)
+            if not item.future.done():
+                item.future.set_exception(RuntimeError("batcher stopped"))
+
+    def _loop(self) -> None:
+        try:
+            self._loop_inner()
+        except Exception as exc:   # a dead loop must not strand callers
+            while True:
+                try:
+                    item = self._q.get_nowait()
+                except queue.Empty:
+                    break
+                if item.stream_q is not None:
+                    item.stream_q.put(None)
+                if not item.future.done():
+                    item.future.set_exception(exc)
+            raise
+
+    def _loop_inner(self) -> None:
+        import torch
+
+        from sentio_amd.engines.tokenizer import EOS_ID
+
+        gen = self.generator
+        dev = gen.device
+        sess = gen.make_slot_session(self.n_slots)
+        state: list[dict | None] = [None] * self.n_slots
+        cur = torch.zeros(self.n_slots, dtype=torch.int64, device=dev)
+        temps = torch.zeros(self.n_slots, device=dev)
+        free_mask = torch.ones(self.n_slots, dtype=torch.bool, device=dev)
+
+        def finish(r: int, reason: str) -> None:
+            s = state[r]
+            state[r] = None
+            free_mask[r] = True
+            self.stats["completed"] += 1
+            it = s["item"]
+            if it.stream_q is not None:
+                it.stream_q.put(None)
+            ids = s["ids"]
+            if it.stop_on_eos and EOS_ID in ids:
+                ids = ids[: ids.index(EOS_ID)]
+            if not it.future.done():
+                it.future.set_result(gen.tokenizer.decode(ids))
+
+        def feed(r: int, tok: int) -> None:
+            s = state[r]
+            s["ids"].append(tok)
+            it = s["item"]
+            if it.stream_q is not None:
+                it.stream_q.put(tok)
+            s["remaining"] -= 1
+            if (s["remaining"] <= 0
+                    or (it.stop_on_eos and tok == EOS_ID)):
+                finish(r, "done")
+
+        while not self._stop.is_set():
+            n_active = sum(s is not None for s in state)
+            # ---- admission: fill free slots from the queue ----
+            free_rows = [i for i, s in enumerate(state) if s is None]
+            want: list[_Item] = []
+            while free_rows[len(want):] and len(want) < self.admit_max:
+                try:
+                    nxt = self._q.get(
+                        timeout=0.05 if (n_active == 0 and not want) else 0)
+                except queue.Empty:
+                    break
+                want.append(nxt)
+            if want:
+                rows = free_rows[: len(want)]
+                try:
+                    logits = gen.prefill_into_slots(
+                        sess, rows, [it.prompt for it in want],
+                        [it.max_new_tokens for it in want])
+                    t_adm = torch.tensor([it.temperature for it in want],
+                                         device=dev)
+                    tok0 = gen.sample_rows(logits, t_adm).cpu().tolist()
+                    for j, (r, it) in enumerate(zip(rows, want)):
+                        state[r] = {"item": it, "ids": [],
+                                    "remaining": it.max_new_tokens}
+                        free_mask[r] = False
+                        temps[r] = it.temperature
+                        cur[r] = tok0[j]
+                        feed(r, tok0[j])
+                    self.stats["requests"] += len(want)
+                    self.stats["admissions"] += 1
+                    self.stats["max_concurrent"] = max(
+                        self.stats["max_concurrent"],
+                        sum(s is not None for s in state))
+                except Exception as exc:
+                    for it in want:
+                        if it.stream_q is not None:
+                            it.stream_q.put(None)
+                        if not it.future.done():
+                            it.future.set_exception(exc)
+            if not any(s is not None for s in state):
+                continue
+            # ---- one decode step over every slot ----
+            # free rows must never advance their KV write position off the
+            # cache end: re-zero their seq_lens each step
+            sess.cache.seq_lens.masked_fill_(free_mask, 0)
+            logits = gen.decode_step_session(sess, cur)
+            toks = gen.sample_rows(logits, temps)
+            cur.copy_(toks)
+            self.stats["steps"] += 1
+            toks_host = toks.cpu().tolist()
+            for r in range(self.n_slots):
+                if state[r] is not None:
+                    feed(r, toks_host[r])
+
+    def health(self) -> dict[str, Any]:
+        return {"queued": self._q.qsize(), "mode": "continuous",
+                "slots": self.n_slots,
+                "running": self._thread is not None and self._thread.is_alive(),
+                **self.stats}
+
+
+class ContinuousGenerator:
+    """Generator frontend for continuous batching: single-prompt calls and
+    streams join the slot loop; multi-prompt (bench-style) calls pass
+    through to the raw engine (they interleave with the loop between
+    steps via the engine lock)."""
+
+    def __init__(self, raw, slots: int = 32):
+        self.raw = raw
+        self.batcher = ContinuousBatcher(raw, slots=slots)
+
+    def generate(self, prompts: list[str], max_new_tokens: int = 128,
+                 temperature: float = 0.3, stop_on_eos: bool = True,
+                 **kwargs) -> list[str]:
+        if len(prompts) != 1 or kwargs.get("on_token") is not None:
+            return self.raw.generate(prompts, max_new_tokens=max_new_tokens,
+                                     temperature=temperature,
+                                     stop_on_eos=stop_on_eos, **kwargs)
+        return [self.batcher.generate(prompts[0],
+                                      max_new_tokens=max_new_tokens,
+                                      temperature=temperature,
+                                      stop_on_eos=stop_on_eos)]
+
+    def stream(self, prompt: str, max_new_tokens: int = 128,
+               temperature: float = 0.3):
+        return self.batcher.generate_stream(prompt,
+                                            max_new_tokens=max_new_tokens,
+                                            temperature=temperature)
+
+    def __getattr__(self, name):
+        return getattr(self.raw, name)
+
+
 @dataclass
 class _MicroItem:
     units: list

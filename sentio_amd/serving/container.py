@@ -112,6 +112,14 @@ class ServiceContainer:
         reference's provider batched server-side)."""
         def make():
             raw = self.generator()
+            if (self.settings.continuous_batching
+                    and self.settings.dynamic_batching
+                    and hasattr(raw, "make_slot_session")):
+                # real engine only: mock engines take the wave batcher
+                from sentio_amd.serving.batcher import ContinuousGenerator
+
+                return ContinuousGenerator(
+                    raw, slots=self.settings.max_batch_size)
             if not self.settings.dynamic_batching:
                 return raw
             from sentio_amd.serving.batcher import BatchedGenerator
@@ -273,6 +281,14 @@ class ServiceContainer:
                                      max_new_tokens=2, temperature=0.0)
             except Exception as exc:
                 logger.warning("generator warm-up failed: %s", exc)
+            try:
+                # capture the continuous batcher's slot-session decode
+                # graph before traffic (same mid-traffic-capture hazard)
+                self.generator_frontend().generate(["warm frontend"],
+                                                   max_new_tokens=2,
+                                                   temperature=0.0)
+            except Exception as exc:
+                logger.warning("frontend warm-up failed: %s", exc)
         self.pipeline()
         self.ingestor()
         self.auth_manager()
