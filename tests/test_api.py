@@ -306,3 +306,17 @@ def test_chat_history_malformed_entries_tolerated(client):
         "history": [{"not_role": "x"}, {}],
     })
     assert r.status_code == 200
+
+
+def test_ui_upload_contract_matches_embed_schema(client):
+    """The built-in UI posts {content, metadata} to /embed — keep the page's
+    JS contract aligned with EmbedRequest (the r1 page sent {text:...},
+    which 422'd in a real browser)."""
+    from sentio_amd.serving.ui import UI_HTML
+
+    assert "content:text" in UI_HTML.replace(" ", "")
+    assert "'/chat/stream'" in UI_HTML          # streaming path wired
+    assert "45000" in UI_HTML                   # reference chunk size
+    r = client.post("/embed", json={"content": "ui contract doc",
+                                    "metadata": {"source": "ui"}})
+    assert r.status_code == 200
