@@ -213,33 +213,63 @@ class GeneratorEngine:
         return _DecodeSession(self, slots, self.max_seq, use_graphs)
 
     @torch.inference_mode()
+    def prefill_admission(self, prompts: list[str],
+                          max_new_list: list[int]) -> dict:
+        """Admission prefill (continuous batching), LOCK-FREE: encode +
+        prefill the prompts into a temporary cache on the CALLER's stream.
+        Touches only the temp cache and read-only weights, so it may run
+        on a side stream CONCURRENTLY with the decode loop's graph
+        replays — a joining request's prefill never stalls resident
+        requests.  integrate_admission() splices the result into slots."""
+        ids_list = []
+        for p, mn in zip(prompts, max_new_list):
+            budget = max(self.max_seq - mn - 1, 8)
+            ids_list.append(self.tokenizer.encode(p[-4 * budget:], budget))
+        lens = [len(i) for i in ids_list]
+        S = max(lens)
+        padded = [i + [0] * (S - len(i)) for i in ids_list]
+        tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
+        lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
+        tmp = KVCache(self.cfg, len(prompts), S, self.device,
+                      self.model.dtype, n_kv_heads=self.model.hkv_local)
+        logits = self.model.prefill(tokens, tmp, lens=lens_t)
+        return {"tmp": tmp, "lens": lens_t, "logits": logits, "S": S}
+
+    @torch.inference_mode()
+    def integrate_admission(self, sess: "_DecodeSession", rows: list[int],
+                            pre: dict,
+                            idx: list[int] | None = None) -> torch.Tensor:
+        """Copy a prefill_admission's KV into sess.cache rows and set their
+        seq_lens (run on the decode loop's stream, AFTER syncing with the
+        admission stream).  `idx` selects a SUBSET of the admission's
+        requests (partial integration when fewer slots are free than the
+        admission holds).  Returns those requests' logits."""
+        tmp, lens_t, S = pre["tmp"], pre["lens"], pre["S"]
+        logits = pre["logits"]
+        with self._gen_lock:
+            rows_t = torch.tensor(rows, dtype=torch.int64, device=self.device)
+            if idx is not None:
+                sel = torch.tensor(idx, dtype=torch.int64, device=self.device)
+                lens_t = lens_t[sel]
+                logits = logits[sel]
+            for li in range(self.cfg.n_layers):
+                src_k, src_v = tmp.k[li], tmp.v[li]
+                if idx is not None:
+                    src_k, src_v = src_k[sel], src_v[sel]
+                sess.cache.k[li][rows_t, :, :S] = src_k
+                sess.cache.v[li][rows_t, :, :S] = src_v
+            sess.cache.seq_lens[rows_t] = lens_t
+        return logits
+
+    @torch.inference_mode()
     def prefill_into_slots(self, sess: "_DecodeSession", rows: list[int],
                            prompts: list[str],
                            max_new_list: list[int]) -> torch.Tensor:
-        """Admission prefill (continuous batching): encode + prefill the
-        prompts in a temporary cache, copy their KV into sess.cache at
-        `rows`, set those rows' seq_lens, and return last-real-position
-        logits [len(rows), V].  Other rows keep decoding untouched —
-        a joining request never stalls resident ones beyond this call."""
+        """prefill_admission + integrate_admission in one call (same-thread
+        admission path / tests)."""
         with self._gen_lock:
-            ids_list = []
-            for p, mn in zip(prompts, max_new_list):
-                budget = max(self.max_seq - mn - 1, 8)
-                ids_list.append(self.tokenizer.encode(p[-4 * budget:], budget))
-            lens = [len(i) for i in ids_list]
-            S = max(lens)
-            padded = [i + [0] * (S - len(i)) for i in ids_list]
-            tokens = torch.tensor(padded, dtype=torch.int64, device=self.device)
-            lens_t = torch.tensor(lens, dtype=torch.int32, device=self.device)
-            tmp = KVCache(self.cfg, len(rows), S, self.device,
-                          self.model.dtype, n_kv_heads=self.model.hkv_local)
-            logits = self.model.prefill(tokens, tmp, lens=lens_t)
-            rows_t = torch.tensor(rows, dtype=torch.int64, device=self.device)
-            for li in range(self.cfg.n_layers):
-                sess.cache.k[li][rows_t, :, :S] = tmp.k[li]
-                sess.cache.v[li][rows_t, :, :S] = tmp.v[li]
-            sess.cache.seq_lens[rows_t] = lens_t
-            return logits
+            pre = self.prefill_admission(prompts, max_new_list)
+        return self.integrate_admission(sess, rows, pre)
 
     @torch.inference_mode()
     def decode_step_session(self, sess: "_DecodeSession",

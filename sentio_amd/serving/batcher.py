@@ -275,24 +275,85 @@ class ContinuousBatcher:
             if self._thread is not None and self._thread.is_alive():
                 return
             self._stop.clear()
+            # admissions prefill on their own thread (+ side HIP stream):
+            # bounded to 2 in flight so temp KV caches stay small
+            self._ready_q = queue.Queue(maxsize=2)
+            self._adm_thread = threading.Thread(
+                target=self._admitter, daemon=True, name="sentio-admit")
+            self._adm_thread.start()
             self._thread = threading.Thread(target=self._loop, daemon=True,
                                             name="sentio-continuous")
             self._thread.start()
 
     def stop(self) -> None:
         self._stop.set()
-        if self._thread is not None:
-            self._thread.join(timeout=5.0)
-            self._thread = None
-        while True:
+        for t in (self._thread, getattr(self, "_adm_thread", None)):
+            if t is not None:
+                t.join(timeout=5.0)
+        self._thread = None
+        self._adm_thread = None
+        for q_ in (self._q, getattr(self, "_ready_q", None)):
+            if q_ is None:
+                continue
+            while True:
+                try:
+                    got = q_.get_nowait()
+                except queue.Empty:
+                    break
+                items = got[0] if isinstance(got, tuple) else [got]
+                for item in items:
+                    if item.stream_q is not None:
+                        item.stream_q.put(None)
+                    if not item.future.done():
+                        item.future.set_exception(
+                            RuntimeError("batcher stopped"))
+
+    def _admitter(self) -> None:
+        """Collect queued requests and prefill them OFF the decode loop:
+        on GPU the prefill runs on a dedicated side stream, concurrent
+        with the loop's decode-graph replays; the loop only pays the
+        KV splice (integrate_admission) between steps."""
+        import torch
+
+        gen = self.generator
+        side = (torch.cuda.Stream()
+                if gen.device != "cpu" and torch.cuda.is_available() else None)
+        while not self._stop.is_set():
+            want: list[_Item] = []
             try:
-                item = self._q.get_nowait()
+                want.append(self._q.get(timeout=0.05))
             except queue.Empty:
-                break
-            if item.stream_q is not None:
-                item.stream_q.put(None)
-            if not item.future.done():
-                item.future.set_exception(RuntimeError("batcher stopped"))
+                continue
+            while len(want) < self.admit_max:
+                try:
+                    want.append(self._q.get_nowait())
+                except queue.Empty:
+                    break
+            try:
+                if side is not None:
+                    with torch.cuda.stream(side):
+                        pre = gen.prefill_admission(
+                            [it.prompt for it in want],
+                            [it.max_new_tokens for it in want])
+                    ev = torch.cuda.Event()
+                    ev.record(side)
+                else:
+                    pre = gen.prefill_admission(
+                        [it.prompt for it in want],
+                        [it.max_new_tokens for it in want])
+                    ev = None
+                while not self._stop.is_set():
+                    try:
+                        self._ready_q.put((want, pre, ev), timeout=0.2)
+                        break
+                    except queue.Full:
+                        continue
+            except Exception as exc:
+                for it in want:
+                    if it.stream_q is not None:
+                        it.stream_q.put(None)
+                    if not it.future.done():
+                        it.future.set_exception(exc)
 
     def _loop(self) -> None:
         try:
@@ -347,46 +408,54 @@ class ContinuousBatcher:
                     or (it.stop_on_eos and tok == EOS_ID)):
                 finish(r, "done")
 
+        pending: tuple | None = None      # an admission waiting for slots
         while not self._stop.is_set():
             n_active = sum(s is not None for s in state)
-            # ---- admission: fill free slots from the queue ----
-            free_rows = [i for i, s in enumerate(state) if s is None]
-            want: list[_Item] = []
-            while free_rows[len(want):] and len(want) < self.admit_max:
+            # ---- integrate a ready admission (prefilled off-loop) ----
+            if pending is None:
                 try:
-                    nxt = self._q.get(
-                        timeout=0.05 if (n_active == 0 and not want) else 0)
+                    want_, pre_, ev_ = self._ready_q.get(
+                        timeout=0.05 if n_active == 0 else 0)
+                    pending = (want_, pre_, ev_, list(range(len(want_))))
                 except queue.Empty:
-                    break
-                want.append(nxt)
-            if want:
-                rows = free_rows[: len(want)]
-                try:
-                    logits = gen.prefill_into_slots(
-                        sess, rows, [it.prompt for it in want],
-                        [it.max_new_tokens for it in want])
-                    t_adm = torch.tensor([it.temperature for it in want],
-                                         device=dev)
-                    tok0 = gen.sample_rows(logits, t_adm).cpu().tolist()
-                    for j, (r, it) in enumerate(zip(rows, want)):
-                        state[r] = {"item": it, "ids": [],
-                                    "remaining": it.max_new_tokens}
-                        free_mask[r] = False
-                        temps[r] = it.temperature
-                        cur[r] = tok0[j]
-                        feed(r, tok0[j])
-                    self.stats["requests"] += len(want)
-                    self.stats["admissions"] += 1
-                    self.stats["max_concurrent"] = max(
-                        self.stats["max_concurrent"],
-                        sum(s is not None for s in state))
-                except Exception as exc:
-                    for it in want:
-                        if it.stream_q is not None:
-                            it.stream_q.put(None)
-                        if not it.future.done():
-                            it.future.set_exception(exc)
-            if not any(s is not None for s in state):
+                    pending = None
+            if pending is not None:
+                want, pre, ev, left = pending
+                free_rows = [i for i, st_ in enumerate(state) if st_ is None]
+                take = left[: len(free_rows)]   # partial: as many as fit
+                if take:
+                    rows = free_rows[: len(take)]
+                    left = left[len(take):]
+                    pending = (want, pre, ev, left) if left else None
+                    try:
+                        if ev is not None:   # admission stream -> loop stream
+                            torch.cuda.current_stream().wait_event(ev)
+                        logits = gen.integrate_admission(sess, rows, pre,
+                                                         idx=take)
+                        items = [want[j] for j in take]
+                        t_adm = torch.tensor([it.temperature for it in items],
+                                             device=dev)
+                        tok0 = gen.sample_rows(logits, t_adm).cpu().tolist()
+                        for j, (r, it) in enumerate(zip(rows, items)):
+                            state[r] = {"item": it, "ids": [],
+                                        "remaining": it.max_new_tokens}
+                            free_mask[r] = False
+                            temps[r] = it.temperature
+                            cur[r] = tok0[j]
+                            feed(r, tok0[j])
+                        self.stats["requests"] += len(items)
+                        self.stats["admissions"] += 1
+                        self.stats["max_concurrent"] = max(
+                            self.stats["max_concurrent"],
+                            sum(st_ is not None for st_ in state))
+                    except Exception as exc:
+                        pending = None
+                        for it in want:
+                            if it.stream_q is not None:
+                                it.stream_q.put(None)
+                            if not it.future.done():
+                                it.future.set_exception(exc)
+            if not any(st_ is not None for st_ in state):
                 continue
             # ---- one decode step over every slot ----
             # free rows must never advance their KV write position off the

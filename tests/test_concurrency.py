@@ -238,8 +238,8 @@ def test_continuous_batching_mid_decode_join_greedy_exact():
     from sentio_amd.engines.generator import GeneratorEngine
     from sentio_amd.serving.batcher import ContinuousGenerator
 
-    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=160)
-    solo_a = eng.generate(["first request about graphs"], max_new_tokens=24,
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=192)
+    solo_a = eng.generate(["first request about graphs"], max_new_tokens=80,
                           temperature=0.0, stop_on_eos=False)[0]
     solo_b = eng.generate(["second one, retrieval topic"], max_new_tokens=10,
                           temperature=0.0, stop_on_eos=False)[0]
@@ -253,9 +253,9 @@ def test_continuous_batching_mid_decode_join_greedy_exact():
                                      temperature=0.0, stop_on_eos=False)[0]
 
         t1 = _th.Thread(target=run,
-                        args=("a", "first request about graphs", 24))
+                        args=("a", "first request about graphs", 80))
         t1.start()
-        _t.sleep(0.15)           # a is mid-decode; b joins now
+        _t.sleep(0.1)            # a is mid-decode; b joins now
         t2 = _th.Thread(target=run,
                         args=("b", "second one, retrieval topic", 10))
         t2.start()
