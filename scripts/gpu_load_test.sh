@@ -29,6 +29,9 @@ cat gpurun_out/load_test_gpu_16c128r.json
 timeout 420 python scripts/load_test.py --clients 32 --requests 192 \
   --stream-frac 0.25 > gpurun_out/load_test_gpu_32c192r.json
 cat gpurun_out/load_test_gpu_32c192r.json
+timeout 420 python scripts/load_test.py --clients 48 --requests 240 \
+  --stream-frac 0.25 > gpurun_out/load_test_gpu_48c240r.json
+cat gpurun_out/load_test_gpu_48c240r.json
 # batcher coalescing evidence + perf counters
 curl -s -m 30 http://127.0.0.1:8000/health/detailed > gpurun_out/load_health_detailed.json
 curl -s -m 30 http://127.0.0.1:8000/metrics/performance > gpurun_out/load_metrics_perf.json
