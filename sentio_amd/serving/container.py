@@ -275,7 +275,14 @@ class ServiceContainer:
             # load test otherwise pays mid-traffic (p95 blowout)
             try:
                 gen = self.generator()
-                for b in (1, 2, 4, 8, 16, 24, 32, 48, 64):
+                # continuous batching serves through ONE slot session;
+                # the wave-batcher's bucket pool would cost ~1 GB of KV
+                # per row (kv_cache_max_tokens) for sessions that mode
+                # never uses (64-slot config OOM'd a 288 GB GPU on the
+                # full pool) — pre-capture only what the active mode runs
+                buckets = ((1,) if self.settings.continuous_batching
+                           else (1, 2, 4, 8, 16, 24, 32, 48, 64))
+                for b in buckets:
                     if b <= self.settings.max_batch_size:
                         gen.generate([f"warm {i}" for i in range(b)],
                                      max_new_tokens=2, temperature=0.0)
