@@ -99,7 +99,7 @@ class Settings:
     tp_degree: int = 1
     index_shards: int = 1                   # = world size when distributed
     kv_cache_max_tokens: int = 8192
-    max_batch_size: int = 32
+    max_batch_size: int = 64          # continuous-batching slots (64 rows of KV ~ 68 GB at 8k ctx)
     dynamic_batching: bool = True           # batch concurrent /chat generations
     continuous_batching: bool = True        # requests join decode mid-flight
     batch_wait_ms: float = 8.0
