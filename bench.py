@@ -268,9 +268,15 @@ def main():
     stage_t: dict[str, float] = {}
 
     def _mark(name: str, t0: float) -> float:
-        """Accumulate per-stage wall time (sync'd) for the stderr breakdown."""
+        """Accumulate per-stage wall time for the stderr breakdown.  Syncs
+        ONLY the current stream: retrieval stages run on a side stream
+        under the previous step's generation, and a device-wide sync made
+        every stage absorb the decode stream's backlog (r1's breakdown
+        showed a 640 ms "embed" stage that is really ~6 ms isolated —
+        VERDICT item 9).  Cross-stream contention still counts; these are
+        in-context times, not isolated kernel rates (see profiles/)."""
         if on_gpu:
-            torch.cuda.synchronize()
+            torch.cuda.current_stream().synchronize()
         t1 = time.perf_counter()
         stage_t[name] = stage_t.get(name, 0.0) + (t1 - t0)
         return t1
