@@ -340,6 +340,7 @@ __global__ void decode_attn_split_kernel(
     const bf16* __restrict__ vc,
     float* __restrict__ ws_o,      // [B, Hkv, SPLITS, G, D]
     float* __restrict__ ws_ml,     // [B, Hkv, SPLITS, G, 2]
+    bf16* __restrict__ out,        // [B, H, D] (written directly at splits==1)
     const int* __restrict__ seq_lens,
     int H, int Hkv, int Smax, int D_, float scale, int splits) {
   constexpr int D = DT;
@@ -535,6 +536,17 @@ __global__ void decode_attn_split_kernel(
   }
   __syncthreads();
 
+  if (splits == 1) {
+    // single split (the 64-slot continuous-serving shape: B*Hkv already
+    // fills the chip): normalize HERE and skip the combine kernel + the
+    // fp32 workspace round-trip entirely
+    for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
+      const int g = i / D, d = i % D;
+      const float den = l_run[g] > 0.f ? l_run[g] : 1.f;
+      out[((long)b * H + hkv * G + g) * D + d] = f2bf(o_sh[i] / den);
+    }
+    return;
+  }
   // write unnormalized partials for this split
   const long base = (((long)b * Hkv + hkv) * splits + split) * G;
   for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
@@ -651,8 +663,8 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
 #define DEC_CASE(GV, DV)                                                      \
   hipLaunchKernelGGL((decode_attn_split_kernel<GV, DV>), grid,                \
                      dim3(DEC_CHUNK), lds, stream, (const bf16*)q,            \
-                     (const bf16*)kc, (const bf16*)vc, ws_o, ws_ml, seq_lens, \
-                     H, Hkv, Smax, D, scale, splits);                         \
+                     (const bf16*)kc, (const bf16*)vc, ws_o, ws_ml,           \
+                     (bf16*)out, seq_lens, H, Hkv, Smax, D, scale, splits);   \
   break;
   switch (G * 1000 + D) {
     case 1064: DEC_CASE(1, 64) case 2064: DEC_CASE(2, 64)
@@ -668,9 +680,11 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
   }
 #undef DEC_CASE
   HIP_CHECK_LAUNCH();
-  hipLaunchKernelGGL(decode_attn_combine_kernel, dim3(H, B), dim3(128), 0,
-                     stream, ws_o, ws_ml, (bf16*)out, H, G, D, splits);
-  HIP_CHECK_LAUNCH();
+  if (splits > 1) {   // splits==1 normalized + wrote out in the split kernel
+    hipLaunchKernelGGL(decode_attn_combine_kernel, dim3(H, B), dim3(128), 0,
+                       stream, ws_o, ws_ml, (bf16*)out, H, G, D, splits);
+    HIP_CHECK_LAUNCH();
+  }
   return hipSuccess;
 }
 

@@ -266,10 +266,11 @@ torch::Tensor decode_attn(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   splits = std::max(1, std::min(splits, max_splits));
   if (const char* ov = std::getenv("SENTIO_DECODE_SPLITS"))
     splits = std::max(1, std::min(atoi(ov), max_splits));
-  auto ws_o = torch::empty({(long)B * Hkv * splits * G * D},
-                           q.options().dtype(torch::kFloat));
-  auto ws_ml = torch::empty({(long)B * Hkv * splits * G * 2},
-                            q.options().dtype(torch::kFloat));
+  // splits==1 writes `out` directly from the split kernel (no combine):
+  // keep only dummy workspaces
+  const long ws_n = splits > 1 ? (long)B * Hkv * splits * G : 1;
+  auto ws_o = torch::empty({ws_n * D}, q.options().dtype(torch::kFloat));
+  auto ws_ml = torch::empty({ws_n * 2}, q.options().dtype(torch::kFloat));
   check_hip(sentio_decode_attn(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                                out.data_ptr(), sl.data_ptr<int>(),
                                ws_o.data_ptr<float>(), ws_ml.data_ptr<float>(),
