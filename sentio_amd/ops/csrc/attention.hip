@@ -562,6 +562,224 @@ __global__ void decode_attn_split_kernel(
   }
 }
 
+// ---------------- cooperative-row decode kernel (occupancy variant) ----
+// Same split-S contract as decode_attn_split_kernel, but phase A loads K
+// rows COOPERATIVELY — a 16-lane group reads one 256 B row per iteration
+// (1 KB coalesced per wave-instruction) — and reduces the per-lane
+// partial dots with group16 shuffles instead of staging K through a
+// 64 KB LDS tile.  LDS drops to ~10 KB and (with half-depth V prefetch)
+// VGPRs to ~170, so residency can rise from 2 to 3 blocks/CU: the PMC
+// wave-state split showed the staged kernel 48% parked on waits/barriers
+// at 2 blocks/CU.  Each lane's q slice (8 dims x G heads) is
+// loop-invariant and lives in registers.  Select with
+// SENTIO_DECODE_COOP=1 (read once at first launch).
+template <int G, int DT>
+__launch_bounds__(DEC_CHUNK, 1)
+__global__ void decode_attn_coop_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ kc,
+    const bf16* __restrict__ vc,
+    float* __restrict__ ws_o, float* __restrict__ ws_ml,
+    bf16* __restrict__ out,
+    const int* __restrict__ seq_lens,
+    int H, int Hkv, int Smax, int D_, float scale, int splits) {
+  constexpr int D = DT;
+  constexpr int ROWB = D / 8;              // lanes that cover one K row
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* p_sh = reinterpret_cast<float*>(smem);            // [G][DEC_CHUNK]
+  float* q_sh = p_sh + G * DEC_CHUNK;                      // [G][D]
+  float* red = q_sh + G * D;                               // [32] scratch
+  float* o_sh = red + 32;                                  // [G][D]
+
+  const int hkv = blockIdx.x;
+  const int b = blockIdx.y;
+  const int split = blockIdx.z;
+  const int slen = seq_lens[b];
+  const int span = (slen + splits - 1) / splits;
+  const int s_begin = split * span;
+  const int s_end = min(slen, s_begin + span);
+  const bf16* kb = kc + ((long)b * Hkv + hkv) * Smax * (long)D;
+  const bf16* vb = vc + ((long)b * Hkv + hkv) * Smax * (long)D;
+
+  for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
+    const int g = i / D, d = i % D;
+    q_sh[i] = bf2f(q[((long)b * H + hkv * G + g) * D + d]);
+    o_sh[i] = 0.f;
+  }
+  __syncthreads();
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int li = lane & 15;                // lane-in-group: dim slice owner
+  const int g4 = lane >> 4;                // group in wave: key owner
+  // loop-invariant per-lane q slice (lanes beyond ROWB duplicate a slice
+  // and are masked out of the reduction)
+  float q_reg[G][8];
+#pragma unroll
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      q_reg[g][e] = q_sh[g * D + (li % ROWB) * 8 + e];
+
+  float m_run[G], l_run[G], alpha[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) { m_run[g] = -INFINITY; l_run[g] = 0.f; }
+
+  const int dgroup = threadIdx.x & 15;
+  const int jslot = threadIdx.x >> 4;
+  const bool dg_ok = dgroup < ROWB;
+  float o_part[G][8];
+#pragma unroll
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o_part[g][e] = 0.f;
+
+  for (int s0 = s_begin; s0 < s_end; s0 += DEC_CHUNK) {
+    const int chunk = min(DEC_CHUNK, s_end - s0);
+    // ---- phase A: cooperative dot.  Wave w owns keys [w*64, w*64+64);
+    // iteration u: group g4 covers key w*64 + u*4 + g4; lane li reads its
+    // 16 B dim slice.  Loads clamp to the valid span (no exec-mask
+    // predication: it forces conservative vmcnt(0) waits); out-of-range
+    // keys are masked in the softmax via row >= chunk.
+    constexpr int AIT = 64 / 4;            // iterations per wave
+    constexpr int ABATCH = 8;              // K rows in flight
+#pragma unroll
+    for (int u0 = 0; u0 < AIT; u0 += ABATCH) {
+      bf16x8 kr[ABATCH];
+#pragma unroll
+      for (int u = 0; u < ABATCH; ++u) {
+        const long key = min((long)(s0 + wid * 64 + (u0 + u) * 4 + g4),
+                             (long)(s_end - 1));
+        kr[u] = nt_load8(reinterpret_cast<const short*>(kb + key * D) +
+                         (li % ROWB) * 8);
+      }
+#pragma unroll
+      for (int u = 0; u < ABATCH; ++u) {
+        float part[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          float acc = 0.f;
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            acc += bits2f(kr[u][e]) * q_reg[g][e];
+          part[g] = (li < ROWB) ? acc : 0.f;
+        }
+#pragma unroll
+        for (int g = 0; g < G; ++g) part[g] = group16_sum(part[g]);
+        if (li == 0) {
+          const int key_local = wid * 64 + (u0 + u) * 4 + g4;
+#pragma unroll
+          for (int g = 0; g < G; ++g)
+            p_sh[g * DEC_CHUNK + key_local] = part[g] * scale;
+        }
+      }
+    }
+    __syncthreads();                       // scores visible to owners
+
+    // ---- V prefetch, first half (second half issues mid-PV)
+    constexpr int JT = DEC_CHUNK / 16;
+    constexpr int JH = JT / 2;
+    bf16x8 v8[JH];
+    auto load_v = [&](int half) {
+#pragma unroll
+      for (int u = 0; u < JH; ++u) {
+        const long j = min((long)(s0 + jslot + (half * JH + u) * 16),
+                           (long)(s_end - 1));
+        v8[u] = nt_load8(reinterpret_cast<const short*>(vb + j * D) +
+                         dgroup * 8);
+      }
+    };
+    if (dg_ok) load_v(0);
+
+    // ---- softmax (thread-per-key; raw score from p_sh, p written back
+    // in place — each thread touches only its own slot before the
+    // publishing barrier)
+    {
+      const int row = threadIdx.x;
+      float sc[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g)
+        sc[g] = (row < chunk) ? p_sh[g * DEC_CHUNK + row] : -INFINITY;
+      float mx[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) mx[g] = sc[g];
+      block_reduce_vec<G, true>(mx, red);
+      float sums[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        const float m_new = fmaxf(m_run[g], mx[g]);
+        float a = (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
+        if (m_new == -INFINITY) a = 1.f;
+        alpha[g] = a;
+        const float p = (sc[g] != -INFINITY) ? __expf(sc[g] - m_new) : 0.f;
+        p_sh[g * DEC_CHUNK + row] = p;
+        sums[g] = p;
+        m_run[g] = m_new;
+      }
+      block_reduce_vec<G, false>(sums, red);  // barrier also publishes p_sh
+#pragma unroll
+      for (int g = 0; g < G; ++g)
+        l_run[g] = l_run[g] * alpha[g] + sums[g];
+    }
+
+    // ---- PV consume in two halves
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) o_part[g][e] *= alpha[g];
+    if (dg_ok) {
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (half) load_v(1);
+#pragma unroll
+        for (int u = 0; u < JH; ++u) {
+          const int j = jslot + (half * JH + u) * 16;
+          if (j >= chunk) continue;        // p is 0 there anyway
+          float vf[8];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) vf[e] = bits2f(v8[u][e]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = p_sh[g * DEC_CHUNK + j];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) o_part[g][e] += p * vf[e];
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (dg_ok) {
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        atomicAdd(&o_sh[g * D + dgroup * 8 + e], o_part[g][e]);
+  }
+  __syncthreads();
+
+  if (splits == 1) {
+    for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
+      const int g = i / D, d = i % D;
+      const float den = l_run[g] > 0.f ? l_run[g] : 1.f;
+      out[((long)b * H + hkv * G + g) * D + d] = f2bf(o_sh[i] / den);
+    }
+    return;
+  }
+  const long base = (((long)b * Hkv + hkv) * splits + split) * G;
+  for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
+    const int g = i / D, d = i % D;
+    ws_o[(base + g) * D + d] = o_sh[i];
+  }
+  if (threadIdx.x == 0) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      ws_ml[(base + g) * 2 + 0] = m_run[g];
+      ws_ml[(base + g) * 2 + 1] = l_run[g];
+    }
+  }
+}
+
 // combine: one block per (b, h); threads over D
 __global__ void decode_attn_combine_kernel(
     const float* __restrict__ ws_o, const float* __restrict__ ws_ml,
@@ -660,11 +878,25 @@ hipError_t sentio_decode_attn(const void* q, const void* kc, const void* vc,
                + (size_t)DEC_CHUNK * D * 2;   // swizzled K stage
   dim3 grid(Hkv, B, splits);
   if (D != 64 && D != 128) return hipErrorInvalidValue;
+  static int use_coop = -1;
+  if (use_coop < 0) {
+    const char* e = std::getenv("SENTIO_DECODE_COOP");
+    use_coop = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (use_coop)
+    lds = (size_t)(G * DEC_CHUNK + 2 * G * D + 32) * sizeof(float);
 #define DEC_CASE(GV, DV)                                                      \
-  hipLaunchKernelGGL((decode_attn_split_kernel<GV, DV>), grid,                \
-                     dim3(DEC_CHUNK), lds, stream, (const bf16*)q,            \
-                     (const bf16*)kc, (const bf16*)vc, ws_o, ws_ml,           \
-                     (bf16*)out, seq_lens, H, Hkv, Smax, D, scale, splits);   \
+  if (use_coop) {                                                             \
+    hipLaunchKernelGGL((decode_attn_coop_kernel<GV, DV>), grid,               \
+                       dim3(DEC_CHUNK), lds, stream, (const bf16*)q,          \
+                       (const bf16*)kc, (const bf16*)vc, ws_o, ws_ml,         \
+                       (bf16*)out, seq_lens, H, Hkv, Smax, D, scale, splits); \
+  } else {                                                                    \
+    hipLaunchKernelGGL((decode_attn_split_kernel<GV, DV>), grid,              \
+                       dim3(DEC_CHUNK), lds, stream, (const bf16*)q,          \
+                       (const bf16*)kc, (const bf16*)vc, ws_o, ws_ml,         \
+                       (bf16*)out, seq_lens, H, Hkv, Smax, D, scale, splits); \
+  }                                                                           \
   break;
   switch (G * 1000 + D) {
     case 1064: DEC_CASE(1, 64) case 2064: DEC_CASE(2, 64)
