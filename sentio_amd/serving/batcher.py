@@ -1,21 +1,24 @@
-"""Dynamic request batching for the generation engine.
+"""Request batching for the serving engines.
 
 The reference served each request's LLM call as its own HTTP round trip to
 a provider that batched internally (reference src/core/llm/providers/
 openai.py:117).  On-device, concurrent `/chat` requests must be batched by
-US: decode is weight-bandwidth-bound, so a batch of 32 decodes costs barely
-more than a batch of 1 (the 16 GB of weights stream once per token either
-way).  This batcher turns per-request `generate()` calls into shared
-batches:
+US: decode is weight-bandwidth-bound, so 64 rows of decode cost barely
+more than 1 (the 16 GB of weights stream once per token either way).
 
-* callers enqueue (prompt, params) and block on a future;
-* a worker thread drains the queue, groups requests by compatible sampling
-  params (temperature bucket, max_tokens), waits up to `max_wait_ms` for
-  stragglers, and runs ONE batched `generator.generate`;
-* results fan back out to the callers.
+Three coalescing layers live here:
 
-Used by ChatHandler when `settings.dynamic_batching` is on; the pipeline
-path is unchanged (it already batches within one bench step).
+* `ContinuousBatcher` (default generation frontend) — TRUE continuous
+  batching: requests join a persistent slot-session decode loop at step
+  boundaries (admission prefill on a side stream, per-row seq_lens and
+  per-row-temperature sampling), stream per step, and free their slot the
+  moment they finish — no wave barrier.
+* `DynamicBatcher` (CONTINUOUS_BATCHING=0 fallback, and the mock-engine
+  path) — wave batching: group by sampling params, keep collecting while
+  the engine runs the previous batch, run ONE batched generate.
+* `MicroBatcher` + `BatchedEncoder`/`BatchedReranker` — concurrent
+  single-query embeds / rerank pair scores coalesce into one engine
+  forward each.
 """
 
 from __future__ import annotations
