@@ -233,7 +233,8 @@ class GeneratorEngine:
         tmp = KVCache(self.cfg, len(prompts), S, self.device,
                       self.model.dtype, n_kv_heads=self.model.hkv_local)
         logits = self.model.prefill(tokens, tmp, lens=lens_t)
-        return {"tmp": tmp, "lens": lens_t, "logits": logits, "S": S}
+        return {"tmp": tmp, "lens": lens_t, "logits": logits, "S": S,
+                "lens_host": lens}
 
     @torch.inference_mode()
     def integrate_admission(self, sess: "_DecodeSession", rows: list[int],
@@ -308,6 +309,9 @@ class GeneratorEngine:
         """
         if not prompts:
             return []
+        # a request cannot generate past the cache: clamp so the prompt
+        # budget stays positive and decode never writes beyond max_seq
+        max_new_tokens = max(1, min(max_new_tokens, self.max_seq - 9))
         import time as _time
 
         from sentio_amd.observability.kernel_timer import get_timer

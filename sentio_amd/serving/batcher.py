@@ -361,8 +361,16 @@ class ContinuousBatcher:
     def _loop(self) -> None:
         try:
             self._loop_inner()
-        except Exception as exc:   # a dead loop must not strand callers
-            while True:
+        except Exception as exc:   # a dead loop must not strand callers:
+            for s in getattr(self, "_live_state", []):   # admitted slots
+                if s is None:
+                    continue
+                it = s["item"]
+                if it.stream_q is not None:
+                    it.stream_q.put(None)
+                if not it.future.done():
+                    it.future.set_exception(exc)
+            while True:            # still-queued requests
                 try:
                     item = self._q.get_nowait()
                 except queue.Empty:
@@ -382,6 +390,7 @@ class ContinuousBatcher:
         dev = gen.device
         sess = gen.make_slot_session(self.n_slots)
         state: list[dict | None] = [None] * self.n_slots
+        self._live_state = state            # _loop's crash handler fails these
         cur = torch.zeros(self.n_slots, dtype=torch.int64, device=dev)
         temps = torch.zeros(self.n_slots, device=dev)
         free_mask = torch.ones(self.n_slots, dtype=torch.bool, device=dev)
@@ -436,12 +445,18 @@ class ContinuousBatcher:
                         logits = gen.integrate_admission(sess, rows, pre,
                                                          idx=take)
                         items = [want[j] for j in take]
+                        lens_host = [pre["lens_host"][j] for j in take]
                         t_adm = torch.tensor([it.temperature for it in items],
                                              device=dev)
                         tok0 = gen.sample_rows(logits, t_adm).cpu().tolist()
                         for j, (r, it) in enumerate(zip(rows, items)):
+                            # decode may never write past the slot cache:
+                            # remaining steps clamp to Smax - prompt_len - 1
+                            cap = max(1, sess.cache.max_seq
+                                      - lens_host[j] - 1)
                             state[r] = {"item": it, "ids": [],
-                                        "remaining": it.max_new_tokens}
+                                        "remaining": min(it.max_new_tokens,
+                                                         cap)}
                             free_mask[r] = False
                             temps[r] = it.temperature
                             cur[r] = tok0[j]

@@ -295,3 +295,37 @@ def test_continuous_batching_stream_and_capacity():
         assert gen.batcher.stats["completed"] >= 7
     finally:
         gen.batcher.stop()
+
+
+def test_continuous_batching_edge_params():
+    """Per-request params vary freely inside one slot loop: oversized
+    max_new_tokens clamps to the cache budget, temperature 0 and >0 mix,
+    stop_on_eos differs — all complete, none corrupts another."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.serving.batcher import ContinuousGenerator
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=96)
+    gen = ContinuousGenerator(eng, slots=3)
+    try:
+        jobs = [
+            ("steady prompt", 8, 0.0, False),
+            ("x", 10_000, 0.0, True),          # max_new >> max_seq: clamps
+            ("warm prompt about graphs", 6, 0.9, True),
+            ("", 4, 0.0, False),               # empty prompt: BOS-only
+            ("last one", 5, 0.5, True),
+        ]
+
+        def run(j):
+            p, mnt, t, eos = j
+            return gen.generate([p], max_new_tokens=mnt, temperature=t,
+                                stop_on_eos=eos)[0]
+
+        with ThreadPoolExecutor(max_workers=5) as ex:
+            outs = list(ex.map(run, jobs))
+        assert len(outs) == 5
+        assert all(isinstance(o, str) for o in outs)
+        assert gen.batcher.stats["completed"] == 5
+    finally:
+        gen.batcher.stop()
