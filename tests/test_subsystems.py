@@ -220,3 +220,45 @@ def test_rate_limit_settings_from_env(monkeypatch):
     s = Settings.from_env()
     assert s.rate_limit_chat_per_min == 555
     assert s.rate_limit_embed_per_min == 44
+
+
+def test_r2_serving_flags_from_env(monkeypatch):
+    """Round-2 batching knobs flow through the env config layer."""
+    from sentio_amd.config import Settings
+
+    monkeypatch.setenv("CONTINUOUS_BATCHING", "false")
+    monkeypatch.setenv("DYNAMIC_BATCHING", "true")
+    monkeypatch.setenv("MAX_BATCH_SIZE", "48")
+    s = Settings.from_env()
+    assert s.continuous_batching is False
+    assert s.dynamic_batching is True
+    assert s.max_batch_size == 48
+    monkeypatch.setenv("CONTINUOUS_BATCHING", "true")
+    assert Settings.from_env().continuous_batching is True
+
+
+def test_frontend_selection_by_flags(monkeypatch):
+    """Container picks the continuous frontend for real engines, the wave
+    batcher when continuous is off, and never wraps mocks in the slot
+    loop (mock engines lack slot sessions)."""
+    from sentio_amd.config import Settings
+    from sentio_amd.serving.container import ServiceContainer
+
+    s = Settings()
+    s.mock_compute = True
+    s.device = "cpu"
+    s.dynamic_batching = True
+    s.continuous_batching = True
+    c = ServiceContainer(s)
+    fe = c.generator_frontend()
+    from sentio_amd.serving.batcher import BatchedGenerator
+
+    assert isinstance(fe, BatchedGenerator)   # mock -> wave batcher
+
+    s2 = Settings()
+    s2.mock_compute = True
+    s2.device = "cpu"
+    s2.dynamic_batching = False
+    c2 = ServiceContainer(s2)
+    fe2 = c2.generator_frontend()
+    assert not hasattr(fe2, "batcher")        # raw engine passthrough
