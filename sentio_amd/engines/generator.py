@@ -285,11 +285,17 @@ class GeneratorEngine:
         """Per-ROW-temperature sampling — continuous batches mix requests
         with different temperatures: greedy rows take argmax, the rest
         Gumbel-argmax (torch eager; the split sampling kernel takes one
-        scalar temperature and stays on the uniform-batch path)."""
+        scalar temperature and stays on the uniform-batch path).  Noise
+        comes from a SEEDED per-engine generator so stochastic decodes are
+        reproducible, matching the wave path's seeded sample_token."""
         self._step_seed += 1
+        rng = getattr(self, "_sample_rng", None)
+        if rng is None or str(rng.device) != str(logits.device):
+            rng = self._sample_rng = torch.Generator(device=logits.device)
+            rng.manual_seed(909)
         t = temps.to(logits.device).view(-1, 1)
-        g = -torch.log(-torch.log(
-            torch.rand_like(logits).clamp_min(1e-20)).clamp_min(1e-20))
+        u = torch.rand(logits.shape, device=logits.device, generator=rng)
+        g = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))
         scores = torch.where(t > 0, logits / t.clamp_min(1e-6) + g, logits)
         return scores.argmax(-1)
 
