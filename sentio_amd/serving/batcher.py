@@ -333,8 +333,11 @@ class ContinuousBatcher:
                 except queue.Empty:
                     break
             try:
+                from sentio_amd.observability.kernel_timer import get_timer
+
                 if side is not None:
-                    with torch.cuda.stream(side):
+                    with torch.cuda.stream(side), \
+                            get_timer("admission_prefill").measure():
                         pre = gen.prefill_admission(
                             [it.prompt for it in want],
                             [it.max_new_tokens for it in want])
@@ -386,6 +389,9 @@ class ContinuousBatcher:
 
         from sentio_amd.engines.tokenizer import EOS_ID
 
+        from sentio_amd.observability.kernel_timer import get_timer
+
+        _step_timer = get_timer("decode_step")
         gen = self.generator
         dev = gen.device
         sess = gen.make_slot_session(self.n_slots)
@@ -479,7 +485,8 @@ class ContinuousBatcher:
             # free rows must never advance their KV write position off the
             # cache end: re-zero their seq_lens each step
             sess.cache.seq_lens.masked_fill_(free_mask, 0)
-            logits = gen.decode_step_session(sess, cur)
+            with _step_timer.measure():   # lazy HIP events, no sync
+                logits = gen.decode_step_session(sess, cur)
             toks = gen.sample_rows(logits, temps)
             cur.copy_(toks)
             self.stats["steps"] += 1
