@@ -348,12 +348,16 @@ class ContinuousBatcher:
                         [it.prompt for it in want],
                         [it.max_new_tokens for it in want])
                     ev = None
+                placed = False
                 while not self._stop.is_set():
                     try:
                         self._ready_q.put((want, pre, ev), timeout=0.2)
+                        placed = True
                         break
                     except queue.Full:
                         continue
+                if not placed:     # shut down while the ready queue was
+                    raise RuntimeError("batcher stopped")   # full: fail, not drop
             except Exception as exc:
                 for it in want:
                     if it.stream_q is not None:
@@ -494,6 +498,27 @@ class ContinuousBatcher:
             for r in range(self.n_slots):
                 if state[r] is not None:
                     feed(r, toks_host[r])
+
+        # graceful stop mid-traffic: active slots and any un-integrated
+        # admission must fail fast, not block callers until their timeout
+        err = RuntimeError("batcher stopped")
+        for r in range(self.n_slots):
+            if state[r] is None:
+                continue
+            it = state[r]["item"]
+            state[r] = None
+            if it.stream_q is not None:
+                it.stream_q.put(None)
+            if not it.future.done():
+                it.future.set_exception(err)
+        if pending is not None:
+            want, _pre, _ev, left = pending
+            for j in left:
+                it = want[j]
+                if it.stream_q is not None:
+                    it.stream_q.put(None)
+                if not it.future.done():
+                    it.future.set_exception(err)
 
     def health(self) -> dict[str, Any]:
         return {"queued": self._q.qsize(), "mode": "continuous",

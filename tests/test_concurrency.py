@@ -388,3 +388,35 @@ def test_continuous_batching_randomized_storm():
                             temperature=0.0)[0] is not None
     finally:
         gen.batcher.stop()
+
+
+def test_continuous_batching_stop_mid_traffic_unblocks_callers():
+    """stop() during active decode fails in-flight requests promptly
+    instead of stranding their futures until the client timeout."""
+    import threading as _th
+    import time as _t
+
+    from sentio_amd.engines.generator import GeneratorEngine
+    from sentio_amd.serving.batcher import ContinuousGenerator
+
+    eng = GeneratorEngine("tiny-decoder64", device="cpu", max_seq=96)
+    gen = ContinuousGenerator(eng, slots=2)
+    results = {}
+
+    def run():
+        try:
+            results["out"] = gen.generate(["long running request"],
+                                          max_new_tokens=5000,
+                                          temperature=0.0,
+                                          stop_on_eos=False)[0]
+        except Exception as exc:
+            results["exc"] = exc
+
+    t = _th.Thread(target=run)
+    t.start()
+    _t.sleep(0.2)                 # request admitted and decoding
+    gen.batcher.stop()
+    t.join(timeout=10)
+    assert not t.is_alive()
+    # either it finished just before the stop, or it failed FAST
+    assert "out" in results or isinstance(results.get("exc"), RuntimeError)
