@@ -121,6 +121,15 @@ def create_app(settings: Settings | None = None,
         yield
         _tracing.stop_otlp_exporter()
         hc.stop()
+        # drain the batching workers so SIGTERM shutdown (k8s) is clean:
+        # in-flight requests were already completed by uvicorn's drain
+        fe = container._cache.get("generator_frontend")
+        if fe is not None and hasattr(fe, "batcher"):
+            fe.batcher.stop()
+        for eng_key in ("encoder", "reranker"):
+            micro = getattr(container._cache.get(eng_key), "micro", None)
+            if micro is not None:
+                micro.stop()
 
     app = FastAPI(title="sentio-amd", version=__import__("sentio_amd").__version__,
                   lifespan=lifespan)
