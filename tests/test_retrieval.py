@@ -180,3 +180,55 @@ def test_web_cache_errors_are_soft():
     h = HybridRetriever(dense=main, cache_retriever=_Boom())
     out = h.retrieve("q", top_k=2)
     assert len(out) == 2 and main.calls == 1
+
+
+def test_retrieval_quality_harness_sane():
+    """The quality harness (scripts/retrieval_quality.py) is deterministic
+    and orders the retrievers sensibly on the lexical topic task: BM25
+    near-perfect, hybrid >= dense (fusion must not destroy the lexical
+    signal)."""
+    import importlib.util
+    import os
+
+    spec = importlib.util.spec_from_file_location(
+        "retrieval_quality",
+        os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), "scripts", "retrieval_quality.py"))
+    rq = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(rq)
+
+    import numpy as np
+
+    from sentio_amd.engines.encoder import EncoderEngine
+    from sentio_amd.index.bm25 import BM25Index
+    from sentio_amd.index.dense import DenseIndex
+    from sentio_amd.models.document import Document
+    from sentio_amd.retrieval.dense import DenseRetriever
+    from sentio_amd.retrieval.hybrid import HybridRetriever
+    from sentio_amd.retrieval.sparse import BM25Retriever
+
+    rng = np.random.RandomState(71)
+    texts, labels = rq.build_corpus(180, rng)
+    queries = rq.build_queries(24, rng)
+    docs = [Document(text=t, metadata={"topic": lab}, id=f"d{i}")
+            for i, (t, lab) in enumerate(zip(texts, labels))]
+    by_id = {d.id: d.metadata["topic"] for d in docs}
+
+    enc = EncoderEngine("tiny-encoder", device="cpu", max_seq=128)
+    didx = DenseIndex(dim=enc.dim, device="cpu")
+    didx.add(docs, enc.embed(texts))
+    bidx = BM25Index()
+    bidx.build([d.id for d in docs], texts)
+    dense = DenseRetriever(enc, didx)
+    sparse = BM25Retriever(bidx, doc_lookup=didx.get_document)
+
+    r_bm25 = rq.evaluate(sparse, queries, by_id, 10)
+    r_dense = rq.evaluate(dense, queries, by_id, 10)
+    hyb = HybridRetriever(dense=dense, sparse=sparse, fusion_method="rrf")
+    r_hyb = rq.evaluate(hyb, queries, by_id, 10)
+
+    assert r_bm25["recall_at_k"] >= 0.95          # lexical task
+    assert 0.0 <= r_dense["recall_at_k"] <= 1.0
+    assert r_hyb["recall_at_k"] >= r_dense["recall_at_k"] - 0.05
+    # determinism
+    assert rq.evaluate(sparse, queries, by_id, 10) == r_bm25
