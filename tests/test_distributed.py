@@ -280,3 +280,28 @@ def _heartbeat_worker(rank, world):
 
 def test_heartbeat_gather_world2():
     _run_workers(_heartbeat_worker, world=2)
+
+
+def test_bench_world2_cpu_end_to_end():
+    """bench.py under torch.distributed.run with 2 CPU ranks (gloo): the
+    driver launches exactly this shape on GPUs for the scaling curve —
+    the collective merge paths must work rank-parallel end to end."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    port = _free_port()
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(root, "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    line = [l for l in proc.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["metric"] == "chat_qps" and out["value"] > 0
+    assert out["config"]["total_docs"] == out["config"]["docs_per_gpu"] * 2
+    assert ">verify" in out["config"]["pipeline"]
